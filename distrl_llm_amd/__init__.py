@@ -1,0 +1,13 @@
+"""distrl_llm_amd — MI355X-native distributed RL fine-tuning engine for LLMs.
+
+A from-scratch CDNA4 (gfx950) implementation of the capabilities of
+BY571/DistRL-LLM: heterogeneous actor/learner GPU pools, batched paged-KV
+rollout generation, 4-bit (nf4) LoRA training with 8-bit Adam, PG and GRPO
+learners, top-k candidate subselection, PEFT-format LoRA checkpoints and the
+same ``train_distributed.py`` CLI — built on PyTorch-ROCm + hand-written
+HIP/CDNA4 kernels + RCCL over xGMI (no Ray, no vLLM, no Unsloth, no Triton).
+"""
+
+__version__ = "0.1.0"
+
+from .config import SamplingParams, GenerationConfig  # noqa: F401

@@ -1,0 +1,360 @@
+"""Paged-KV generation engine: prefill + continuous-batching decode.
+
+Native replacement for the reference's in-process vLLM engine
+(`policy.fast_generate`, reference distributed_actor.py:147-172): batched
+sampling with a paged KV cache, n-candidate fan-out per prompt (prompt KV
+blocks shared across the n candidates via refcounting — prefill runs once
+per prompt, not n times), per-call SamplingParams, and EOS/max-token
+termination. Decode is the hot loop; on GPU every custom
+op routes through the gfx950 HIP extension via ops.functional and the
+step can be hipGraph-captured (engine.graph).
+
+The engine reads the SAME weight/LoRA tensors the learner trains, so
+in-process weight sync is free and cross-process sync is one RCCL
+broadcast into these tensors.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Dict, List, Optional, Sequence as Seq
+
+import torch
+import torch.nn.functional as F
+
+from ..config import EngineConfig, SamplingParams
+from ..models.model import CausalLM
+from ..ops import functional as OF
+from ..ops import reference as R
+from .kvcache import BlockAllocator, KVCachePool, Sequence
+
+
+class Engine:
+    def __init__(self, model: CausalLM, cfg: EngineConfig,
+                 device: Optional[torch.device] = None, seed: int = 0):
+        self.model = model
+        self.cfg = cfg
+        self.spec = model.spec
+        self.device = device if device is not None else model.device
+        self.dtype = model.dtype_
+        self.scale = self.spec.head_dim ** -0.5
+        self._seq_counter = 0
+        self.generator = torch.Generator(device=self.device)
+        self.generator.manual_seed(seed)
+
+        num_blocks = cfg.num_kv_blocks
+        if num_blocks <= 0:
+            num_blocks = self._derive_num_blocks()
+        self.pool = KVCachePool(self.spec.num_layers, num_blocks,
+                                cfg.kv_block_size, self.spec.num_kv_heads,
+                                self.spec.head_dim, self.dtype, self.device)
+        self._graph_runner = None
+
+    # ------------------------------------------------------------ sizing
+
+    def _derive_num_blocks(self) -> int:
+        """Size the KV pool from free HBM (re-derived for 288 GB per GPU,
+        not copied from the reference's 24 GB fractions — SURVEY.md
+        §2.6-10)."""
+        s = self.spec
+        bs = self.cfg.kv_block_size
+        per_block = KVCachePool.pool_size_bytes(1, s.num_layers, bs,
+                                                s.num_kv_heads, s.head_dim,
+                                                self.dtype)
+        if self.device.type == "cuda":
+            free, _total = torch.cuda.mem_get_info(self.device)
+            budget = int(free * self.cfg.gpu_memory_utilization)
+        else:
+            budget = 64 * 1024 * 1024  # CPU tests: 64 MB pool
+        n = max(budget // per_block, 16)
+        max_needed = self.cfg.max_num_seqs * KVCachePool.blocks_for(
+            self.cfg.max_seq_length, bs)
+        return int(min(n, max_needed))
+
+    # ---------------------------------------------------------- weights
+
+    def _proj(self, mod, x):
+        """Projection through a LoRALinear (base GEMM + adapter)."""
+        return mod(x)
+
+    # ---------------------------------------------------------- prefill
+
+    @torch.no_grad()
+    def _prefill_batch(self, seqs: List[Sequence]) -> torch.Tensor:
+        """Run the prompt phase for a batch of fresh sequences (one per
+        prompt; fan-out happens after). Writes prompt KV into the pool and
+        returns last-token logits (len(seqs), V)."""
+        s = self.spec
+        device = self.device
+        lens = [len(q.prompt_ids) for q in seqs]
+        total = sum(lens)
+
+        # flat varlen layout
+        input_ids = torch.tensor([t for q in seqs for t in q.prompt_ids],
+                                 dtype=torch.long, device=device)
+        positions = torch.tensor([p for L in lens for p in range(L)],
+                                 dtype=torch.long, device=device)
+        slot_mapping = []
+        for q in seqs:
+            L = len(q.prompt_ids)
+            nb = KVCachePool.blocks_for(L, self.pool.block_size)
+            q.block_table = [self.pool.allocator.alloc() for _ in range(nb)]
+            for p in range(L):
+                blk = q.block_table[p // self.pool.block_size]
+                slot_mapping.append(blk * self.pool.block_size
+                                    + p % self.pool.block_size)
+            q.context_len = L
+        slot_mapping = torch.tensor(slot_mapping, dtype=torch.long, device=device)
+
+        cos, sin = R.rope_cos_sin(positions, s.head_dim, s.rope_theta,
+                                  device=device)
+
+        x = self.model.model.embed_tokens(input_ids)
+        for li, layer in enumerate(self.model.model.layers):
+            h = layer.input_layernorm(x)
+            q = self._proj(layer.self_attn.q_proj, h).view(total, s.num_heads, s.head_dim)
+            k = self._proj(layer.self_attn.k_proj, h).view(total, s.num_kv_heads, s.head_dim)
+            v = self._proj(layer.self_attn.v_proj, h).view(total, s.num_kv_heads, s.head_dim)
+            q, k = R.apply_rope(q, k, cos, sin)
+            OF.kv_cache_scatter(k, v, self.pool.key[li], self.pool.value[li],
+                                slot_mapping)
+            o = self._prefill_attention(q, k, v, lens)
+            x = x + self._proj(layer.self_attn.o_proj, o.reshape(total, s.q_size))
+            h2 = layer.post_attention_layernorm(x)
+            g = self._proj(layer.mlp.gate_proj, h2)
+            u = self._proj(layer.mlp.up_proj, h2)
+            x = x + self._proj(layer.mlp.down_proj, OF.silu_mul(g, u))
+        x = self.model.model.norm(x)
+
+        last_idx = torch.tensor([sum(lens[:i + 1]) - 1 for i in range(len(lens))],
+                                dtype=torch.long, device=device)
+        return self.model.logits(x[last_idx])
+
+    def _prefill_attention(self, q, k, v, lens: List[int]) -> torch.Tensor:
+        """Causal attention over concatenated prompts. GPU: torch SDPA per
+        prompt (library path; custom flash prefill kernel is an upgrade
+        path — decode is the hot loop); CPU: reference varlen."""
+        if q.is_cuda:
+            group = self.spec.num_heads // self.spec.num_kv_heads
+            outs = []
+            start = 0
+            for L in lens:
+                qs = q[start:start + L].transpose(0, 1).unsqueeze(0)
+                ks = k[start:start + L].repeat_interleave(group, 1).transpose(0, 1).unsqueeze(0)
+                vs = v[start:start + L].repeat_interleave(group, 1).transpose(0, 1).unsqueeze(0)
+                o = F.scaled_dot_product_attention(qs, ks, vs, is_causal=True,
+                                                   scale=self.scale)
+                outs.append(o.squeeze(0).transpose(0, 1))
+                start += L
+            return torch.cat(outs, 0)
+        return R.varlen_prefill_attention(q, k, v, lens, self.scale)
+
+    # ------------------------------------------------------------ decode
+
+    @torch.no_grad()
+    def _decode_step(self, seqs: List[Sequence]) -> torch.Tensor:
+        """One decode iteration for all running sequences: forward their
+        latest sampled token, write its KV, return next-token logits."""
+        s = self.spec
+        device = self.device
+        N = len(seqs)
+        bs = self.pool.block_size
+
+        tokens, positions, slots, ctx_lens = [], [], [], []
+        max_nb = 0
+        for q in seqs:
+            pos = q.total_len - 1
+            tokens.append(q.output_ids[-1] if q.output_ids else q.prompt_ids[-1])
+            positions.append(pos)
+            blk_idx = pos // bs
+            if blk_idx == len(q.block_table):
+                q.block_table.append(self.pool.allocator.alloc())
+            slots.append(q.block_table[blk_idx] * bs + pos % bs)
+            q.context_len = pos + 1
+            ctx_lens.append(q.context_len)
+            max_nb = max(max_nb, len(q.block_table))
+
+        input_ids = torch.tensor(tokens, dtype=torch.long, device=device)
+        positions_t = torch.tensor(positions, dtype=torch.long, device=device)
+        slot_mapping = torch.tensor(slots, dtype=torch.long, device=device)
+        block_tables = torch.zeros(N, max_nb, dtype=torch.int32, device=device)
+        for i, q in enumerate(seqs):
+            block_tables[i, :len(q.block_table)] = torch.tensor(
+                q.block_table, dtype=torch.int32)
+        context_lens = torch.tensor(ctx_lens, dtype=torch.int32, device=device)
+
+        cos, sin = R.rope_cos_sin(positions_t, s.head_dim, s.rope_theta,
+                                  device=device)
+
+        x = self.model.model.embed_tokens(input_ids)
+        for li, layer in enumerate(self.model.model.layers):
+            h = layer.input_layernorm(x)
+            q = self._proj(layer.self_attn.q_proj, h).view(N, s.num_heads, s.head_dim)
+            k = self._proj(layer.self_attn.k_proj, h).view(N, s.num_kv_heads, s.head_dim)
+            v = self._proj(layer.self_attn.v_proj, h).view(N, s.num_kv_heads, s.head_dim)
+            q, k = R.apply_rope(q, k, cos, sin)
+            OF.kv_cache_scatter(k, v, self.pool.key[li], self.pool.value[li],
+                                slot_mapping)
+            o = OF.paged_attention_decode(q, self.pool.key[li], self.pool.value[li],
+                                          block_tables, context_lens, self.scale)
+            x = x + self._proj(layer.self_attn.o_proj, o.reshape(N, s.q_size))
+            h2 = layer.post_attention_layernorm(x)
+            g = self._proj(layer.mlp.gate_proj, h2)
+            u = self._proj(layer.mlp.up_proj, h2)
+            x = x + self._proj(layer.mlp.down_proj, OF.silu_mul(g, u))
+        x = self.model.model.norm(x)
+        return self.model.logits(x)
+
+    # ----------------------------------------------------------- forking
+
+    def _fork(self, parent: Sequence, n: int, first_tokens: List[int]) -> List[Sequence]:
+        """Create n candidate sequences sharing the parent's full prompt
+        blocks; a partial tail block is copied per candidate."""
+        bs = self.pool.block_size
+        L = len(parent.prompt_ids)
+        full = L // bs  # number of completely-filled blocks
+        children = []
+        for i in range(n):
+            self._seq_counter += 1
+            child = Sequence(self._seq_counter, parent.prompt_ids,
+                             parent.parent_prompt)
+            table = []
+            for b in parent.block_table[:full]:
+                self.pool.allocator.incref(b)
+                table.append(b)
+            if full < len(parent.block_table):
+                nb = self.pool.allocator.alloc()
+                self.pool.copy_block(parent.block_table[full], nb)
+                table.append(nb)
+            child.block_table = table
+            child.context_len = L
+            child.output_ids = [first_tokens[i]]
+            children.append(child)
+        # parent's own references are released (children hold their own)
+        for b in parent.block_table:
+            self.pool.allocator.free(b)
+        parent.block_table = []
+        return children
+
+    def _finish(self, seq: Sequence) -> None:
+        seq.finished = True
+        for b in seq.block_table:
+            self.pool.allocator.free(b)
+        seq.block_table = []
+
+    # ---------------------------------------------------------- generate
+
+    @torch.no_grad()
+    def generate(self, prompts: List[List[int]], sp: SamplingParams,
+                 eos_token_id: Optional[int] = None,
+                 prefill_token_budget: int = 8192) -> List[List[List[int]]]:
+        """Generate sp.n completions per prompt.
+
+        prompts: token-id lists. Returns per prompt a list of n output
+        token-id lists (EOS included when emitted, like vLLM's
+        ``o.token_ids``).
+        """
+        was_training = self.model.training
+        self.model.eval()
+        try:
+            return self._generate_inner(prompts, sp, eos_token_id,
+                                        prefill_token_budget)
+        finally:
+            if was_training:
+                self.model.train()
+
+    def _generate_inner(self, prompts, sp, eos_token_id, prefill_token_budget):
+        bs = self.pool.block_size
+        max_total = self.cfg.max_seq_length
+        results: List[List[List[int]]] = [[] for _ in prompts]
+
+        waiting = list(range(len(prompts)))
+        running: List[Sequence] = []
+
+        def blocks_needed(pi: int) -> int:
+            L = min(len(prompts[pi]), max_total - 1)
+            worst = min(L + sp.max_tokens, max_total)
+            return (KVCachePool.blocks_for(L, bs)
+                    + sp.n * (KVCachePool.blocks_for(worst, bs)
+                              - L // bs))
+
+        def future_need() -> int:
+            """Blocks the already-running sequences may still allocate
+            (worst case) — reserved so decode never exhausts the pool."""
+            need = 0
+            for q in running:
+                worst = min(len(q.prompt_ids) + sp.max_tokens, max_total)
+                need += KVCachePool.blocks_for(worst, bs) - len(q.block_table)
+            return need
+
+        def try_admit():
+            """Prefill + fork as many waiting prompts as memory allows
+            (continuous batching admission)."""
+            batch: List[Sequence] = []
+            batch_tokens = 0
+            reserved = future_need()
+            while waiting:
+                pi = waiting[0]
+                need = blocks_needed(pi)
+                L = min(len(prompts[pi]), max_total - 1)
+                if need > self.pool.allocator.num_free - reserved - len(batch):
+                    break
+                reserved += need
+                if batch and batch_tokens + L > prefill_token_budget:
+                    break
+                if len(running) + len(batch) * sp.n > self.cfg.max_num_seqs:
+                    break
+                waiting.pop(0)
+                self._seq_counter += 1
+                seq = Sequence(self._seq_counter, prompts[pi][:L], pi)
+                batch.append(seq)
+                batch_tokens += L
+            if not batch:
+                return
+            logits = self._prefill_batch(batch)
+            for i, parent in enumerate(batch):
+                lg = logits[i:i + 1].expand(sp.n, -1).contiguous()
+                first = OF.sample_tokens(lg, sp.temperature, sp.top_p, sp.top_k,
+                                         generator=self.generator)
+                children = self._fork(parent, sp.n, first.tolist())
+                for c in children:
+                    if ((eos_token_id is not None
+                         and c.output_ids[-1] == eos_token_id)
+                            or len(c.output_ids) >= sp.max_tokens
+                            or c.total_len >= max_total):
+                        # still need its KV? No: sequence is done.
+                        results[c.parent_prompt].append(c.output_ids)
+                        self._finish(c)
+                    else:
+                        running.append(c)
+
+        try_admit()
+        while running or waiting:
+            if not running:
+                try_admit()
+                if not running:
+                    if waiting:
+                        raise MemoryError(
+                            "KV pool too small to admit any waiting prompt")
+                    break
+            logits = self._decode_step(running)
+            next_tokens = OF.sample_tokens(logits, sp.temperature, sp.top_p,
+                                           sp.top_k, generator=self.generator)
+            next_list = next_tokens.tolist()
+            still = []
+            for i, q in enumerate(running):
+                t = next_list[i]
+                q.output_ids.append(t)
+                done = ((eos_token_id is not None and t == eos_token_id)
+                        or len(q.output_ids) >= sp.max_tokens
+                        or q.total_len >= max_total)
+                if done:
+                    results[q.parent_prompt].append(q.output_ids)
+                    self._finish(q)
+                else:
+                    still.append(q)
+            running = still
+            if waiting:
+                try_admit()
+
+        return results

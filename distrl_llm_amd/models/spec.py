@@ -1,0 +1,84 @@
+"""Model architecture registry.
+
+Covers the model families the reference targets (BASELINE.json configs:
+Qwen2.5-{0.5B,7B,32B}, Llama-3-8B) plus tiny variants for CPU tests. Specs
+are resolved from the model-name string the CLI passes (the reference
+resolves names through HF/Unsloth; offline we map known names to
+architectures and random-init weights — BASELINE.json: synthetic prompts /
+random-init weights).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+
+@dataclass(frozen=True)
+class ModelSpec:
+    name: str
+    vocab_size: int
+    hidden_size: int
+    intermediate_size: int
+    num_layers: int
+    num_heads: int
+    num_kv_heads: int
+    head_dim: int
+    rope_theta: float
+    rms_norm_eps: float
+    tie_word_embeddings: bool
+    qkv_bias: bool  # Qwen2 uses biases on q/k/v projections
+    max_position: int = 32768
+
+    @property
+    def q_size(self) -> int:
+        return self.num_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.num_kv_heads * self.head_dim
+
+
+_REGISTRY = {
+    "qwen2.5-0.5b": ModelSpec("qwen2.5-0.5b", 151936, 896, 4864, 24, 14, 2, 64,
+                              1e6, 1e-6, True, True),
+    "qwen2.5-1.5b": ModelSpec("qwen2.5-1.5b", 151936, 1536, 8960, 28, 12, 2, 128,
+                              1e6, 1e-6, True, True),
+    "qwen2.5-7b": ModelSpec("qwen2.5-7b", 152064, 3584, 18944, 28, 28, 4, 128,
+                            1e6, 1e-6, False, True),
+    "qwen2.5-32b": ModelSpec("qwen2.5-32b", 152064, 5120, 27648, 64, 40, 8, 128,
+                             1e6, 1e-5, False, True),
+    "llama-3-8b": ModelSpec("llama-3-8b", 128256, 4096, 14336, 32, 32, 8, 128,
+                            5e5, 1e-5, False, False),
+    # tiny models for CPU tests / the gloo plumbing config
+    "tiny-qwen2": ModelSpec("tiny-qwen2", 2048, 64, 128, 2, 4, 2, 16,
+                            1e4, 1e-6, True, True, max_position=512),
+    "tiny-llama": ModelSpec("tiny-llama", 2048, 64, 128, 2, 4, 2, 16,
+                            1e4, 1e-5, False, False, max_position=512),
+}
+
+
+def get_spec(model_name: str) -> ModelSpec:
+    """Resolve a model-name string (e.g. the reference's default
+    'unsloth/Qwen2.5-7B-Instruct-bnb-4bit') to an architecture spec."""
+    low = model_name.lower()
+    for key in ("tiny-qwen2", "tiny-llama"):
+        if key in low:
+            return _REGISTRY[key]
+    if "qwen2.5-0.5b" in low or "qwen2-0.5b" in low:
+        return _REGISTRY["qwen2.5-0.5b"]
+    if "qwen2.5-1.5b" in low:
+        return _REGISTRY["qwen2.5-1.5b"]
+    if "qwen2.5-7b" in low:
+        return _REGISTRY["qwen2.5-7b"]
+    if "qwen2.5-32b" in low:
+        return _REGISTRY["qwen2.5-32b"]
+    if "llama-3" in low and "8b" in low:
+        return _REGISTRY["llama-3-8b"]
+    raise ValueError(f"Unknown model architecture for name: {model_name!r}; "
+                     f"known: {sorted(_REGISTRY)}")
+
+
+def is_4bit_model_name(model_name: str) -> bool:
+    """The reference selects 4-bit via the '-bnb-4bit' model suffix and the
+    LOAD_IN_4BIT constant (reference distributed_actor.py:17)."""
+    return "4bit" in model_name.lower()

@@ -1,0 +1,226 @@
+"""Dispatch layer: HIP/CDNA4 kernels on ROCm devices, torch reference on CPU.
+
+Policy (see package docstring): a CUDA(ROCm) tensor REQUIRES the compiled
+gfx950 extension — if it is missing the op raises instead of silently
+falling back to eager (so GPU tests can never pass on a non-native path).
+Set ``DISTRL_ALLOW_EAGER_GPU=1`` only for debugging.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference as R
+from .build import get_extension, extension_available
+
+
+def _require_ext(op: str):
+    ext = get_extension()
+    if ext is None:
+        if os.environ.get("DISTRL_ALLOW_EAGER_GPU", "0") == "1":
+            return None
+        raise RuntimeError(
+            f"distrl_llm_amd op '{op}' called on a GPU tensor but the gfx950 "
+            f"extension is not built/loadable. Run `python -m "
+            f"distrl_llm_amd.ops.build` (or __graft_entry__.build()) first.")
+    return ext
+
+
+# --------------------------------------------------------------- rmsnorm
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        ext = _require_ext("rmsnorm")
+        if ext is None:
+            y = R.rmsnorm(x, weight, eps)
+            ctx.save_for_backward(x, weight)
+            ctx.eps = eps
+            ctx.used_ext = False
+            return y
+        y = ext.rmsnorm_fwd(x, weight, eps)
+        ctx.save_for_backward(x, weight)
+        ctx.eps = eps
+        ctx.used_ext = True
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        if ctx.used_ext:
+            ext = get_extension()
+            dx = ext.rmsnorm_bwd(dy.contiguous(), x, weight, ctx.eps)
+        else:
+            x32 = x.float()
+            w32 = weight.float()
+            dyw = dy.float() * w32
+            var = x32.pow(2).mean(-1, keepdim=True)
+            r = torch.rsqrt(var + ctx.eps)
+            dot = (dyw * x32).mean(-1, keepdim=True)
+            dx = (dyw * r - x32 * dot * r.pow(3)).to(x.dtype)
+        return dx, None, None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if not x.is_cuda:
+        return R.rmsnorm(x, weight, eps)
+    if not torch.is_grad_enabled() or not x.requires_grad:
+        ext = _require_ext("rmsnorm")
+        if ext is None:
+            return R.rmsnorm(x, weight, eps)
+        return ext.rmsnorm_fwd(x.contiguous(), weight, eps)
+    return _RMSNormFn.apply(x.contiguous(), weight, eps)
+
+
+# -------------------------------------------------------------- silu_mul
+
+class _SiluMulFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        ctx.save_for_backward(gate, up)
+        if gate.is_cuda:
+            ext = _require_ext("silu_mul")
+            if ext is not None:
+                return ext.silu_mul_fwd(gate.contiguous(), up.contiguous())
+        return R.silu_mul(gate, up)
+
+    @staticmethod
+    def backward(ctx, dy):
+        gate, up = ctx.saved_tensors
+        if gate.is_cuda and extension_available():
+            ext = get_extension()
+            dg, du = ext.silu_mul_bwd(dy.contiguous(), gate.contiguous(), up.contiguous())
+            return dg, du
+        g32 = gate.float()
+        sig = torch.sigmoid(g32)
+        silu = g32 * sig
+        dsilu = sig * (1 + g32 * (1 - sig))
+        dg = (dy.float() * up.float() * dsilu).to(gate.dtype)
+        du = (dy.float() * silu).to(up.dtype)
+        return dg, du
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if not gate.is_cuda:
+        return R.silu_mul(gate, up)
+    if not torch.is_grad_enabled() or not (gate.requires_grad or up.requires_grad):
+        ext = _require_ext("silu_mul")
+        if ext is None:
+            return R.silu_mul(gate, up)
+        return ext.silu_mul_fwd(gate.contiguous(), up.contiguous())
+    return _SiluMulFn.apply(gate, up)
+
+
+# ------------------------------------------------------------------ rope
+
+def apply_rope_inplace(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
+                       head_dim: int, theta: float) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Inference-path RoPE: q (T, H, D), k (T, KV, D), positions (T,).
+    On GPU uses the fused kernel (in-place); on CPU returns rotated copies."""
+    if q.is_cuda:
+        ext = _require_ext("rope")
+        if ext is not None:
+            ext.rope_inplace(q, k, positions.to(torch.int32), float(theta))
+            return q, k
+    cos, sin = R.rope_cos_sin(positions, head_dim, theta, device=q.device)
+    return R.apply_rope(q, k, cos, sin)
+
+
+# ------------------------------------------------------------- kv cache
+
+def kv_cache_scatter(k, v, key_cache, value_cache, slot_mapping):
+    if k.is_cuda:
+        ext = _require_ext("kv_cache_scatter")
+        if ext is not None:
+            ext.kv_cache_scatter(k.contiguous(), v.contiguous(), key_cache,
+                                 value_cache, slot_mapping.to(torch.int32))
+            return
+    R.kv_cache_scatter(k, v, key_cache, value_cache, slot_mapping)
+
+
+# ---------------------------------------------------- paged attn decode
+
+def paged_attention_decode(q, key_cache, value_cache, block_tables,
+                           context_lens, scale: float):
+    if q.is_cuda:
+        ext = _require_ext("paged_attention_decode")
+        if ext is not None:
+            return ext.paged_attention_decode(
+                q.contiguous(), key_cache, value_cache,
+                block_tables.to(torch.int32), context_lens.to(torch.int32),
+                float(scale))
+    return R.paged_attention_decode(q, key_cache, value_cache, block_tables,
+                                    context_lens, scale)
+
+
+# -------------------------------------------------------------- sampling
+
+def sample_tokens(logits: torch.Tensor, temperature: float, top_p: float,
+                  top_k: int, seeds: Optional[torch.Tensor] = None,
+                  generator: Optional[torch.Generator] = None) -> torch.Tensor:
+    if logits.is_cuda:
+        ext = _require_ext("sample_tokens")
+        if ext is not None and temperature > 0.0:
+            if seeds is None:
+                seeds = torch.randint(0, 2**31 - 1, (logits.shape[0],),
+                                      device=logits.device, dtype=torch.int64,
+                                      generator=generator)
+            return ext.sample_tokens(logits.contiguous(), float(temperature),
+                                     float(top_p), int(top_k), seeds)
+    return R.sample_tokens(logits, temperature, top_p, top_k, generator=generator)
+
+
+# ------------------------------------------------------------ fused loss
+
+class _LogprobLossFn(torch.autograd.Function):
+    """Fused per-token log-softmax + gather + masked PG/GRPO loss.
+
+    forward returns the scalar loss; backward emits dlogits directly
+    without materializing a (B, T, V) log-prob tensor (SURVEY.md §2.4-B
+    north star). loss = -mean_b( (sum_t logp*mask / sum_t mask) * R_b ).
+    """
+
+    @staticmethod
+    def forward(ctx, logits, targets, mask, rewards, loss_scale):
+        ext = get_extension() if logits.is_cuda else None
+        m = mask.to(torch.float32)
+        denom = m.sum(-1).clamp_min(1.0)
+        coef = -rewards.to(torch.float32) / denom / logits.shape[0] * loss_scale
+        if ext is not None:
+            loss, lse = ext.logprob_loss_fwd(logits, targets, m, coef)
+            ctx.save_for_backward(logits, targets, m, coef, lse)
+            ctx.used_ext = True
+            return loss
+        logp = logits.float().log_softmax(-1)
+        tok = logp.gather(-1, targets.unsqueeze(-1)).squeeze(-1)
+        loss = ((tok * m).sum(-1) * coef).sum()
+        ctx.save_for_backward(logits, targets, m, coef,
+                              torch.logsumexp(logits.float(), -1))
+        ctx.used_ext = False
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, targets, m, coef, lse = ctx.saved_tensors
+        if ctx.used_ext:
+            ext = get_extension()
+            dlogits = ext.logprob_loss_bwd(logits, targets, m, coef, lse,
+                                           dloss.float())
+        else:
+            probs = (logits.float() - lse.unsqueeze(-1)).exp()
+            w = (m * coef.unsqueeze(-1)) * dloss.float()
+            dlogits = probs * (-w).unsqueeze(-1)
+            dlogits.scatter_add_(-1, targets.unsqueeze(-1), w.unsqueeze(-1))
+            dlogits = dlogits.to(logits.dtype)
+        return dlogits, None, None, None, None
+
+
+def logprob_loss(logits: torch.Tensor, targets: torch.Tensor, mask: torch.Tensor,
+                 rewards: torch.Tensor, loss_scale: float = 1.0) -> torch.Tensor:
+    """PG/GRPO loss on answer-region logits. logits (B, T, V); targets/mask
+    (B, T); rewards (B,). ``loss_scale`` folds the reference's
+    loss/num_batches scaling (reference distributed_actor.py:382)."""
+    return _LogprobLossFn.apply(logits, targets, mask, rewards, loss_scale)

@@ -1,0 +1,292 @@
+"""SPMD RL Trainer: the reference Trainer's orchestration, re-built on
+collectives instead of Ray RPC.
+
+Every rank runs the same program (one process per GPU). Rank 0 drives the
+control flow by broadcasting (command, payload) tuples over the gloo
+control plane; all ranks execute each command's collective sequence in
+lockstep. This replaces the reference's Ray remote-call surface
+(SURVEY.md §1 L4->L3: generate / train / compute_gradients /
+apply_merged_gradients / save_adapter / save_checkpoint).
+
+Round structure matches reference distributed_trainer.py:232-382:
+generate fan-out across ALL workers (learners generate too —
+reference README:19), rank-0 reward + advantage + top-k math, learner
+update (gradient all-reduce over the learner subgroup instead of CPU
+gather to learner 0), weight sync (RCCL broadcast instead of adapter disk
+round-trip, plus the adapter save kept for checkpoint parity), metrics
+with the reference's exact key set, periodic eval
+(pass@1 / BoN — distributed_trainer.py:384-416) and checkpoint cadence.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Dict, List, Optional
+
+import numpy as np
+
+from ..config import SamplingParams
+from ..models import lora as lora_io
+from ..parallel.fabric import Fabric
+from .advantage import even_chunk_sizes, merge_candidates, process_candidates
+from .sched import split_dict_lists, worker_chunk_sizes
+
+
+class Trainer:
+    def __init__(self, fabric: Fabric, config: Dict, engine, learner,
+                 tokenizer, train_dataset=None, test_dataset=None,
+                 reward_function=None, logger=None):
+        self.fabric = fabric
+        self.config = config
+        self.engine = engine
+        self.learner = learner  # None on pure actor ranks
+        self.tokenizer = tokenizer
+        self.train_dataset = train_dataset  # rank 0 only needs these
+        self.test_dataset = test_dataset
+        self.reward_function = reward_function
+        self.logger = logger
+
+        c = config
+        self.batch_size = c["batch_size"]
+        self.num_candidates = c["num_candidates"]
+        self.learner_chunk_size = c["learner_chunk_size"]
+        self.topk = c["topk"]
+        self.learner_type = c["learner"]
+        self.episodes = c["episodes"]
+        self.save_every = c["save_every"]
+        self.eval_every = c["eval_every"]
+        self.lora_save_path = c["lora_save_path"]
+        self.run_directory = f"run_{c.get('run_name') or 'default'}"
+        self.sampling_params = SamplingParams(
+            max_tokens=c["max_new_tokens"], temperature=c["temperature"],
+            n=c["num_candidates"], top_p=0.95)
+        # eval params mirror reference distributed_trainer.py:53-58
+        self.eval_sampling_params = SamplingParams(
+            max_tokens=c["max_new_tokens"], temperature=0.6, top_p=0.95, n=8)
+        self.eos_token_id = getattr(tokenizer, "eos_token_id", None)
+
+    # ----------------------------------------------------------- plumbing
+
+    def _cmd(self, name: str, payload=None):
+        """Rank 0: broadcast a command then execute it locally."""
+        assert self.fabric.rank == 0
+        self.fabric.broadcast_obj((name, payload), src=0)
+        return self._handle(name, payload)
+
+    def follower_loop(self):
+        """Ranks != 0: execute the command stream until stop."""
+        while True:
+            name, payload = self.fabric.broadcast_obj(src=0)
+            if name == "stop":
+                break
+            self._handle(name, payload)
+
+    def _handle(self, name: str, payload):
+        if name == "generate":
+            return self._generate_handler(payload)
+        if name == "update":
+            return self._update_handler(payload)
+        if name == "sync_weights":
+            params = self._lora_params()
+            self.fabric.broadcast_lora(params)
+            return None
+        if name == "save_adapter":
+            if self.fabric.rank == 0:
+                self._save_adapter(payload)
+            return None
+        if name == "save_checkpoint":
+            if self.fabric.rank == 0:
+                self._save_adapter(payload)
+            return None
+        raise ValueError(f"unknown command {name!r}")
+
+    def _lora_params(self):
+        model = self.engine.model
+        return [p for p in model.parameters() if p.requires_grad]
+
+    def _save_adapter(self, path: str):
+        c = self.config
+        lora_io.save_adapter(self.engine.model, path, c["model"],
+                             c["max_lora_rank"], c["lora_alpha"],
+                             c["lora_dropout"])
+
+    # ----------------------------------------------------------- generate
+
+    def _generate_handler(self, payload):
+        """All ranks: generate this rank's chunk, gather to rank 0.
+
+        payload: (batch_dict, sp_dict). Returns (candidates, duration) on
+        rank 0, None elsewhere. Candidate dicts use the reference wire
+        format: answers / token_lengths /
+        solution / problem each replicated n-per-prompt.
+        """
+        batch, sp_dict = payload
+        sp = SamplingParams(**sp_dict)
+        t0 = time.time()
+        bsz = len(batch["problem"])
+        sizes = worker_chunk_sizes(bsz, self.fabric.num_actors,
+                                   self.fabric.num_learners,
+                                   self.learner_chunk_size)
+        chunks = split_dict_lists(batch, sizes)
+        my_task = chunks[self.fabric.rank]
+        result = None
+        if len(my_task["problem"]) > 0:
+            result = self._generate_task(my_task, sp)
+        gathered = self.fabric.gather_obj(result, dst=0)
+        if self.fabric.rank != 0:
+            return None
+        candidates = [g for g in gathered if g is not None]
+        return candidates, time.time() - t0
+
+    def _generate_task(self, task: Dict, sp: SamplingParams) -> Dict:
+        """Run the engine on one task dict — the native vllm_generate
+        (reference distributed_actor.py:147-172)."""
+        prompt_ids = [self.tokenizer.encode(p) for p in task["problem"]]
+        outs = self.engine.generate(prompt_ids, sp,
+                                    eos_token_id=self.eos_token_id)
+        task = dict(task)
+        task["answers"] = [[self.tokenizer.decode(ids, skip_special_tokens=True)
+                            for ids in per_prompt] for per_prompt in outs]
+        task["token_lengths"] = [[len(ids) for ids in per_prompt]
+                                 for per_prompt in outs]
+        task["solution"] = [[s] * sp.n for s in task["solution"]]
+        task["problem"] = [[p] * sp.n for p in task["problem"]]
+        return task
+
+    def _compute_rewards(self, candidates: List[Dict]) -> float:
+        """Rank 0: per-group reward arrays (reference
+        distributed_trainer.py:205-219)."""
+        t0 = time.time()
+        for cand in candidates:
+            rewards = []
+            for answers, solutions in zip(cand["answers"], cand["solution"]):
+                rewards.append(self.reward_function(answers, solutions))
+            cand["rewards"] = rewards
+        return time.time() - t0
+
+    # ------------------------------------------------------------- update
+
+    def _update_handler(self, payload):
+        """Learners: accumulate grads on their chunk, all-reduce, step.
+        Returns this rank's loss (gathered by rank 0)."""
+        chunks = payload  # list of (problems, answers, rewards) per learner
+        loss = None
+        if self.fabric.is_learner:
+            problems, answers, rewards = chunks[self.fabric.learner_index]
+            if len(problems) > 0:
+                loss = self.learner.accumulate_gradients(problems, answers,
+                                                         rewards)
+            else:
+                loss = 0.0
+            self.fabric.allreduce_mean_grads(self.learner.params)
+            self.learner.step()
+            loss = self.fabric.allreduce_mean_scalar(loss)
+        gathered = self.fabric.gather_obj(loss, dst=0)
+        if self.fabric.rank != 0:
+            return None
+        losses = [x for x in gathered if x is not None]
+        return float(np.mean(losses)) if losses else 0.0
+
+    # -------------------------------------------------------------- train
+
+    def train(self):
+        """Entry point on every rank."""
+        if self.fabric.rank != 0:
+            return self.follower_loop()
+        try:
+            self._train_rank0()
+        finally:
+            self.fabric.broadcast_obj(("stop", None), src=0)
+
+    def _train_rank0(self):
+        total_batch_steps = 0
+        total_samples = 0
+        sp_dict = self.sampling_params.__dict__
+
+        if self.eval_every > 0:
+            self.evaluate(total_batch_steps)
+
+        for episode in range(self.episodes):
+            dataset = self.train_dataset.shuffle()
+            for batch in dataset.iter(batch_size=self.batch_size):
+                total_batch_steps += 1
+                total_samples += len(batch["problem"])
+
+                candidates, gen_dur = self._cmd("generate", (batch, sp_dict))
+                reward_dur = self._compute_rewards(candidates)
+                candidates, stats = process_candidates(candidates,
+                                                       self.learner_type,
+                                                       self.topk)
+
+                # split flattened training samples across learners evenly
+                # (reference distributed_trainer.py:311-322)
+                t0 = time.time()
+                problems, answers, rewards = merge_candidates(candidates)
+                sizes = even_chunk_sizes(len(problems), self.fabric.num_learners)
+                chunks, start = [], 0
+                for sz in sizes:
+                    chunks.append((problems[start:start + sz],
+                                   answers[start:start + sz],
+                                   rewards[start:start + sz]))
+                    start += sz
+                loss = self._cmd("update", chunks)
+                self._cmd("sync_weights")
+                update_dur = time.time() - t0
+
+                self._cmd("save_adapter", self.lora_save_path)
+
+                if self.logger is not None:
+                    self.logger.log({
+                        "loss": loss,
+                        "mean_format_reward": stats["mean_format_reward"],
+                        "mean_accuracy_reward": stats["mean_accuracy_reward"],
+                        "min_accuracy_reward": stats["min_accuracy_reward"],
+                        "max_accuracy_reward": stats["max_accuracy_reward"],
+                        "mean_token_length": stats["mean_token_length"],
+                        "episode": episode,
+                        "total_batch_steps": total_batch_steps,
+                        "total_samples_processed": total_samples,
+                        "timing/update_duration": update_dur,
+                        "timing/reward_duration": reward_dur,
+                        "timing/generation_duration": gen_dur,
+                    }, step=total_batch_steps)
+
+                if self.eval_every > 0 and total_batch_steps % self.eval_every == 0:
+                    self.evaluate(total_batch_steps)
+                if self.save_every > 0 and total_batch_steps % self.save_every == 0:
+                    self._cmd("save_checkpoint",
+                              os.path.join(self.run_directory,
+                                           f"model_{total_batch_steps}"))
+            self._cmd("save_checkpoint",
+                      os.path.join(self.run_directory,
+                                   f"model_{total_batch_steps}"))
+
+    # --------------------------------------------------------------- eval
+
+    def evaluate(self, total_steps: int):
+        """Rank 0 drives; same fan-out machinery with eval sampling params
+        (reference distributed_trainer.py:384-416)."""
+        t0 = time.time()
+        sp_dict = self.eval_sampling_params.__dict__
+        total_passed = 0.0
+        total_max = 0.0
+        total_problems = 0
+        token_lengths = []
+        for batch in self.test_dataset.iter(batch_size=self.batch_size):
+            candidates, _dur = self._cmd("generate", (batch, sp_dict))
+            self._compute_rewards(candidates)
+            for cand in candidates:
+                for r, toks in zip(cand["rewards"], cand["token_lengths"]):
+                    token_lengths.append(float(np.mean(toks)))
+                    total_passed += float(np.mean(r[:, 1]))
+                    total_max += float(np.max(r[:, 1]))
+                    total_problems += 1
+        n = self.eval_sampling_params.n
+        if self.logger is not None and total_problems > 0:
+            self.logger.log({
+                f"eval/pass@1(mean{n})": total_passed / total_problems,
+                f"eval/BoN({n})": total_max / total_problems,
+                "eval/mean_token_length": float(np.mean(token_lengths)),
+                "timing/eval_duration": time.time() - t0,
+            }, step=total_steps)

@@ -1,0 +1,91 @@
+import pytest
+import torch
+
+from distrl_llm_amd.config import EngineConfig, SamplingParams
+from distrl_llm_amd.engine import Engine
+from distrl_llm_amd.models import CausalLM, get_spec
+
+
+@pytest.fixture(scope="module")
+def setup():
+    spec = get_spec("tiny-qwen2")
+    model = CausalLM(spec, lora_r=4, lora_alpha=8, dtype=torch.float32)
+    model.random_init(seed=7)
+    cfg = EngineConfig(max_seq_length=128, kv_block_size=8, num_kv_blocks=256,
+                       max_num_seqs=64)
+    return model, Engine(model, cfg, device=torch.device("cpu"), seed=0)
+
+
+def _naive_greedy(model, prompt_ids, steps):
+    """Full-recompute greedy decoding as ground truth."""
+    ids = list(prompt_ids)
+    out = []
+    for _ in range(steps):
+        logits = model(torch.tensor([ids]))[0, -1]
+        t = int(logits.argmax())
+        out.append(t)
+        ids.append(t)
+    return out
+
+
+def test_greedy_matches_naive(setup):
+    model, engine = setup
+    prompts = [[1, 5, 9, 2, 7], [3, 3, 8]]
+    sp = SamplingParams(max_tokens=6, temperature=0.0, n=1)
+    results = engine.generate(prompts, sp, eos_token_id=None)
+    for p, res in zip(prompts, results):
+        assert len(res) == 1
+        expected = _naive_greedy(model, p, 6)
+        assert res[0] == expected, (res[0], expected)
+
+
+def test_fanout_counts_and_pool_freed(setup):
+    model, engine = setup
+    prompts = [list(range(1, 12)), [2, 4, 6]]
+    sp = SamplingParams(max_tokens=5, temperature=1.0, n=4, top_p=0.9)
+    before_free = engine.pool.allocator.num_free
+    results = engine.generate(prompts, sp, eos_token_id=None)
+    assert [len(r) for r in results] == [4, 4]
+    for r in results:
+        for ids in r:
+            assert len(ids) == 5
+    # all KV blocks returned to the pool
+    assert engine.pool.allocator.num_free == before_free
+
+
+def test_eos_termination(setup):
+    model, engine = setup
+    # find the greedy-argmax token after a short prompt and declare it EOS:
+    prompt = [1, 2, 3]
+    first = _naive_greedy(model, prompt, 1)[0]
+    sp = SamplingParams(max_tokens=8, temperature=0.0, n=1)
+    res = engine.generate([prompt], sp, eos_token_id=first)
+    assert res[0][0] == [first]  # stopped immediately, EOS included
+
+
+def test_shared_prefill_isolation(setup):
+    """n candidates of one prompt must match n separate single runs in
+    greedy mode (shared prompt blocks don't corrupt each other)."""
+    model, engine = setup
+    prompt = list(range(1, 18))  # spans 2+ blocks of 8 with a partial tail
+    sp = SamplingParams(max_tokens=4, temperature=0.0, n=3)
+    res = engine.generate([prompt], sp, eos_token_id=None)
+    expected = _naive_greedy(model, prompt, 4)
+    for ids in res[0]:
+        assert ids == expected
+
+
+def test_continuous_batching_admission(setup):
+    """More prompts than the pool can hold at once must still all finish."""
+    model, engine = setup
+    spec = model.spec
+    cfg = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=40,
+                       max_num_seqs=64)
+    small = Engine(model, cfg, device=torch.device("cpu"), seed=0)
+    prompts = [[i + 1, i + 2, i + 3, i + 4] for i in range(10)]
+    sp = SamplingParams(max_tokens=6, temperature=0.0, n=2)
+    results = small.generate(prompts, sp, eos_token_id=None)
+    assert [len(r) for r in results] == [2] * 10
+    expected0 = _naive_greedy(model, prompts[0], 6)
+    assert results[0][0] == expected0
+    assert small.pool.allocator.num_free == 40

@@ -1,0 +1,122 @@
+import os
+
+import pytest
+import torch
+
+from distrl_llm_amd.models import CausalLM, get_spec
+from distrl_llm_amd.models.lora import (load_adapter, lora_state_dict,
+                                        save_adapter)
+from distrl_llm_amd.ops import reference as R
+
+
+@pytest.fixture(scope="module")
+def tiny_model():
+    spec = get_spec("tiny-qwen2")
+    m = CausalLM(spec, lora_r=4, lora_alpha=8, dtype=torch.float32)
+    m.random_init(seed=0)
+    return m
+
+
+def test_forward_shapes(tiny_model):
+    B, T = 2, 12
+    ids = torch.randint(0, tiny_model.spec.vocab_size, (B, T))
+    logits = tiny_model(ids)
+    assert logits.shape == (B, T, tiny_model.spec.vocab_size)
+    assert torch.isfinite(logits).all()
+
+
+def test_fresh_lora_is_noop(tiny_model):
+    """B zeros ==> adapter contributes nothing (reference first-round
+    behavior)."""
+    ids = torch.randint(0, tiny_model.spec.vocab_size, (1, 8))
+    logits = tiny_model(ids)
+    # disable lora by zeroing A as well; with B=0 output must be unchanged
+    with torch.no_grad():
+        saved = [p.clone() for p in tiny_model.parameters() if p.requires_grad]
+        for mod in tiny_model.modules():
+            if hasattr(mod, "lora_A") and mod.lora_A is not None:
+                mod.lora_A.zero_()
+    logits2 = tiny_model(ids)
+    torch.testing.assert_close(logits, logits2)
+    with torch.no_grad():
+        for p, s in zip([p for p in tiny_model.parameters() if p.requires_grad], saved):
+            p.copy_(s)
+
+
+def test_left_pad_invariance(tiny_model):
+    """Left-padding must not change the logits of real tokens."""
+    ids = torch.randint(0, tiny_model.spec.vocab_size, (1, 6))
+    logits = tiny_model(ids)
+    padded = torch.cat([torch.zeros(1, 3, dtype=torch.long), ids], dim=1)
+    mask = torch.cat([torch.zeros(1, 3, dtype=torch.long),
+                      torch.ones(1, 6, dtype=torch.long)], dim=1)
+    logits_p = tiny_model(padded, mask)
+    torch.testing.assert_close(logits, logits_p[:, 3:], rtol=1e-4, atol=1e-4)
+
+
+def test_grads_only_on_lora(tiny_model):
+    ids = torch.randint(0, tiny_model.spec.vocab_size, (2, 8))
+    loss = tiny_model(ids).float().pow(2).mean()
+    loss.backward()
+    for name, p in tiny_model.named_parameters():
+        if "lora_" in name:
+            assert p.requires_grad
+        else:
+            assert not p.requires_grad and p.grad is None
+    tiny_model.zero_grad(set_to_none=True)
+
+
+def test_adapter_roundtrip(tmp_path, tiny_model):
+    with torch.no_grad():
+        for p in tiny_model.parameters():
+            if p.requires_grad:
+                p.add_(torch.randn_like(p) * 0.01)
+    path = str(tmp_path / "adapter")
+    save_adapter(tiny_model, path, "tiny-qwen2", r=4, alpha=8)
+    assert os.path.exists(os.path.join(path, "adapter_config.json"))
+    assert os.path.exists(os.path.join(path, "adapter_model.safetensors"))
+    before = {k: v.clone() for k, v in lora_state_dict(tiny_model).items()}
+
+    spec = get_spec("tiny-qwen2")
+    m2 = CausalLM(spec, lora_r=4, lora_alpha=8, dtype=torch.float32)
+    m2.random_init(seed=0)
+    n = load_adapter(m2, path)
+    assert n == 7 * spec.num_layers
+    after = lora_state_dict(m2)
+    for k in before:
+        torch.testing.assert_close(before[k], after[k])
+
+
+def test_peft_key_format(tiny_model):
+    keys = list(lora_state_dict(tiny_model))
+    assert all(k.startswith("base_model.model.model.layers.") for k in keys)
+    assert any(k.endswith(".self_attn.q_proj.lora_A.weight") for k in keys)
+    assert any(k.endswith(".mlp.down_proj.lora_B.weight") for k in keys)
+
+
+def test_nf4_roundtrip():
+    w = torch.randn(128, 64)
+    packed, absmax = R.quantize_nf4(w, block_size=64)
+    assert packed.dtype == torch.uint8 and packed.numel() == w.numel() // 2
+    deq = R.dequantize_nf4(packed, absmax, w.shape, 64)
+    # nf4 quantization error is bounded by half the largest code gap
+    # (0.304/2 = 0.152) x absmax
+    err = (w - deq).abs()
+    bound = absmax.repeat_interleave(64).view(w.shape) * 0.1521 + 1e-6
+    assert (err <= bound).all()
+    # exact codebook values round-trip exactly
+    w2 = torch.tensor([R.NF4_CODE.tolist() * 4]) * 3.0
+    p2, a2 = R.quantize_nf4(w2, 64)
+    torch.testing.assert_close(R.dequantize_nf4(p2, a2, w2.shape, 64), w2)
+
+
+def test_quantize_model_attaches_sidecar():
+    spec = get_spec("tiny-qwen2")
+    m = CausalLM(spec, lora_r=0, dtype=torch.float32).random_init(1)
+    before = m.model.layers[0].self_attn.q_proj.weight.clone()
+    m.quantize_nf4_()
+    mod = m.model.layers[0].self_attn.q_proj
+    assert mod.weight_nf4 is not None and mod.weight_absmax is not None
+    # weight replaced by its quantized image: close but not identical
+    assert not torch.equal(before, mod.weight)
+    assert (before - mod.weight).abs().max() < 0.1
