@@ -132,10 +132,13 @@ class CausalLM(nn.Module):
         (BASELINE.json: random-init weights). The reference seeds LoRA with
         random_state=3407 (reference helper.py:43)."""
         dev = next(self.parameters()).device
-        g = torch.Generator(device="cpu").manual_seed(seed)
+        # generate on-device when possible (7B on CPU would take minutes);
+        # deterministic for a given (seed, architecture, device type)
+        g = torch.Generator(device=dev).manual_seed(seed)
 
         def fill(t, std=0.02):
-            t.copy_(torch.randn(t.shape, generator=g, dtype=torch.float32).mul_(std).to(t.dtype))
+            t.copy_(torch.randn(t.shape, generator=g, dtype=torch.float32,
+                                device=dev).mul_(std).to(t.dtype))
 
         fill(self.model.embed_tokens.weight)
         for layer in self.model.layers:
@@ -146,7 +149,8 @@ class CausalLM(nn.Module):
                 if mod.bias is not None:
                     mod.bias.zero_()
                 if mod.r > 0:
-                    a = torch.empty(mod.lora_A.shape, dtype=torch.float32)
+                    a = torch.empty(mod.lora_A.shape, dtype=torch.float32,
+                                    device=dev)
                     nn.init.kaiming_uniform_(a, a=math.sqrt(5), generator=g)
                     mod.lora_A.copy_(a.to(mod.lora_A.dtype))
                     mod.lora_B.zero_()

@@ -91,6 +91,14 @@ class Trainer:
             params = self._lora_params()
             self.fabric.broadcast_lora(params)
             return None
+        if name == "barrier":
+            import torch
+            if self.fabric.device.type == "cuda":
+                torch.cuda.synchronize()
+            self.fabric.barrier()
+            if self.fabric.device.type == "cuda":
+                torch.cuda.synchronize()
+            return None
         if name == "save_adapter":
             if self.fabric.rank == 0:
                 self._save_adapter(payload)
@@ -199,10 +207,44 @@ class Trainer:
         finally:
             self.fabric.broadcast_obj(("stop", None), src=0)
 
+    def rl_round(self, batch: Dict, sp_dict: Optional[Dict] = None) -> Dict:
+        """Rank 0: one full RL round — generate fan-out, rewards,
+        advantage/top-k, learner update, weight sync. Returns the round's
+        metrics dict (also the unit of work bench.py times)."""
+        sp_dict = sp_dict or self.sampling_params.__dict__
+        candidates, gen_dur = self._cmd("generate", (batch, sp_dict))
+        reward_dur = self._compute_rewards(candidates)
+        candidates, stats = process_candidates(candidates, self.learner_type,
+                                               self.topk)
+
+        # split flattened training samples across learners evenly
+        # (reference distributed_trainer.py:311-322)
+        t0 = time.time()
+        problems, answers, rewards = merge_candidates(candidates)
+        sizes = even_chunk_sizes(len(problems), self.fabric.num_learners)
+        chunks, start = [], 0
+        for sz in sizes:
+            chunks.append((problems[start:start + sz],
+                           answers[start:start + sz],
+                           rewards[start:start + sz]))
+            start += sz
+        loss = self._cmd("update", chunks)
+        self._cmd("sync_weights")
+        update_dur = time.time() - t0
+
+        stats = dict(stats)
+        stats.update({
+            "loss": loss,
+            "num_samples": len(problems),
+            "timing/update_duration": update_dur,
+            "timing/reward_duration": reward_dur,
+            "timing/generation_duration": gen_dur,
+        })
+        return stats
+
     def _train_rank0(self):
         total_batch_steps = 0
         total_samples = 0
-        sp_dict = self.sampling_params.__dict__
 
         if self.eval_every > 0:
             self.evaluate(total_batch_steps)
@@ -213,32 +255,12 @@ class Trainer:
                 total_batch_steps += 1
                 total_samples += len(batch["problem"])
 
-                candidates, gen_dur = self._cmd("generate", (batch, sp_dict))
-                reward_dur = self._compute_rewards(candidates)
-                candidates, stats = process_candidates(candidates,
-                                                       self.learner_type,
-                                                       self.topk)
-
-                # split flattened training samples across learners evenly
-                # (reference distributed_trainer.py:311-322)
-                t0 = time.time()
-                problems, answers, rewards = merge_candidates(candidates)
-                sizes = even_chunk_sizes(len(problems), self.fabric.num_learners)
-                chunks, start = [], 0
-                for sz in sizes:
-                    chunks.append((problems[start:start + sz],
-                                   answers[start:start + sz],
-                                   rewards[start:start + sz]))
-                    start += sz
-                loss = self._cmd("update", chunks)
-                self._cmd("sync_weights")
-                update_dur = time.time() - t0
-
+                stats = self.rl_round(batch)
                 self._cmd("save_adapter", self.lora_save_path)
 
                 if self.logger is not None:
                     self.logger.log({
-                        "loss": loss,
+                        "loss": stats["loss"],
                         "mean_format_reward": stats["mean_format_reward"],
                         "mean_accuracy_reward": stats["mean_accuracy_reward"],
                         "min_accuracy_reward": stats["min_accuracy_reward"],
@@ -247,9 +269,9 @@ class Trainer:
                         "episode": episode,
                         "total_batch_steps": total_batch_steps,
                         "total_samples_processed": total_samples,
-                        "timing/update_duration": update_dur,
-                        "timing/reward_duration": reward_dur,
-                        "timing/generation_duration": gen_dur,
+                        "timing/update_duration": stats["timing/update_duration"],
+                        "timing/reward_duration": stats["timing/reward_duration"],
+                        "timing/generation_duration": stats["timing/generation_duration"],
                     }, step=total_batch_steps)
 
                 if self.eval_every > 0 and total_batch_steps % self.eval_every == 0:
