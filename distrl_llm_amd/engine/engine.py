@@ -42,6 +42,10 @@ class Engine:
         self.generator = torch.Generator(device=self.device)
         self.generator.manual_seed(seed)
 
+        d = self.spec.head_dim
+        self._inv_freq = 1.0 / (self.spec.rope_theta ** (
+            torch.arange(0, d, 2, device=self.device, dtype=torch.float32) / d))
+
         num_blocks = cfg.num_kv_blocks
         if num_blocks <= 0:
             num_blocks = self._derive_num_blocks()
@@ -106,16 +110,14 @@ class Engine:
             q.context_len = L
         slot_mapping = torch.tensor(slot_mapping, dtype=torch.long, device=device)
 
-        cos, sin = R.rope_cos_sin(positions, s.head_dim, s.rope_theta,
-                                  device=device)
-
         x = self.model.model.embed_tokens(input_ids)
         for li, layer in enumerate(self.model.model.layers):
             h = layer.input_layernorm(x)
             q = self._proj(layer.self_attn.q_proj, h).view(total, s.num_heads, s.head_dim)
             k = self._proj(layer.self_attn.k_proj, h).view(total, s.num_kv_heads, s.head_dim)
             v = self._proj(layer.self_attn.v_proj, h).view(total, s.num_kv_heads, s.head_dim)
-            q, k = R.apply_rope(q, k, cos, sin)
+            q, k = OF.apply_rope_inplace(q, k, positions, self._inv_freq,
+                                         s.rope_theta)
             OF.kv_cache_scatter(k, v, self.pool.key[li], self.pool.value[li],
                                 slot_mapping)
             o = self._prefill_attention(q, k, v, lens)
@@ -183,16 +185,14 @@ class Engine:
                 q.block_table, dtype=torch.int32)
         context_lens = torch.tensor(ctx_lens, dtype=torch.int32, device=device)
 
-        cos, sin = R.rope_cos_sin(positions_t, s.head_dim, s.rope_theta,
-                                  device=device)
-
         x = self.model.model.embed_tokens(input_ids)
         for li, layer in enumerate(self.model.model.layers):
             h = layer.input_layernorm(x)
             q = self._proj(layer.self_attn.q_proj, h).view(N, s.num_heads, s.head_dim)
             k = self._proj(layer.self_attn.k_proj, h).view(N, s.num_kv_heads, s.head_dim)
             v = self._proj(layer.self_attn.v_proj, h).view(N, s.num_kv_heads, s.head_dim)
-            q, k = R.apply_rope(q, k, cos, sin)
+            q, k = OF.apply_rope_inplace(q, k, positions_t, self._inv_freq,
+                                         s.rope_theta)
             OF.kv_cache_scatter(k, v, self.pool.key[li], self.pool.value[li],
                                 slot_mapping)
             o = OF.paged_attention_decode(q, self.pool.key[li], self.pool.value[li],

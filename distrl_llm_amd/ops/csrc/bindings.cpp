@@ -1,0 +1,47 @@
+// Python bindings for the gfx950 kernel library.
+#include <torch/extension.h>
+#include <vector>
+
+torch::Tensor rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+torch::Tensor rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                          double eps);
+torch::Tensor silu_mul_fwd(torch::Tensor gate, torch::Tensor up);
+std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor gate,
+                                        torch::Tensor up);
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+                  torch::Tensor inv_freq);
+void kv_cache_scatter(torch::Tensor k, torch::Tensor v,
+                      torch::Tensor key_cache, torch::Tensor value_cache,
+                      torch::Tensor slot_mapping);
+torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kcache,
+                                     torch::Tensor vcache,
+                                     torch::Tensor block_tables,
+                                     torch::Tensor ctx_lens, double scale);
+torch::Tensor sample_tokens(torch::Tensor logits, double temperature,
+                            double top_p, int64_t top_k, torch::Tensor seeds);
+std::vector<torch::Tensor> logprob_lse_fwd(torch::Tensor logits,
+                                           torch::Tensor targets);
+torch::Tensor logprob_loss_bwd(torch::Tensor logits, torch::Tensor targets,
+                               torch::Tensor w, torch::Tensor lse);
+void adam8bit_step(torch::Tensor p, torch::Tensor g, torch::Tensor m_q,
+                   torch::Tensor v_q, torch::Tensor m_absmax,
+                   torch::Tensor v_absmax, double lr, double b1, double b2,
+                   double eps, double wd, int64_t step);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (gfx950)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward dx (gfx950)");
+  m.def("silu_mul_fwd", &silu_mul_fwd, "fused SiLU*mul forward");
+  m.def("silu_mul_bwd", &silu_mul_bwd, "fused SiLU*mul backward");
+  m.def("rope_inplace", &rope_inplace, "fused in-place RoPE (q,k)");
+  m.def("kv_cache_scatter", &kv_cache_scatter, "paged KV cache scatter");
+  m.def("paged_attention_decode", &paged_attention_decode,
+        "paged GQA decode attention");
+  m.def("sample_tokens", &sample_tokens,
+        "fused temperature/top-k/top-p categorical sampling");
+  m.def("logprob_lse_fwd", &logprob_lse_fwd,
+        "per-token log p(target) + LSE, one streaming pass");
+  m.def("logprob_loss_bwd", &logprob_loss_bwd,
+        "dlogits for the masked PG/GRPO loss");
+  m.def("adam8bit_step", &adam8bit_step, "fused blockwise 8-bit Adam step");
+}

@@ -1,0 +1,64 @@
+#include "hip/hip_runtime.h"
+// Fused in-place RoPE for the generation path (rotate_half / neox
+// convention, matching HF Qwen2/Llama). Replaces vLLM's pos_encoding
+// kernels. cos/sin computed once per token into LDS and
+// reused across all q + kv heads.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+__global__ void rope_inplace_kernel(__hip_bfloat16* __restrict__ q,
+                                    __hip_bfloat16* __restrict__ k,
+                                    const int* __restrict__ positions,
+                                    const float* __restrict__ inv_freq,
+                                    int H, int KV, int D) {
+  extern __shared__ float smem[];  // [D/2 cos][D/2 sin]
+  const int half = D / 2;
+  float* cs = smem;
+  float* sn = smem + half;
+  const int t = blockIdx.x;
+  const float pos = (float)positions[t];
+  for (int i = threadIdx.x; i < half; i += blockDim.x) {
+    float a = pos * inv_freq[i];
+    sn[i] = __sinf(a);
+    cs[i] = __cosf(a);
+  }
+  __syncthreads();
+
+  const int total = (H + KV) * half;
+  for (int idx = threadIdx.x; idx < total; idx += blockDim.x) {
+    const int head = idx / half;
+    const int i = idx % half;
+    __hip_bfloat16* ptr = (head < H)
+        ? q + ((int64_t)t * H + head) * D
+        : k + ((int64_t)t * KV + (head - H)) * D;
+    float x1 = bf2f(ptr[i]);
+    float x2 = bf2f(ptr[i + half]);
+    float c = cs[i], s = sn[i];
+    ptr[i] = f2bf(x1 * c - x2 * s);
+    ptr[i + half] = f2bf(x2 * c + x1 * s);
+  }
+}
+
+}  // namespace
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+                  torch::Tensor inv_freq) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(positions.scalar_type() == at::kInt);
+  const int T = q.size(0), H = q.size(1), D = q.size(2);
+  const int KV = k.size(1);
+  TORCH_CHECK(k.size(0) == T && k.size(2) == D);
+  if (T == 0) return;
+  const int smem = D * sizeof(float);
+  hipLaunchKernelGGL(rope_inplace_kernel, dim3(T), dim3(256), smem,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     reinterpret_cast<__hip_bfloat16*>(q.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
+                     positions.data_ptr<int>(), inv_freq.data_ptr<float>(),
+                     H, KV, D);
+  HIP_CHECK_LAST();
+}

@@ -117,15 +117,16 @@ def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
 # ------------------------------------------------------------------ rope
 
 def apply_rope_inplace(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
-                       head_dim: int, theta: float) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Inference-path RoPE: q (T, H, D), k (T, KV, D), positions (T,).
-    On GPU uses the fused kernel (in-place); on CPU returns rotated copies."""
+                       inv_freq: torch.Tensor, theta: float) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Inference-path RoPE: q (T, H, D), k (T, KV, D), positions (T,),
+    inv_freq (D/2,) fp32 precomputed. On GPU uses the fused in-place
+    kernel; on CPU returns rotated copies."""
     if q.is_cuda:
         ext = _require_ext("rope")
         if ext is not None:
-            ext.rope_inplace(q, k, positions.to(torch.int32), float(theta))
+            ext.rope_inplace(q, k, positions.to(torch.int32), inv_freq)
             return q, k
-    cos, sin = R.rope_cos_sin(positions, head_dim, theta, device=q.device)
+    cos, sin = R.rope_cos_sin(positions, q.shape[-1], theta, device=q.device)
     return R.apply_rope(q, k, cos, sin)
 
 
@@ -185,33 +186,31 @@ class _LogprobLossFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, logits, targets, mask, rewards, loss_scale):
-        ext = get_extension() if logits.is_cuda else None
+        ext = _require_ext("logprob_loss") if logits.is_cuda else None
         m = mask.to(torch.float32)
         denom = m.sum(-1).clamp_min(1.0)
         coef = -rewards.to(torch.float32) / denom / logits.shape[0] * loss_scale
+        logits = logits.contiguous()
         if ext is not None:
-            loss, lse = ext.logprob_loss_fwd(logits, targets, m, coef)
-            ctx.save_for_backward(logits, targets, m, coef, lse)
-            ctx.used_ext = True
-            return loss
-        logp = logits.float().log_softmax(-1)
-        tok = logp.gather(-1, targets.unsqueeze(-1)).squeeze(-1)
+            tok, lse = ext.logprob_lse_fwd(logits, targets.contiguous())
+        else:
+            logp = logits.float().log_softmax(-1)
+            tok = logp.gather(-1, targets.unsqueeze(-1)).squeeze(-1)
+            lse = torch.logsumexp(logits.float(), -1)
         loss = ((tok * m).sum(-1) * coef).sum()
-        ctx.save_for_backward(logits, targets, m, coef,
-                              torch.logsumexp(logits.float(), -1))
-        ctx.used_ext = False
+        ctx.save_for_backward(logits, targets, m, coef, lse)
+        ctx.used_ext = ext is not None
         return loss
 
     @staticmethod
     def backward(ctx, dloss):
         logits, targets, m, coef, lse = ctx.saved_tensors
+        w = (m * coef.unsqueeze(-1)) * dloss.float()
         if ctx.used_ext:
             ext = get_extension()
-            dlogits = ext.logprob_loss_bwd(logits, targets, m, coef, lse,
-                                           dloss.float())
+            dlogits = ext.logprob_loss_bwd(logits, targets, w, lse)
         else:
             probs = (logits.float() - lse.unsqueeze(-1)).exp()
-            w = (m * coef.unsqueeze(-1)) * dloss.float()
             dlogits = probs * (-w).unsqueeze(-1)
             dlogits.scatter_add_(-1, targets.unsqueeze(-1), w.unsqueeze(-1))
             dlogits = dlogits.to(logits.dtype)

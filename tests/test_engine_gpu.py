@@ -1,0 +1,89 @@
+"""GPU engine + learner integration: the full native path (HIP kernels for
+norm/rope/cache/attention/sampling/loss) on a small bf16 model."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def setup():
+    from distrl_llm_amd.config import EngineConfig
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    spec = get_spec("small-qwen2")
+    model = CausalLM(spec, lora_r=8, lora_alpha=16, dtype=torch.bfloat16,
+                     device=torch.device("cuda:0"))
+    model.random_init(seed=7)
+    cfg = EngineConfig(max_seq_length=512, kv_block_size=16,
+                       num_kv_blocks=2048, max_num_seqs=256)
+    return model, Engine(model, cfg, device=torch.device("cuda:0"), seed=0)
+
+
+def _naive_greedy(model, prompt_ids, steps):
+    ids = list(prompt_ids)
+    out = []
+    for _ in range(steps):
+        with torch.no_grad():
+            logits = model(torch.tensor([ids], device="cuda:0"))[0, -1]
+        t = int(logits.argmax())
+        out.append(t)
+        ids.append(t)
+    return out
+
+
+def test_gpu_greedy_matches_naive(setup):
+    model, engine = setup
+    from distrl_llm_amd.config import SamplingParams
+    prompts = [[1, 5, 9, 2, 7, 11, 200, 3000], [3, 3, 8]]
+    sp = SamplingParams(max_tokens=8, temperature=0.0, n=1)
+    results = engine.generate(prompts, sp, eos_token_id=None)
+    for p, res in zip(prompts, results):
+        expected = _naive_greedy(model, p, 8)
+        # bf16 paged path vs bf16 recompute path: tiny numeric divergence
+        # can flip argmax; require near-total agreement
+        agree = sum(a == b for a, b in zip(res[0], expected))
+        assert agree >= 7, (res[0], expected)
+
+
+def test_gpu_fanout_and_long_context(setup):
+    model, engine = setup
+    from distrl_llm_amd.config import SamplingParams
+    prompts = [list(torch.randint(0, 4000, (200,)).tolist()) for _ in range(3)]
+    sp = SamplingParams(max_tokens=32, temperature=1.0, n=4, top_p=0.95)
+    before = engine.pool.allocator.num_free
+    results = engine.generate(prompts, sp, eos_token_id=None)
+    assert [len(r) for r in results] == [4, 4, 4]
+    for r in results:
+        for ids in r:
+            assert len(ids) == 32
+            assert all(0 <= t < model.spec.vocab_size for t in ids)
+    assert engine.pool.allocator.num_free == before
+
+
+def test_gpu_learner_round(setup):
+    model, engine = setup
+    from distrl_llm_amd.train.learner import Learner
+    from distrl_llm_amd.utils.tokenizer import ByteTokenizer
+    tok = ByteTokenizer(vocab_size=model.spec.vocab_size)
+    learner = Learner(model, tok, lr=1e-3, max_prompt_tokens=32,
+                      max_new_tokens=32, train_batch_size=2)
+    problems = ["what is 1+1?", "compute 2*3", "evaluate 5-2", "sum 1..4"]
+    answers = ["<answer>2</answer>", "<answer>6</answer>",
+               "<answer>3</answer>", "<answer>10</answer>"]
+    lora_before = model.model.layers[0].self_attn.q_proj.lora_B.clone()
+    loss = learner.accumulate_gradients(problems, answers, [1.0, -0.5, 0.3, 0.1])
+    assert torch.isfinite(torch.tensor(loss))
+    learner.step()
+    assert not torch.equal(lora_before,
+                           model.model.layers[0].self_attn.q_proj.lora_B)
+
+
+def test_native_extension_is_loaded():
+    """The HIP path must be the one that runs on GPU (no silent eager
+    fallback)."""
+    from distrl_llm_amd.ops.build import get_extension
+    ext = get_extension()
+    assert ext is not None, "gfx950 extension must load on the GPU box"
+    assert "_build" in ext.__file__  # in-tree .so
