@@ -69,7 +69,7 @@ def main():
     model_name = args.model
     max_new, max_prompt = args.max_new_tokens, args.max_prompt_tokens
     if args.tiny:
-        model_name = "tiny-qwen2"
+        model_name = "tiny-qwen2" if not use_cuda else "small-qwen2"
         max_new, max_prompt = 16, 64
 
     num_learners = world_size if args.actors < 0 else world_size - args.actors
