@@ -148,6 +148,10 @@ def main():
         stats = trainer.rl_round(batches[args.warmup + i], sp_dict)
         total_samples += stats["num_samples"]
         per_step.append(time.time() - ts)
+        print(f"[bench] step {i}: gen={stats['timing/generation_duration']:.2f}s "
+              f"reward={stats['timing/reward_duration']:.2f}s "
+              f"update={stats['timing/update_duration']:.2f}s "
+              f"total={per_step[-1]:.2f}s", file=sys.stderr, flush=True)
     trainer._cmd("barrier", None)
     elapsed = time.time() - t0
 
