@@ -110,7 +110,23 @@ def main():
     from distrl_llm_amd.parallel.worker import build_worker
     from distrl_llm_amd.rl.data import (ListDataset, process_dataset,
                                         r1_preprompt, synthetic_math_dataset)
-    from distrl_llm_amd.rl.rewards import reward_function
+    from distrl_llm_amd.rl.rewards import reward_function as real_reward_function
+
+    def reward_function(completions, solutions):
+        """Real reward stack + a deterministic per-candidate jitter.
+
+        Random-init weights produce uniformly zero rewards, which would make
+        every GRPO advantage zero and trip the degenerate-group skip — the
+        learner would do no work and the bench would be a lie. The jitter
+        (hash of the completion text, <=0.1) restores the reward spread real
+        training has, so the timed region contains the full learner
+        workload: forward, fused loss, backward, all-reduce, Adam step."""
+        import zlib
+        r = real_reward_function(completions, solutions)
+        jit = np.array([(zlib.crc32(c[-256:].encode()) % 1000) / 1000.0 * 0.1
+                        for c in completions])
+        r[:, 0] += jit
+        return r
 
     n_prompts = batch_size * (args.steps + args.warmup)
     rows = process_dataset(None, synthetic_math_dataset(n_prompts, seed=17),

@@ -1,0 +1,164 @@
+"""Device-resident decode session with hipGraph-captured steps.
+
+The eager decode loop pays per-step Python state assembly + ~700 kernel
+launches per step on a 28-layer model. A DecodeSession instead keeps ALL
+per-sequence state (tokens, positions, context lengths, block tables,
+finished mask, output buffer, sampler seed/step counters) in device
+tensors, pre-allocates each sequence's worst-case KV blocks up front (the
+admission control already reserved them), and expresses one decode step as
+a fixed tensor program over those buffers — embed -> 28 layers (HIP
+rmsnorm / rope / kv-scatter / paged-attention + hipBLASLt projections) ->
+LM head -> fused sampler -> in-graph state advance. That program is
+captured once into a hipGraph and replayed per token (SURVEY.md §2.4-A:
+"hipGraph-captured decode step" north star); the host loop is just
+graph.replay() with one finished-mask sync per chunk of steps.
+
+Sampling randomness inside the graph: the fused sampler hashes
+(per-seq seed, device step counter), and the counter increments in-graph,
+so every replay draws fresh randomness with zero host work.
+
+Frozen (finished) sequences stop advancing their position and re-forward
+their last token idempotently; their recorded garbage steps are dropped at
+extraction (per-seq output length = final_position - prompt_len + 1).
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from ..config import SamplingParams
+from .kvcache import KVCachePool, Sequence
+
+
+class DecodeSession:
+    def __init__(self, engine, seqs: List[Sequence], sp: SamplingParams,
+                 eos_token_id: Optional[int], use_graph: bool = True):
+        self.engine = engine
+        self.seqs = seqs
+        self.sp = sp
+        self.eos = eos_token_id
+        dev = engine.device
+        bs = engine.pool.block_size
+        N = len(seqs)
+        max_total = engine.cfg.max_seq_length
+
+        prompt_lens = [len(q.prompt_ids) for q in seqs]
+        limits = [min(pl + sp.max_tokens, max_total) - 1 for pl in prompt_lens]
+
+        # pre-allocate every block each sequence can ever need
+        for q, lim in zip(seqs, limits):
+            need = KVCachePool.blocks_for(lim + 1, bs)
+            while len(q.block_table) < need:
+                q.block_table.append(engine.pool.allocator.alloc())
+        max_nb = max(len(q.block_table) for q in seqs)
+
+        bt = torch.zeros(N, max_nb, dtype=torch.int32)
+        for i, q in enumerate(seqs):
+            bt[i, :len(q.block_table)] = torch.tensor(q.block_table,
+                                                      dtype=torch.int32)
+        self.block_tables = bt.to(dev, non_blocking=True)
+        self.tokens = torch.tensor([q.output_ids[-1] for q in seqs],
+                                   dtype=torch.long, device=dev)
+        self.positions = torch.tensor(prompt_lens, dtype=torch.long, device=dev)
+        self.prompt_lens = list(prompt_lens)
+        self.ctx_lens = (self.positions + 1).to(torch.int32)
+        self.limit_pos = torch.tensor(limits, dtype=torch.long, device=dev)
+        self.finished = torch.zeros(N, dtype=torch.bool, device=dev)
+        self.step_idx = torch.zeros(1, dtype=torch.long, device=dev)
+        self.seeds = torch.randint(0, 2**31 - 1, (N,), dtype=torch.int64,
+                                   device=dev, generator=engine.generator)
+        self.max_steps = max(lim - pl for lim, pl in zip(limits, prompt_lens))
+        self.out_buf = torch.zeros(max(self.max_steps, 1), N,
+                                   dtype=torch.long, device=dev)
+        self.graph = None
+        self.use_graph = use_graph and dev.type == "cuda"
+
+    # ------------------------------------------------------------- step
+
+    def _step(self):
+        e = self.engine
+        bs = e.pool.block_size
+        pos = self.positions
+        blk = torch.div(pos, bs, rounding_mode="floor")
+        slot = (self.block_tables.gather(1, blk.unsqueeze(1)).squeeze(1).long()
+                * bs + pos % bs)
+        logits = e._decode_forward(self.tokens, pos, slot, self.block_tables,
+                                   self.ctx_lens)
+        if self.sp.temperature > 0.0:
+            from ..ops.build import get_extension
+            ext = get_extension()
+            sampled = ext.sample_tokens(logits, float(self.sp.temperature),
+                                        float(self.sp.top_p),
+                                        int(self.sp.top_k), self.seeds,
+                                        self.step_idx)
+        else:
+            sampled = logits.argmax(-1)
+        self.out_buf.index_copy_(0, self.step_idx, sampled.unsqueeze(0))
+
+        was_finished = self.finished
+        new_fin = was_finished | (self.positions + 1 >= self.limit_pos)
+        if self.eos is not None:
+            new_fin = new_fin | (~was_finished & sampled.eq(self.eos))
+        adv = (~was_finished).long()
+        self.tokens.copy_(torch.where(was_finished, self.tokens, sampled))
+        self.positions.add_(adv)
+        self.ctx_lens.add_(adv.to(torch.int32))
+        self.finished.copy_(new_fin)
+        self.step_idx.add_(1)
+
+    # ---------------------------------------------------------- capture
+
+    def _capture(self):
+        state = [self.tokens, self.positions, self.ctx_lens, self.finished,
+                 self.step_idx]
+        saved = [t.clone() for t in state]
+        torch.cuda.synchronize()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._step()
+            self._step()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        for t, sv in zip(state, saved):
+            t.copy_(sv)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._step()
+        self.graph = g
+
+    # -------------------------------------------------------------- run
+
+    def run(self, chunk: int = 64) -> List[List[int]]:
+        if self.max_steps > 0:
+            if self.use_graph:
+                try:
+                    self._capture()
+                except Exception as err:  # capture unsupported -> eager
+                    import sys
+                    print(f"[engine] hipGraph capture failed ({err}); "
+                          f"eager decode", file=sys.stderr)
+                    self.graph = None
+            steps = 0
+            while steps < self.max_steps:
+                n = min(chunk, self.max_steps - steps)
+                if self.graph is not None:
+                    for _ in range(n):
+                        self.graph.replay()
+                else:
+                    for _ in range(n):
+                        self._step()
+                steps += n
+                if bool(self.finished.all()):
+                    break
+
+        # ---- extraction ----
+        final_pos = self.positions.cpu().tolist()
+        out = self.out_buf.cpu()
+        results = []
+        for i, q in enumerate(self.seqs):
+            n_new = final_pos[i] - self.prompt_lens[i]
+            results.append(q.output_ids[:1] + out[:n_new, i].tolist())
+        return results

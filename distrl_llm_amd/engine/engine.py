@@ -184,14 +184,23 @@ class Engine:
             block_tables[i, :len(q.block_table)] = torch.tensor(
                 q.block_table, dtype=torch.int32)
         context_lens = torch.tensor(ctx_lens, dtype=torch.int32, device=device)
+        return self._decode_forward(input_ids, positions_t, slot_mapping,
+                                    block_tables, context_lens)
 
+    def _decode_forward(self, input_ids: torch.Tensor, positions: torch.Tensor,
+                        slot_mapping: torch.Tensor, block_tables: torch.Tensor,
+                        context_lens: torch.Tensor) -> torch.Tensor:
+        """Single-token batched decode forward over device-side state —
+        the hipGraph-capturable step body (all inputs are device tensors)."""
+        s = self.spec
+        N = input_ids.shape[0]
         x = self.model.model.embed_tokens(input_ids)
         for li, layer in enumerate(self.model.model.layers):
             h = layer.input_layernorm(x)
             q = self._proj(layer.self_attn.q_proj, h).view(N, s.num_heads, s.head_dim)
             k = self._proj(layer.self_attn.k_proj, h).view(N, s.num_kv_heads, s.head_dim)
             v = self._proj(layer.self_attn.v_proj, h).view(N, s.num_kv_heads, s.head_dim)
-            q, k = OF.apply_rope_inplace(q, k, positions_t, self._inv_freq,
+            q, k = OF.apply_rope_inplace(q, k, positions, self._inv_freq,
                                          s.rope_theta)
             OF.kv_cache_scatter(k, v, self.pool.key[li], self.pool.value[li],
                                 slot_mapping)
@@ -329,6 +338,26 @@ class Engine:
                         running.append(c)
 
         try_admit()
+
+        if self.device.type == "cuda" and not self.cfg.enforce_eager:
+            # wave-based decode sessions (device state + hipGraph replay)
+            from .decode_session import DecodeSession
+            while running or waiting:
+                if not running:
+                    try_admit()
+                    if not running:
+                        if waiting:
+                            raise MemoryError(
+                                "KV pool too small to admit any waiting prompt")
+                        break
+                session = DecodeSession(self, running, sp, eos_token_id)
+                outs = session.run()
+                for q, ids in zip(running, outs):
+                    results[q.parent_prompt].append(ids)
+                    self._finish(q)
+                running = []
+            return results
+
         while running or waiting:
             if not running:
                 try_admit()

@@ -48,6 +48,8 @@ class Attention(nn.Module):
         self.scale = spec.head_dim ** -0.5
 
     def forward(self, x, cos, sin, attn_bias):
+        """attn_bias: additive (B,1,T,T) mask, or None for pure causal
+        (right-padded layouts — lets SDPA pick its flash backend)."""
         B, T, _ = x.shape
         s = self.spec
         q = self.q_proj(x).view(B, T, s.num_heads, s.head_dim)
@@ -58,8 +60,12 @@ class Attention(nn.Module):
         k = k.repeat_interleave(group, dim=2)
         v = v.repeat_interleave(group, dim=2)
         q, k, v = (t.transpose(1, 2) for t in (q, k, v))  # (B, H, T, D)
-        o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias,
-                                           scale=self.scale)
+        if attn_bias is None:
+            o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                               scale=self.scale)
+        else:
+            o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias,
+                                               scale=self.scale)
         o = o.transpose(1, 2).reshape(B, T, s.q_size)
         return self.o_proj(o)
 
@@ -201,12 +207,20 @@ class CausalLM(nn.Module):
         position_ids = (attention_mask.long().cumsum(-1) - 1).clamp_min(0)
         cos, sin = self._rope_tables(position_ids)
 
-        # causal + padding mask -> additive bias (B, 1, T, T)
-        causal = torch.ones(T, T, dtype=torch.bool, device=input_ids.device).tril()
-        pad = attention_mask.bool().view(B, 1, 1, T)
-        allow = causal.view(1, 1, T, T) & pad
-        bias = torch.zeros(B, 1, T, T, dtype=self.dtype_, device=input_ids.device)
-        bias.masked_fill_(~allow, torch.finfo(self.dtype_).min)
+        # Right-pad-only layouts need no mask at all: trailing pads never
+        # influence real tokens under causal attention -> SDPA can use its
+        # flash backend. Left-padded rows need the full additive bias.
+        right_pad_only = bool(
+            (attention_mask[:, 0] == 1).all()
+            and (attention_mask.long().diff(dim=-1) <= 0).all())
+        if right_pad_only:
+            bias = None
+        else:
+            causal = torch.ones(T, T, dtype=torch.bool, device=input_ids.device).tril()
+            pad = attention_mask.bool().view(B, 1, 1, T)
+            allow = causal.view(1, 1, T, T) & pad
+            bias = torch.zeros(B, 1, T, T, dtype=self.dtype_, device=input_ids.device)
+            bias.masked_fill_(~allow, torch.finfo(self.dtype_).min)
 
         x = self.model.embed_tokens(input_ids)
         for layer in self.model.layers:

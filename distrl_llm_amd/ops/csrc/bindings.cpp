@@ -18,7 +18,8 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kcache,
                                      torch::Tensor block_tables,
                                      torch::Tensor ctx_lens, double scale);
 torch::Tensor sample_tokens(torch::Tensor logits, double temperature,
-                            double top_p, int64_t top_k, torch::Tensor seeds);
+                            double top_p, int64_t top_k, torch::Tensor seeds,
+                            torch::Tensor step);
 std::vector<torch::Tensor> logprob_lse_fwd(torch::Tensor logits,
                                            torch::Tensor targets);
 torch::Tensor logprob_loss_bwd(torch::Tensor logits, torch::Tensor targets,

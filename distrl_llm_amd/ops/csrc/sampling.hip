@@ -33,6 +33,7 @@ template <typename T>
 __global__ __launch_bounds__(256)
 void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
                    float top_p, int top_k, const int64_t* __restrict__ seeds,
+                   const int64_t* __restrict__ step,
                    int64_t* __restrict__ out) {
   __shared__ float red[16];
   __shared__ float bin_p[NBINS];
@@ -43,7 +44,9 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
 
   const int row = blockIdx.x;
   const T* lr = logits + (int64_t)row * V;
-  const uint64_t seed = (uint64_t)seeds[row];
+  // step counter read from device memory so a hipGraph replay of this
+  // kernel draws fresh randomness every step (the graph increments *step)
+  const uint64_t seed = (uint64_t)seeds[row] ^ splitmix64(0x5D21u + (uint64_t)*step);
 
   // ---- pass A: max ----
   float m = -INFINITY;
@@ -129,10 +132,12 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
 }  // namespace
 
 torch::Tensor sample_tokens(torch::Tensor logits, double temperature,
-                            double top_p, int64_t top_k, torch::Tensor seeds) {
+                            double top_p, int64_t top_k, torch::Tensor seeds,
+                            torch::Tensor step) {
   TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() && logits.dim() == 2);
   TORCH_CHECK(temperature > 0.0, "temperature 0 is greedy: use argmax");
   TORCH_CHECK(seeds.scalar_type() == at::kLong);
+  TORCH_CHECK(step.scalar_type() == at::kLong && step.is_cuda());
   const int B = logits.size(0), V = logits.size(1);
   auto out = torch::empty({B}, logits.options().dtype(at::kLong));
   if (B == 0) return out;
@@ -143,13 +148,14 @@ torch::Tensor sample_tokens(torch::Tensor logits, double temperature,
     hipLaunchKernelGGL(sample_kernel<__hip_bfloat16>, grid, block, 0, stream,
                        reinterpret_cast<const __hip_bfloat16*>(logits.data_ptr()),
                        V, inv_t, (float)top_p, (int)top_k,
-                       seeds.data_ptr<int64_t>(), out.data_ptr<int64_t>());
+                       seeds.data_ptr<int64_t>(), step.data_ptr<int64_t>(),
+                       out.data_ptr<int64_t>());
   } else {
     TORCH_CHECK(logits.scalar_type() == at::kFloat);
     hipLaunchKernelGGL(sample_kernel<float>, grid, block, 0, stream,
                        logits.data_ptr<float>(), V, inv_t, (float)top_p,
                        (int)top_k, seeds.data_ptr<int64_t>(),
-                       out.data_ptr<int64_t>());
+                       step.data_ptr<int64_t>(), out.data_ptr<int64_t>());
   }
   HIP_CHECK_LAST();
   return out;

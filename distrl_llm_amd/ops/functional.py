@@ -161,6 +161,7 @@ def paged_attention_decode(q, key_cache, value_cache, block_tables,
 
 def sample_tokens(logits: torch.Tensor, temperature: float, top_p: float,
                   top_k: int, seeds: Optional[torch.Tensor] = None,
+                  step: Optional[torch.Tensor] = None,
                   generator: Optional[torch.Generator] = None) -> torch.Tensor:
     if logits.is_cuda:
         ext = _require_ext("sample_tokens")
@@ -169,8 +170,10 @@ def sample_tokens(logits: torch.Tensor, temperature: float, top_p: float,
                 seeds = torch.randint(0, 2**31 - 1, (logits.shape[0],),
                                       device=logits.device, dtype=torch.int64,
                                       generator=generator)
+            if step is None:
+                step = torch.zeros(1, dtype=torch.int64, device=logits.device)
             return ext.sample_tokens(logits.contiguous(), float(temperature),
-                                     float(top_p), int(top_k), seeds)
+                                     float(top_p), int(top_k), seeds, step)
     return R.sample_tokens(logits, temperature, top_p, top_k, generator=generator)
 
 

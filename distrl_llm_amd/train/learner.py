@@ -51,39 +51,52 @@ class Learner:
 
     # ------------------------------------------------------- tokenization
 
-    def _encode_batch(self, problems: Sequence[str], answers: Sequence[str]
-                      ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
-        """Fixed-shape layout: prompt left-pad to max_prompt_tokens (keep
-        the first tokens on truncation, like HF truncation=True), answer
-        right-pad to max_new_tokens."""
+    def _encode_batch(self, problems: Sequence[str], answers: Sequence[str]):
+        """Packed right-padded layout: [prompt(<=max_prompt) | answer
+        (<=max_new) | pad]. Keeps the reference's truncation semantics
+        (prompt keeps its first max_prompt_tokens, answers their first
+        max_new_tokens — distributed_actor.py:217-229) but packs instead of
+        left-padding, so causal attention needs no mask (flash-capable) and
+        trailing pad compute shrinks to the batch max.
+        """
         P, A = self.max_prompt_tokens, self.max_new_tokens
         B = len(problems)
-        input_ids = torch.full((B, P + A), self.pad_token_id, dtype=torch.long)
-        attn = torch.zeros(B, P + A, dtype=torch.long)
-        targets = torch.full((B, A), self.pad_token_id, dtype=torch.long)
-        ans_mask = torch.zeros(B, A, dtype=torch.long)
-        for i, (p, a) in enumerate(zip(problems, answers)):
-            pi = self.tokenizer.encode(p)[:P]
-            ai = self.tokenizer.encode(a)[:A]
-            input_ids[i, P - len(pi):P] = torch.tensor(pi)
-            attn[i, P - len(pi):P] = 1
+        enc = [(self.tokenizer.encode(p)[:P], self.tokenizer.encode(a)[:A])
+               for p, a in zip(problems, answers)]
+        T = max(len(pi) + len(ai) for pi, ai in enc)
+        T = min((T + 7) // 8 * 8, P + A)  # pad batch length to multiple of 8
+        A_max = max(max(len(ai) for _, ai in enc), 1)
+        input_ids = torch.full((B, T), self.pad_token_id, dtype=torch.long)
+        attn = torch.zeros(B, T, dtype=torch.long)
+        targets = torch.full((B, A_max), self.pad_token_id, dtype=torch.long)
+        ans_mask = torch.zeros(B, A_max, dtype=torch.long)
+        gather_idx = torch.zeros(B, A_max, dtype=torch.long)
+        for i, (pi, ai) in enumerate(enc):
+            L = len(pi) + len(ai)
+            input_ids[i, :L] = torch.tensor(pi + ai)
+            attn[i, :L] = 1
             if ai:
-                input_ids[i, P:P + len(ai)] = torch.tensor(ai)
-                attn[i, P:P + len(ai)] = 1
                 targets[i, :len(ai)] = torch.tensor(ai)
                 ans_mask[i, :len(ai)] = 1
+            # hidden positions predicting the answer tokens:
+            # len(pi)-1 .. len(pi)+len(ai)-2 (clamped into range)
+            start = max(len(pi) - 1, 0)
+            gather_idx[i] = torch.arange(start, start + A_max).clamp_max(T - 1)
         dev = self.device
-        return (input_ids.to(dev), attn.to(dev), targets.to(dev), ans_mask.to(dev))
+        return (input_ids.to(dev), attn.to(dev), targets.to(dev),
+                ans_mask.to(dev), gather_idx.to(dev))
 
     # -------------------------------------------------------------- loss
 
     def _micro_loss(self, problems, answers, rewards: torch.Tensor,
                     loss_scale: float) -> torch.Tensor:
-        P = self.max_prompt_tokens
-        input_ids, attn, targets, ans_mask = self._encode_batch(problems, answers)
+        input_ids, attn, targets, ans_mask, gidx = self._encode_batch(
+            problems, answers)
         hidden = self.model.forward_hidden(input_ids, attn)
-        # answer-region hidden: positions P-1 .. P+A-2 predict tokens P..P+A-1
-        ans_logits = self.model.logits(hidden[:, P - 1:-1, :])
+        H = hidden.shape[-1]
+        ans_hidden = hidden.gather(
+            1, gidx.unsqueeze(-1).expand(-1, -1, H))
+        ans_logits = self.model.logits(ans_hidden)
         return OF.logprob_loss(ans_logits, targets, ans_mask, rewards, loss_scale)
 
     def accumulate_gradients(self, problems: List[str], answers: List[str],
