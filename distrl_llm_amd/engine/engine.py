@@ -53,6 +53,11 @@ class Engine:
                                 cfg.kv_block_size, self.spec.num_kv_heads,
                                 self.spec.head_dim, self.dtype, self.device)
         self._graph_runner = None
+        if self.device.type == "cuda":
+            from .weights import FusedWeights
+            self.fused = FusedWeights(model)
+        else:
+            self.fused = None
 
     # ------------------------------------------------------------ sizing
 
@@ -191,7 +196,16 @@ class Engine:
                         slot_mapping: torch.Tensor, block_tables: torch.Tensor,
                         context_lens: torch.Tensor) -> torch.Tensor:
         """Single-token batched decode forward over device-side state —
-        the hipGraph-capturable step body (all inputs are device tensors)."""
+        the hipGraph-capturable step body (all inputs are device tensors).
+
+        GPU path: merged-LoRA fused weights (one qkv GEMM, one gate|up
+        GEMM per layer), fused rope+KV-scatter, strided-q paged attention,
+        fused residual-add+RMSNorm and packed SiLU*mul — 9 kernels/layer.
+        """
+        if self.device.type == "cuda":
+            return self._decode_forward_fused(input_ids, positions,
+                                              slot_mapping, block_tables,
+                                              context_lens)
         s = self.spec
         N = input_ids.shape[0]
         x = self.model.model.embed_tokens(input_ids)
@@ -213,6 +227,39 @@ class Engine:
             x = x + self._proj(layer.mlp.down_proj, OF.silu_mul(g, u))
         x = self.model.model.norm(x)
         return self.model.logits(x)
+
+    def _decode_forward_fused(self, input_ids, positions, slot_mapping,
+                              block_tables, context_lens):
+        from ..ops.build import get_extension
+        ext = get_extension()
+        if ext is None:
+            raise RuntimeError("fused decode requires the gfx950 extension")
+        s = self.spec
+        m = self.model
+        qs, kvs = s.q_size, s.kv_size
+        eps = s.rms_norm_eps
+        lws = self.fused.layers
+        pos32 = positions.to(torch.int32)
+        res = m.model.embed_tokens(input_ids)
+        h = ext.rmsnorm_fwd(res, lws[0].in_norm, eps)
+        for li, lw in enumerate(lws):
+            qkv = F.linear(h, lw.qkv_w, lw.qkv_b)
+            ext.rope_scatter_qkv(qkv, pos32, slot_mapping, self._inv_freq,
+                                 self.pool.key[li], self.pool.value[li],
+                                 s.num_heads, s.num_kv_heads, s.head_dim)
+            attn = ext.paged_attention_decode_strided(
+                qkv, s.num_heads, s.head_dim, qs + 2 * kvs,
+                self.pool.key[li], self.pool.value[li], block_tables,
+                context_lens, self.scale)
+            o = F.linear(attn.view(-1, qs), lw.o_w)
+            h, res = ext.add_rmsnorm_fwd(res, o, lw.post_norm, eps)
+            gu = F.linear(h, lw.gateup_w)
+            act = ext.silu_mul_packed(gu)
+            d = F.linear(act, lw.down_w)
+            next_w = (lws[li + 1].in_norm if li + 1 < len(lws)
+                      else m.model.norm.weight)
+            h, res = ext.add_rmsnorm_fwd(res, d, next_w, eps)
+        return m.logits(h)
 
     # ----------------------------------------------------------- forking
 
@@ -265,6 +312,10 @@ class Engine:
         """
         was_training = self.model.training
         self.model.eval()
+        if self.fused is not None:
+            # fold the current LoRA into the merged decode weights
+            # (once per generation round, never inside the decode loop)
+            self.fused.refresh()
         try:
             return self._generate_inner(prompts, sp, eos_token_id,
                                         prefill_token_budget)

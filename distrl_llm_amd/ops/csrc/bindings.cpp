@@ -5,11 +5,22 @@
 torch::Tensor rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
 torch::Tensor rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
                           double eps);
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor h,
+                                           torch::Tensor w, double eps);
 torch::Tensor silu_mul_fwd(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor gate,
                                         torch::Tensor up);
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   torch::Tensor inv_freq);
+void rope_scatter_qkv(torch::Tensor qkv, torch::Tensor positions,
+                      torch::Tensor slots, torch::Tensor inv_freq,
+                      torch::Tensor key_cache, torch::Tensor value_cache,
+                      int64_t H, int64_t KV, int64_t D);
+torch::Tensor silu_mul_packed(torch::Tensor gu);
+torch::Tensor paged_attention_decode_strided(
+    torch::Tensor q, int64_t n_heads, int64_t head_dim, int64_t q_row_stride,
+    torch::Tensor kcache, torch::Tensor vcache, torch::Tensor block_tables,
+    torch::Tensor ctx_lens, double scale);
 void kv_cache_scatter(torch::Tensor k, torch::Tensor v,
                       torch::Tensor key_cache, torch::Tensor value_cache,
                       torch::Tensor slot_mapping);
@@ -32,9 +43,16 @@ void adam8bit_step(torch::Tensor p, torch::Tensor g, torch::Tensor m_q,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (gfx950)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward dx (gfx950)");
+  m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd,
+        "fused residual add + RMSNorm forward");
   m.def("silu_mul_fwd", &silu_mul_fwd, "fused SiLU*mul forward");
   m.def("silu_mul_bwd", &silu_mul_bwd, "fused SiLU*mul backward");
   m.def("rope_inplace", &rope_inplace, "fused in-place RoPE (q,k)");
+  m.def("rope_scatter_qkv", &rope_scatter_qkv,
+        "fused packed-qkv RoPE + paged KV scatter");
+  m.def("silu_mul_packed", &silu_mul_packed, "SiLU*mul on packed [gate|up]");
+  m.def("paged_attention_decode_strided", &paged_attention_decode_strided,
+        "paged GQA decode attention (packed-q rows)");
   m.def("kv_cache_scatter", &kv_cache_scatter, "paged KV cache scatter");
   m.def("paged_attention_decode", &paged_attention_decode,
         "paged GQA decode attention");

@@ -29,14 +29,39 @@ void logprob_lse_kernel(const T* __restrict__ logits,
   const int64_t tgt = targets[row];
 
   float m = -INFINITY, s = 0.f;
-  for (int i = threadIdx.x; i < V; i += blockDim.x) {
-    float x = ld(lr, i);
-    if (i == tgt) tgt_logit_s = x;   // exactly one thread hits it
-    if (x > m) {
-      s = s * __expf(m - x) + 1.f;
-      m = x;
-    } else {
-      s += __expf(x - m);
+  if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+    const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
+    const int nvec = V / 8;
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      bf16x8 v = l8[i];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float x = bf2f(v.v[j]);
+        if (i * 8 + j == (int)tgt) tgt_logit_s = x;
+        if (x > m) {
+          s = s * __expf(m - x) + 1.f;
+          m = x;
+        } else {
+          s += __expf(x - m);
+        }
+      }
+    }
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += blockDim.x) {
+      float x = ld(lr, i);
+      if (i == (int)tgt) tgt_logit_s = x;
+      if (x > m) { s = s * __expf(m - x) + 1.f; m = x; }
+      else s += __expf(x - m);
+    }
+  } else {
+    for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      float x = ld(lr, i);
+      if (i == tgt) tgt_logit_s = x;   // exactly one thread hits it
+      if (x > m) {
+        s = s * __expf(m - x) + 1.f;
+        m = x;
+      } else {
+        s += __expf(x - m);
+      }
     }
   }
   // combine per-thread (m, s): M = max m; S = sum s_i * exp(m_i - M)
@@ -62,8 +87,34 @@ void loss_bwd_kernel(const T* __restrict__ logits,
   const float wv = w[row];
   const float l = lse[row];
   const int64_t tgt = targets[row];
+  if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+    const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
+    bf16x8* d8 = reinterpret_cast<bf16x8*>(dr);
+    const int nvec = V / 8;
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      bf16x8 o;
+      if (wv == 0.f) {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) o.v[j] = f2bf(0.f);
+      } else {
+        bf16x8 v = l8[i];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float p = __expf(bf2f(v.v[j]) - l);
+          o.v[j] = f2bf(wv * ((i * 8 + j == (int)tgt ? 1.f : 0.f) - p));
+        }
+      }
+      d8[i] = o;
+    }
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += blockDim.x) {
+      float g = 0.f;
+      if (wv != 0.f)
+        g = wv * ((i == (int)tgt ? 1.f : 0.f) - __expf(ld(lr, i) - l));
+      dr[i] = f2bf(g);
+    }
+    return;
+  }
   if (wv == 0.f) {
-    // masked row: zero gradient (vectorized store)
     for (int i = threadIdx.x; i < V; i += blockDim.x)
       dr[i] = (T)(0.f);
     return;
@@ -71,10 +122,7 @@ void loss_bwd_kernel(const T* __restrict__ logits,
   for (int i = threadIdx.x; i < V; i += blockDim.x) {
     float p = __expf(ld(lr, i) - l);
     float g = wv * ((i == tgt ? 1.f : 0.f) - p);
-    if constexpr (std::is_same<T, __hip_bfloat16>::value)
-      dr[i] = f2bf(g);
-    else
-      dr[i] = g;
+    dr[i] = g;
   }
 }
 

@@ -6,12 +6,14 @@
 //   grid = (num_seqs, n_kv_heads); one workgroup owns one (sequence,
 //   kv-head) pair and its whole group of query heads (GQA 7:1 on
 //   Qwen2.5-7B). 4 waves / 256 threads.
-//   Phase 1: 16-lane thread groups stream K rows (one bf16x8 = 16 B per
-//   lane), dot against the group's Q (staged in LDS as fp32), shfl-reduce
-//   within the 16-lane group, scores -> LDS.
+//   Phase 1 (scores): each LANE owns one context token — it streams that
+//   token's K row with bf16x8 loads and keeps one fp32 partial per query
+//   head in registers (GROUP is a template param so the per-head array
+//   stays in VGPRs); Q is staged in LDS and read as wave-uniform
+//   broadcasts. No cross-lane reduction at all.
 //   Phase 2: per-head softmax (wave-strided max/sum over the LDS scores).
-//   Phase 3: V streamed through LDS in 32-token tiles, each thread
-//   accumulates one (head, 8-dim) slice in registers, epilogue divides by
+//   Phase 3 (PV): V streamed through LDS in 32-token tiles; each thread
+//   accumulates one (head, 8-dim) slice in registers; epilogue divides by
 //   the softmax denominator and stores bf16x8.
 //
 // K and V are each read exactly once from HBM per step; the kernel is
@@ -22,10 +24,9 @@
 
 namespace {
 
-constexpr int TPG = 16;          // threads per K-token group (D<=128: 8 B..16 B/lane)
 constexpr int VTILE = 32;        // V tokens staged per LDS tile
 
-template <int D>
+template <int D, int GROUP>
 __global__ __launch_bounds__(256)
 void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
                          const __hip_bfloat16* __restrict__ kcache, // (nb,bs,KV,D)
@@ -34,57 +35,53 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
                          const int* __restrict__ ctx_lens,          // (N,)
                          __hip_bfloat16* __restrict__ out,          // (N,H,D)
                          int H, int KV, int max_nb, int block_size,
-                         int Lpad, float scale) {
+                         int Lpad, float scale, int64_t q_row_stride) {
   const int seq = blockIdx.x;
   const int kv = blockIdx.y;
-  const int group = H / KV;
   const int L = ctx_lens[seq];
   const int tid = threadIdx.x;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* q_lds = reinterpret_cast<float*>(smem_raw);          // group*D
-  float* denom = q_lds + group * D;                           // group
-  float* scores = denom + ((group + 3) & ~3);                 // group*Lpad
+  float* q_lds = reinterpret_cast<float*>(smem_raw);          // GROUP*D
+  float* denom = q_lds + GROUP * D;                           // padded to 4
+  float* scores = denom + ((GROUP + 3) & ~3);                 // GROUP*Lpad
   __hip_bfloat16* v_lds = reinterpret_cast<__hip_bfloat16*>(
-      scores + (size_t)group * Lpad);                         // VTILE*D
+      scores + (size_t)GROUP * Lpad);                         // VTILE*D
 
   const int* bt = block_tables + (int64_t)seq * max_nb;
   const int64_t kv_row = (int64_t)KV * D;
 
   // ---- stage Q (group heads of this kv head) into LDS as fp32 ----
-  for (int i = tid; i < group * D; i += blockDim.x) {
-    const int h = kv * group + i / D;
-    q_lds[i] = bf2f(q[((int64_t)seq * H + h) * D + i % D]);
+  for (int i = tid; i < GROUP * D; i += blockDim.x) {
+    const int h = kv * GROUP + i / D;
+    q_lds[i] = bf2f(q[(int64_t)seq * q_row_stride + h * D + i % D]) * scale;
   }
   __syncthreads();
 
-  // ---- phase 1: scores ----
-  const int gid = tid / TPG;            // 16 token-groups in flight
-  const int lane = tid % TPG;
-  constexpr int EPT = D / TPG;          // elems per thread (8 for D=128)
-  for (int t = gid; t < L; t += blockDim.x / TPG) {
+  // ---- phase 1: one token per lane, per-head partials in VGPRs ----
+  for (int t = tid; t < L; t += blockDim.x) {
     const int64_t row = (int64_t)bt[t / block_size] * block_size
                         + t % block_size;
-    const __hip_bfloat16* kp = kcache + row * kv_row + (int64_t)kv * D
-                               + lane * EPT;
-    float kf[EPT];
-    if constexpr (EPT == 8) {
-      bf16x8 kvec = *reinterpret_cast<const bf16x8*>(kp);
+    const bf16x8* kp = reinterpret_cast<const bf16x8*>(
+        kcache + row * kv_row + (int64_t)kv * D);
+    float part[GROUP];
+    #pragma unroll
+    for (int h = 0; h < GROUP; ++h) part[h] = 0.f;
+    #pragma unroll
+    for (int c = 0; c < D / 8; ++c) {
+      const bf16x8 kvec = kp[c];
+      float kf[8];
       #pragma unroll
       for (int j = 0; j < 8; ++j) kf[j] = bf2f(kvec.v[j]);
-    } else {
       #pragma unroll
-      for (int j = 0; j < EPT; ++j) kf[j] = bf2f(kp[j]);
+      for (int h = 0; h < GROUP; ++h) {
+        const float* qh = q_lds + h * D + c * 8;   // wave-uniform broadcast
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) part[h] += qh[j] * kf[j];
+      }
     }
-    for (int h = 0; h < group; ++h) {
-      const float* qh = q_lds + h * D + lane * EPT;
-      float p = 0.f;
-      #pragma unroll
-      for (int j = 0; j < EPT; ++j) p += qh[j] * kf[j];
-      #pragma unroll
-      for (int off = TPG / 2; off > 0; off >>= 1) p += __shfl_xor(p, off, WAVE);
-      if (lane == 0) scores[(size_t)h * Lpad + t] = p * scale;
-    }
+    #pragma unroll
+    for (int h = 0; h < GROUP; ++h) scores[(size_t)h * Lpad + t] = part[h];
   }
   __syncthreads();
 
@@ -92,7 +89,7 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   const int wid = tid / WAVE;
   const int wlane = tid % WAVE;
   const int nw = blockDim.x / WAVE;
-  for (int h = wid; h < group; h += nw) {
+  for (int h = wid; h < GROUP; h += nw) {
     float* s = scores + (size_t)h * Lpad;
     float m = -INFINITY;
     for (int t = wlane; t < L; t += WAVE) m = fmaxf(m, s[t]);
@@ -110,9 +107,9 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
 
   // ---- phase 3: PV accumulation through LDS V tiles ----
   constexpr int DV = D / 8;             // bf16x8 units per row
-  const int units = group * DV;         // <= 8*16 = 128
+  constexpr int UNITS = GROUP * DV;     // <= 16*16 = 256
   float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  const int u = tid;                    // one unit per thread (tid < units)
+  const int u = tid;                    // one unit per thread (tid < UNITS)
   const int uh = u / DV, ud = u % DV;
 
   for (int base = 0; base < L; base += VTILE) {
@@ -127,7 +124,7 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
               vcache + row * kv_row + (int64_t)kv * D + (i % DV) * 8);
     }
     __syncthreads();
-    if (u < units) {
+    if (u < UNITS) {
       const float* ps = scores + (size_t)uh * Lpad + base;
       #pragma unroll 4
       for (int j = 0; j < tile; ++j) {
@@ -140,9 +137,9 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
     __syncthreads();
   }
 
-  if (u < units) {
+  if (u < UNITS) {
     const float inv = 1.f / denom[uh];
-    const int h = kv * group + uh;
+    const int h = kv * GROUP + uh;
     bf16x8 o;
     #pragma unroll
     for (int e = 0; e < 8; ++e) o.v[e] = f2bf(acc[e] * inv);
@@ -150,24 +147,43 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   }
 }
 
+template <int D, int GROUP>
+void launch(const torch::Tensor& q, const torch::Tensor& kcache,
+            const torch::Tensor& vcache, const torch::Tensor& bt,
+            const torch::Tensor& ctx, torch::Tensor& out, int H, int KV,
+            int max_nb, int block_size, int Lpad, float scale, size_t smem,
+            int64_t q_row_stride) {
+  dim3 grid(q.size(0), KV), block(256);
+  hipLaunchKernelGGL((paged_decode_kernel<D, GROUP>), grid, block, smem,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(kcache.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(vcache.data_ptr()),
+                     bt.data_ptr<int>(), ctx.data_ptr<int>(),
+                     reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                     H, KV, max_nb, block_size, Lpad, scale, q_row_stride);
+}
+
 }  // namespace
 
-torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kcache,
-                                     torch::Tensor vcache,
-                                     torch::Tensor block_tables,
-                                     torch::Tensor ctx_lens, double scale) {
+torch::Tensor paged_attention_decode_strided(
+    torch::Tensor q, int64_t n_heads, int64_t head_dim, int64_t q_row_stride,
+    torch::Tensor kcache, torch::Tensor vcache, torch::Tensor block_tables,
+    torch::Tensor ctx_lens, double scale) {
+  // q: (N, q_row_stride) packed rows (e.g. the fused qkv GEMM output);
+  // query heads occupy the first n_heads*head_dim elements of each row.
   TORCH_CHECK(q.is_cuda() && q.is_contiguous());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt &&
               ctx_lens.scalar_type() == at::kInt);
-  const int N = q.size(0), H = q.size(1), D = q.size(2);
+  const int N = q.size(0), H = (int)n_heads, D = (int)head_dim;
   const int KV = kcache.size(2);
   const int block_size = kcache.size(1);
   const int max_nb = block_tables.size(1);
   const int group = H / KV;
   TORCH_CHECK(H % KV == 0 && group <= 16, "GQA group too large");
 
-  auto out = torch::empty_like(q);
+  auto out = torch::empty({N, (int64_t)H, (int64_t)D}, q.options());
   if (N == 0) return out;
 
   const int max_ctx = max_nb * block_size;
@@ -177,27 +193,28 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kcache,
   TORCH_CHECK(smem <= 160 * 1024,
               "context too long for single-pass decode kernel: ", max_ctx);
 
-  dim3 grid(N, KV), block(256);
-  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  if (D == 128) {
-    hipLaunchKernelGGL(paged_decode_kernel<128>, grid, block, smem, stream,
-                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-                       reinterpret_cast<const __hip_bfloat16*>(kcache.data_ptr()),
-                       reinterpret_cast<const __hip_bfloat16*>(vcache.data_ptr()),
-                       block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
-                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
-                       H, KV, max_nb, block_size, Lpad, (float)scale);
-  } else if (D == 64) {
-    hipLaunchKernelGGL(paged_decode_kernel<64>, grid, block, smem, stream,
-                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-                       reinterpret_cast<const __hip_bfloat16*>(kcache.data_ptr()),
-                       reinterpret_cast<const __hip_bfloat16*>(vcache.data_ptr()),
-                       block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
-                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
-                       H, KV, max_nb, block_size, Lpad, (float)scale);
-  } else {
-    TORCH_CHECK(false, "paged decode: head_dim 64 or 128 only, got ", D);
-  }
-  HIP_CHECK_LAST();
-  return out;
+  auto bt = block_tables;
+  auto ctx = ctx_lens;
+  const float sc = (float)scale;
+  #define CASE(DD, GG) \
+    if (D == DD && group == GG) { \
+      launch<DD, GG>(q, kcache, vcache, bt, ctx, out, H, KV, max_nb, \
+                     block_size, Lpad, sc, smem, q_row_stride); \
+      HIP_CHECK_LAST(); return out; }
+  CASE(128, 7) CASE(128, 5) CASE(128, 4) CASE(128, 8) CASE(128, 6)
+  CASE(128, 2) CASE(128, 1) CASE(64, 7) CASE(64, 4) CASE(64, 2) CASE(64, 1)
+  #undef CASE
+  TORCH_CHECK(false, "paged decode: unsupported (head_dim=", D,
+              ", gqa group=", group, ")");
+}
+
+
+torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kcache,
+                                     torch::Tensor vcache,
+                                     torch::Tensor block_tables,
+                                     torch::Tensor ctx_lens, double scale) {
+  TORCH_CHECK(q.dim() == 3);
+  return paged_attention_decode_strided(
+      q.view({q.size(0), q.size(1) * q.size(2)}), q.size(1), q.size(2),
+      q.size(1) * q.size(2), kcache, vcache, block_tables, ctx_lens, scale);
 }

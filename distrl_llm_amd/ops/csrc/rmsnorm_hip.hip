@@ -170,3 +170,67 @@ torch::Tensor rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
   HIP_CHECK_LAST();
   return dx;
 }
+
+// ---- fused residual-add + RMSNorm (decode path): res = x + h;
+// y = rmsnorm(res) * w. Saves one full elementwise pass per layer site.
+namespace {
+
+__global__ void add_rmsnorm_fwd_bf16(const __hip_bfloat16* __restrict__ x,
+                                     const __hip_bfloat16* __restrict__ h,
+                                     const __hip_bfloat16* __restrict__ w,
+                                     __hip_bfloat16* __restrict__ y,
+                                     __hip_bfloat16* __restrict__ res,
+                                     int hidden, float eps) {
+  __shared__ float red[16];
+  const int64_t row = blockIdx.x;
+  const bf16x8* xr = reinterpret_cast<const bf16x8*>(x + row * hidden);
+  const bf16x8* hr = reinterpret_cast<const bf16x8*>(h + row * hidden);
+  const bf16x8* wr = reinterpret_cast<const bf16x8*>(w);
+  bf16x8* yr = reinterpret_cast<bf16x8*>(y + row * hidden);
+  bf16x8* rr = reinterpret_cast<bf16x8*>(res + row * hidden);
+  const int nvec = hidden / 8;
+
+  float ss = 0.f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8 xv = xr[i], hv = hr[i], s;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(xv.v[j]) + bf2f(hv.v[j]);
+      s.v[j] = f2bf(f);
+      ss += f * f;
+    }
+    rr[i] = s;
+  }
+  ss = block_sum(ss, red);
+  const float r = rsqrtf(ss / hidden + eps);
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8 sv = rr[i], wv = wr[i], o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) o.v[j] = f2bf(bf2f(sv.v[j]) * r * bf2f(wv.v[j]));
+    yr[i] = o;
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor h,
+                                           torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && h.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16);
+  const int hidden = x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0);
+  const int64_t rows = x.numel() / hidden;
+  auto y = torch::empty_like(x);
+  auto res = torch::empty_like(x);
+  if (rows == 0) return {y, res};
+  hipLaunchKernelGGL(add_rmsnorm_fwd_bf16, dim3(rows), dim3(256), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(h.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(res.data_ptr()),
+                     hidden, (float)eps);
+  HIP_CHECK_LAST();
+  return {y, res};
+}

@@ -45,13 +45,26 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
   const int row = blockIdx.x;
   const T* lr = logits + (int64_t)row * V;
   // step counter read from device memory so a hipGraph replay of this
-  // kernel draws fresh randomness every step (the graph increments *step)
+  // kernel draws fresh randomness every steps (the graph increments *step)
   const uint64_t seed = (uint64_t)seeds[row] ^ splitmix64(0x5D21u + (uint64_t)*step);
+  constexpr bool BF16 = std::is_same<T, __hip_bfloat16>::value;
+  const int nvec = BF16 ? V / 8 : 0;
 
-  // ---- pass A: max ----
+  // ---- pass A: max (bf16: 16 B/lane vector loads) ----
   float m = -INFINITY;
-  for (int i = threadIdx.x; i < V; i += blockDim.x)
-    m = fmaxf(m, ld(lr, i));
+  if constexpr (BF16) {
+    const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      bf16x8 v = l8[i];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) m = fmaxf(m, bf2f(v.v[j]));
+    }
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += blockDim.x)
+      m = fmaxf(m, ld(lr, i));
+  } else {
+    for (int i = threadIdx.x; i < V; i += blockDim.x)
+      m = fmaxf(m, ld(lr, i));
+  }
   m = block_max(m, red);
 
   const bool filtering = (top_p < 1.f) || (top_k > 0 && top_k < V);
@@ -63,12 +76,36 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
       bin_c[i] = 0;
     }
     __syncthreads();
-    for (int i = threadIdx.x; i < V; i += blockDim.x) {
-      float z = (ld(lr, i) - m) * inv_temp;
-      if (z > -ZRANGE) {
-        int b = min(NBINS - 1, (int)(-z * (NBINS / ZRANGE)));
-        atomicAdd(&bin_p[b], __expf(z));
-        atomicAdd(&bin_c[b], 1);
+    if constexpr (BF16) {
+      const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
+      for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+        bf16x8 v = l8[i];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float z = (bf2f(v.v[j]) - m) * inv_temp;
+          if (z > -ZRANGE) {
+            int b = min(NBINS - 1, (int)(-z * (NBINS / ZRANGE)));
+            atomicAdd(&bin_p[b], __expf(z));
+            atomicAdd(&bin_c[b], 1);
+          }
+        }
+      }
+      for (int i = nvec * 8 + threadIdx.x; i < V; i += blockDim.x) {
+        float z = (ld(lr, i) - m) * inv_temp;
+        if (z > -ZRANGE) {
+          int b = min(NBINS - 1, (int)(-z * (NBINS / ZRANGE)));
+          atomicAdd(&bin_p[b], __expf(z));
+          atomicAdd(&bin_c[b], 1);
+        }
+      }
+    } else {
+      for (int i = threadIdx.x; i < V; i += blockDim.x) {
+        float z = (ld(lr, i) - m) * inv_temp;
+        if (z > -ZRANGE) {
+          int b = min(NBINS - 1, (int)(-z * (NBINS / ZRANGE)));
+          atomicAdd(&bin_p[b], __expf(z));
+          atomicAdd(&bin_c[b], 1);
+        }
       }
     }
     __syncthreads();
@@ -97,13 +134,36 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
   // ---- pass C: Gumbel-argmax over the kept set ----
   float best = -INFINITY;
   int64_t best_i = 0;
-  for (int i = threadIdx.x; i < V; i += blockDim.x) {
-    float z = (ld(lr, i) - m) * inv_temp;
-    if (z >= z_min) {
-      float u = hash_uniform(seed, (uint64_t)i);
-      float g = -__logf(-__logf(u));
-      float key = z + g;
-      if (key > best) { best = key; best_i = i; }
+  if constexpr (BF16) {
+    const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      bf16x8 v = l8[i];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float z = (bf2f(v.v[j]) - m) * inv_temp;
+        if (z >= z_min) {
+          float u = hash_uniform(seed, (uint64_t)(i * 8 + j));
+          float key = z - __logf(-__logf(u));
+          if (key > best) { best = key; best_i = i * 8 + j; }
+        }
+      }
+    }
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += blockDim.x) {
+      float z = (ld(lr, i) - m) * inv_temp;
+      if (z >= z_min) {
+        float u = hash_uniform(seed, (uint64_t)i);
+        float key = z - __logf(-__logf(u));
+        if (key > best) { best = key; best_i = i; }
+      }
+    }
+  } else {
+    for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      float z = (ld(lr, i) - m) * inv_temp;
+      if (z >= z_min) {
+        float u = hash_uniform(seed, (uint64_t)i);
+        float key = z - __logf(-__logf(u));
+        if (key > best) { best = key; best_i = i; }
+      }
     }
   }
   // wave reduce (val, idx)

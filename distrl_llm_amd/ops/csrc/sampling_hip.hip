@@ -34,6 +34,7 @@ template <typename T>
 __global__ __launch_bounds__(256)
 void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
                    float top_p, int top_k, const int64_t* __restrict__ seeds,
+                   const int64_t* __restrict__ step,
                    int64_t* __restrict__ out) {
   __shared__ float red[16];
   __shared__ float bin_p[NBINS];
@@ -44,12 +45,27 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
 
   const int row = blockIdx.x;
   const T* lr = logits + (int64_t)row * V;
-  const uint64_t seed = (uint64_t)seeds[row];
+  // step counter read from device memory so a hipGraph replay of this
+  // kernel draws fresh randomness every steps (the graph increments *step)
+  const uint64_t seed = (uint64_t)seeds[row] ^ splitmix64(0x5D21u + (uint64_t)*step);
+  constexpr bool BF16 = std::is_same<T, __hip_bfloat16>::value;
+  const int nvec = BF16 ? V / 8 : 0;
 
-  // ---- pass A: max ----
+  // ---- pass A: max (bf16: 16 B/lane vector loads) ----
   float m = -INFINITY;
-  for (int i = threadIdx.x; i < V; i += blockDim.x)
-    m = fmaxf(m, ld(lr, i));
+  if constexpr (BF16) {
+    const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      bf16x8 v = l8[i];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) m = fmaxf(m, bf2f(v.v[j]));
+    }
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += blockDim.x)
+      m = fmaxf(m, ld(lr, i));
+  } else {
+    for (int i = threadIdx.x; i < V; i += blockDim.x)
+      m = fmaxf(m, ld(lr, i));
+  }
   m = block_max(m, red);
 
   const bool filtering = (top_p < 1.f) || (top_k > 0 && top_k < V);
@@ -61,12 +77,36 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
       bin_c[i] = 0;
     }
     __syncthreads();
-    for (int i = threadIdx.x; i < V; i += blockDim.x) {
-      float z = (ld(lr, i) - m) * inv_temp;
-      if (z > -ZRANGE) {
-        int b = min(NBINS - 1, (int)(-z * (NBINS / ZRANGE)));
-        atomicAdd(&bin_p[b], __expf(z));
-        atomicAdd(&bin_c[b], 1);
+    if constexpr (BF16) {
+      const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
+      for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+        bf16x8 v = l8[i];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float z = (bf2f(v.v[j]) - m) * inv_temp;
+          if (z > -ZRANGE) {
+            int b = min(NBINS - 1, (int)(-z * (NBINS / ZRANGE)));
+            atomicAdd(&bin_p[b], __expf(z));
+            atomicAdd(&bin_c[b], 1);
+          }
+        }
+      }
+      for (int i = nvec * 8 + threadIdx.x; i < V; i += blockDim.x) {
+        float z = (ld(lr, i) - m) * inv_temp;
+        if (z > -ZRANGE) {
+          int b = min(NBINS - 1, (int)(-z * (NBINS / ZRANGE)));
+          atomicAdd(&bin_p[b], __expf(z));
+          atomicAdd(&bin_c[b], 1);
+        }
+      }
+    } else {
+      for (int i = threadIdx.x; i < V; i += blockDim.x) {
+        float z = (ld(lr, i) - m) * inv_temp;
+        if (z > -ZRANGE) {
+          int b = min(NBINS - 1, (int)(-z * (NBINS / ZRANGE)));
+          atomicAdd(&bin_p[b], __expf(z));
+          atomicAdd(&bin_c[b], 1);
+        }
       }
     }
     __syncthreads();
@@ -95,13 +135,36 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
   // ---- pass C: Gumbel-argmax over the kept set ----
   float best = -INFINITY;
   int64_t best_i = 0;
-  for (int i = threadIdx.x; i < V; i += blockDim.x) {
-    float z = (ld(lr, i) - m) * inv_temp;
-    if (z >= z_min) {
-      float u = hash_uniform(seed, (uint64_t)i);
-      float g = -__logf(-__logf(u));
-      float key = z + g;
-      if (key > best) { best = key; best_i = i; }
+  if constexpr (BF16) {
+    const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      bf16x8 v = l8[i];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float z = (bf2f(v.v[j]) - m) * inv_temp;
+        if (z >= z_min) {
+          float u = hash_uniform(seed, (uint64_t)(i * 8 + j));
+          float key = z - __logf(-__logf(u));
+          if (key > best) { best = key; best_i = i * 8 + j; }
+        }
+      }
+    }
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += blockDim.x) {
+      float z = (ld(lr, i) - m) * inv_temp;
+      if (z >= z_min) {
+        float u = hash_uniform(seed, (uint64_t)i);
+        float key = z - __logf(-__logf(u));
+        if (key > best) { best = key; best_i = i; }
+      }
+    }
+  } else {
+    for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      float z = (ld(lr, i) - m) * inv_temp;
+      if (z >= z_min) {
+        float u = hash_uniform(seed, (uint64_t)i);
+        float key = z - __logf(-__logf(u));
+        if (key > best) { best = key; best_i = i; }
+      }
     }
   }
   // wave reduce (val, idx)
@@ -130,10 +193,12 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
 }  // namespace
 
 torch::Tensor sample_tokens(torch::Tensor logits, double temperature,
-                            double top_p, int64_t top_k, torch::Tensor seeds) {
+                            double top_p, int64_t top_k, torch::Tensor seeds,
+                            torch::Tensor step) {
   TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() && logits.dim() == 2);
   TORCH_CHECK(temperature > 0.0, "temperature 0 is greedy: use argmax");
   TORCH_CHECK(seeds.scalar_type() == at::kLong);
+  TORCH_CHECK(step.scalar_type() == at::kLong && step.is_cuda());
   const int B = logits.size(0), V = logits.size(1);
   auto out = torch::empty({B}, logits.options().dtype(at::kLong));
   if (B == 0) return out;
@@ -144,13 +209,14 @@ torch::Tensor sample_tokens(torch::Tensor logits, double temperature,
     hipLaunchKernelGGL(sample_kernel<__hip_bfloat16>, grid, block, 0, stream,
                        reinterpret_cast<const __hip_bfloat16*>(logits.data_ptr()),
                        V, inv_t, (float)top_p, (int)top_k,
-                       seeds.data_ptr<int64_t>(), out.data_ptr<int64_t>());
+                       seeds.data_ptr<int64_t>(), step.data_ptr<int64_t>(),
+                       out.data_ptr<int64_t>());
   } else {
     TORCH_CHECK(logits.scalar_type() == at::kFloat);
     hipLaunchKernelGGL(sample_kernel<float>, grid, block, 0, stream,
                        logits.data_ptr<float>(), V, inv_t, (float)top_p,
                        (int)top_k, seeds.data_ptr<int64_t>(),
-                       out.data_ptr<int64_t>());
+                       step.data_ptr<int64_t>(), out.data_ptr<int64_t>());
   }
   HIP_CHECK_LAST();
   return out;

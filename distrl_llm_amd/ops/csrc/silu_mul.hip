@@ -113,3 +113,42 @@ std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor gate,
   HIP_CHECK_LAST();
   return {dgate, dup};
 }
+
+// ---- packed variant: input (N, 2F) = [gate | up] rows straight out of
+// the fused gate|up GEMM; output (N, F). Decode fast path.
+namespace {
+__global__ void silu_mul_packed_bf16(const __hip_bfloat16* __restrict__ gu,
+                                     __hip_bfloat16* __restrict__ out,
+                                     int fvec /* F/8 */) {
+  const int64_t row = blockIdx.x;
+  const bf16x8* g8 = reinterpret_cast<const bf16x8*>(gu + row * 2 * fvec * 8);
+  const bf16x8* u8 = g8 + fvec;
+  bf16x8* o8 = reinterpret_cast<bf16x8*>(out + row * fvec * 8);
+  for (int i = threadIdx.x; i < fvec; i += blockDim.x) {
+    bf16x8 g = g8[i], u = u8[i], o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g.v[j]);
+      o.v[j] = f2bf(gf / (1.f + __expf(-gf)) * bf2f(u.v[j]));
+    }
+    o8[i] = o;
+  }
+}
+}  // namespace
+
+torch::Tensor silu_mul_packed(torch::Tensor gu) {
+  TORCH_CHECK(gu.is_cuda() && gu.is_contiguous() && gu.dim() == 2);
+  TORCH_CHECK(gu.scalar_type() == at::kBFloat16);
+  const int64_t N = gu.size(0);
+  const int64_t F = gu.size(1) / 2;
+  TORCH_CHECK(F % 8 == 0);
+  auto out = torch::empty({N, F}, gu.options());
+  if (N == 0) return out;
+  hipLaunchKernelGGL(silu_mul_packed_bf16, dim3(N), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     reinterpret_cast<const __hip_bfloat16*>(gu.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                     (int)(F / 8));
+  HIP_CHECK_LAST();
+  return out;
+}
