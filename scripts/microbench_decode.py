@@ -39,6 +39,8 @@ def main():
     engine = Engine(model, EngineConfig(max_seq_length=1550,
                                         gpu_memory_utilization=0.35),
                     device=dev, seed=0)
+    if engine.fused is not None:
+        engine.fused.refresh()
     sp = SamplingParams(max_tokens=1200, temperature=1.2, n=1, top_p=0.95)
 
     # fabricate prompts and prefill
