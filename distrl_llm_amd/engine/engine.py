@@ -239,11 +239,37 @@ class Engine:
         qs, kvs = s.q_size, s.kv_size
         eps = s.rms_norm_eps
         lws = self.fused.layers
+        nf4 = self.fused.nf4
+        r = self.fused.lora_r
         pos32 = positions.to(torch.int32)
+        N_batch = input_ids.shape[0]
+
+        def proj(x, lw, site, N_out, K_in, r_mult):
+            """One fused projection: nf4 GEMM + LoRA (or merged bf16)."""
+            if not nf4:
+                if site == "qkv":
+                    return F.linear(x, lw.qkv_w, lw.qkv_b)
+                return F.linear(x, getattr(lw, f"{site}_w"))
+            bias = lw.qkv_b if site == "qkv" else None
+            if r > 0:
+                rt = r * r_mult
+                u = torch.empty(N_batch, rt, dtype=torch.float32,
+                                device=x.device)
+                ksplit = max(1, min(8, 512 // max(1, ((N_batch + 15) // 16)
+                                                 * (rt // 16))))
+                ext.lora_u(x, getattr(lw, f"{site}_afrag"), u, rt, ksplit)
+                return ext.nf4_gemm(x, getattr(lw, f"{site}_w4"),
+                                    getattr(lw, f"{site}_amax"), bias, u,
+                                    getattr(lw, f"{site}_bfrag"),
+                                    N_out, K_in, rt)
+            return ext.nf4_gemm(x, getattr(lw, f"{site}_w4"),
+                                getattr(lw, f"{site}_amax"), bias, None,
+                                None, N_out, K_in, 0)
+
         res = m.model.embed_tokens(input_ids)
         h = ext.rmsnorm_fwd(res, lws[0].in_norm, eps)
         for li, lw in enumerate(lws):
-            qkv = F.linear(h, lw.qkv_w, lw.qkv_b)
+            qkv = proj(h, lw, "qkv", qs + 2 * kvs, s.hidden_size, 3)
             ext.rope_scatter_qkv(qkv, pos32, slot_mapping, self._inv_freq,
                                  self.pool.key[li], self.pool.value[li],
                                  s.num_heads, s.num_kv_heads, s.head_dim)
@@ -251,11 +277,12 @@ class Engine:
                 qkv, s.num_heads, s.head_dim, qs + 2 * kvs,
                 self.pool.key[li], self.pool.value[li], block_tables,
                 context_lens, self.scale)
-            o = F.linear(attn.view(-1, qs), lw.o_w)
+            o = proj(attn.view(-1, qs), lw, "o", s.hidden_size, qs, 1)
             h, res = ext.add_rmsnorm_fwd(res, o, lw.post_norm, eps)
-            gu = F.linear(h, lw.gateup_w)
+            gu = proj(h, lw, "gateup", 2 * s.intermediate_size,
+                      s.hidden_size, 2)
             act = ext.silu_mul_packed(gu)
-            d = F.linear(act, lw.down_w)
+            d = proj(act, lw, "down", s.hidden_size, s.intermediate_size, 1)
             next_w = (lws[li + 1].in_norm if li + 1 < len(lws)
                       else m.model.norm.weight)
             h, res = ext.add_rmsnorm_fwd(res, d, next_w, eps)

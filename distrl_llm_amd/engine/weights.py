@@ -22,7 +22,13 @@ import torch
 
 class LayerWeights:
     __slots__ = ("qkv_w", "qkv_b", "o_w", "gateup_w", "down_w",
-                 "in_norm", "post_norm")
+                 "in_norm", "post_norm",
+                 # nf4 fragment packs (base, built once) + adapter fragment
+                 # packs (rebuilt per refresh) for the fused nf4 GEMM path
+                 "qkv_w4", "qkv_amax", "o_w4", "o_amax",
+                 "gateup_w4", "gateup_amax", "down_w4", "down_amax",
+                 "qkv_afrag", "qkv_bfrag", "o_afrag", "o_bfrag",
+                 "gateup_afrag", "gateup_bfrag", "down_afrag", "down_bfrag")
 
 
 def _effective(mod) -> torch.Tensor:
@@ -34,32 +40,85 @@ def _effective(mod) -> torch.Tensor:
 
 
 class FusedWeights:
-    def __init__(self, model):
+    def __init__(self, model, use_nf4: bool = True):
         self.model = model
         self.layers: List[LayerWeights] = []
         self._built = False
+        # nf4 fused-GEMM decode path: available when the model carries the
+        # packed nf4 sidecars (load_in_4bit) and the kernels are built
+        q0 = model.model.layers[0].self_attn.q_proj
+        s = model.spec
+        shapes_ok = (
+            (s.q_size + 2 * s.kv_size) % 256 == 0
+            and s.hidden_size % 256 == 0
+            and (2 * s.intermediate_size) % 256 == 0
+            and s.hidden_size % 64 == 0 and s.q_size % 64 == 0
+            and s.intermediate_size % 64 == 0)
+        self.nf4 = bool(use_nf4 and q0.weight_nf4 is not None
+                        and q0.weight.is_cuda and shapes_ok)
+        self.lora_r = q0.r
+
+    @staticmethod
+    def _nf4_cat(mods):
+        """Concatenate the flat nf4 packs / absmax of modules along N
+        (rows are K-contiguous, so flat concatenation is row concat)."""
+        packed = torch.cat([m.weight_nf4 for m in mods])
+        absmax = torch.cat([m.weight_absmax for m in mods])
+        N = sum(m.out_features for m in mods)
+        K = mods[0].in_features
+        return packed, absmax, N, K
+
+    def _build_nf4_base(self, lw: "LayerWeights", layer):
+        from ..models.quant import prepack_nf4_fragments
+        at, mlp = layer.self_attn, layer.mlp
+        for name, mods in (("qkv", [at.q_proj, at.k_proj, at.v_proj]),
+                           ("o", [at.o_proj]),
+                           ("gateup", [mlp.gate_proj, mlp.up_proj]),
+                           ("down", [mlp.down_proj])):
+            packed, absmax, N, K = self._nf4_cat(mods)
+            w4f, amaxf = prepack_nf4_fragments(packed, absmax, N, K)
+            setattr(lw, f"{name}_w4", w4f)
+            setattr(lw, f"{name}_amax", amaxf)
+
+    def _refresh_nf4_adapters(self, lw: "LayerWeights", layer):
+        """Per-round: stack the A matrices and build the block-diagonal,
+        scale-folded B matrix per fused site, prepacked into fragments."""
+        from ..models.quant import prepack_bf16_fragments
+        at, mlp = layer.self_attn, layer.mlp
+        for name, mods in (("qkv", [at.q_proj, at.k_proj, at.v_proj]),
+                           ("o", [at.o_proj]),
+                           ("gateup", [mlp.gate_proj, mlp.up_proj]),
+                           ("down", [mlp.down_proj])):
+            A = torch.cat([m.lora_A.detach() for m in mods], dim=0)  # (r_tot, K)
+            r_tot = A.shape[0]
+            N = sum(m.out_features for m in mods)
+            B = A.new_zeros(N, r_tot)
+            n0, r0 = 0, 0
+            for m in mods:
+                B[n0:n0 + m.out_features, r0:r0 + m.r] = (
+                    m.lora_B.detach() * m.scale)
+                n0 += m.out_features
+                r0 += m.r
+            setattr(lw, f"{name}_afrag",
+                    prepack_bf16_fragments(A.to(torch.bfloat16)))
+            setattr(lw, f"{name}_bfrag",
+                    prepack_bf16_fragments(B.to(torch.bfloat16)))
 
     @torch.no_grad()
     def refresh(self):
         """(Re)build the merged decode weights from the current base + LoRA
-        tensors. Called once per weight sync, never inside the decode loop."""
+        tensors. Called once per weight sync, never inside the decode loop.
+
+        nf4 mode: the quantized base fragments are built ONCE (frozen);
+        only the adapter fragment packs (A stacks, block-diagonal scaled B)
+        are rebuilt per refresh — the adapter stays exact bf16.
+        """
         model = self.model
         first = not self._built
         for li, layer in enumerate(model.model.layers):
             at = layer.self_attn
-            mlp = layer.mlp
-            qkv = torch.cat([_effective(at.q_proj), _effective(at.k_proj),
-                             _effective(at.v_proj)], dim=0).contiguous()
-            gateup = torch.cat([_effective(mlp.gate_proj),
-                                _effective(mlp.up_proj)], dim=0).contiguous()
-            o_w = _effective(at.o_proj).contiguous()
-            down_w = _effective(mlp.down_proj).contiguous()
             if first:
                 lw = LayerWeights()
-                lw.qkv_w = qkv
-                lw.gateup_w = gateup
-                lw.o_w = o_w
-                lw.down_w = down_w
                 if at.q_proj.bias is not None:
                     lw.qkv_b = torch.cat([at.q_proj.bias, at.k_proj.bias,
                                           at.v_proj.bias]).contiguous()
@@ -70,8 +129,29 @@ class FusedWeights:
                 self.layers.append(lw)
             else:
                 lw = self.layers[li]
-                lw.qkv_w.copy_(qkv)
-                lw.gateup_w.copy_(gateup)
-                lw.o_w.copy_(o_w)
-                lw.down_w.copy_(down_w)
+            if self.nf4:
+                if first:
+                    self._build_nf4_base(lw, layer)
+                self._refresh_nf4_adapters(lw, layer)
+            else:
+                self._refresh_merged(lw, layer, first)
         self._built = True
+
+    def _refresh_merged(self, lw: "LayerWeights", layer, first: bool):
+        at, mlp = layer.self_attn, layer.mlp
+        qkv = torch.cat([_effective(at.q_proj), _effective(at.k_proj),
+                         _effective(at.v_proj)], dim=0).contiguous()
+        gateup = torch.cat([_effective(mlp.gate_proj),
+                            _effective(mlp.up_proj)], dim=0).contiguous()
+        o_w = _effective(at.o_proj).contiguous()
+        down_w = _effective(mlp.down_proj).contiguous()
+        if first:
+            lw.qkv_w = qkv
+            lw.gateup_w = gateup
+            lw.o_w = o_w
+            lw.down_w = down_w
+        else:
+            lw.qkv_w.copy_(qkv)
+            lw.gateup_w.copy_(gateup)
+            lw.o_w.copy_(o_w)
+            lw.down_w.copy_(down_w)

@@ -35,6 +35,14 @@ std::vector<torch::Tensor> logprob_lse_fwd(torch::Tensor logits,
                                            torch::Tensor targets);
 torch::Tensor logprob_loss_bwd(torch::Tensor logits, torch::Tensor targets,
                                torch::Tensor w, torch::Tensor lse);
+torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
+                       c10::optional<torch::Tensor> bias,
+                       c10::optional<torch::Tensor> u,
+                       c10::optional<torch::Tensor> bfrag,
+                       int64_t N, int64_t K, int64_t r);
+void lora_u(torch::Tensor x, torch::Tensor afrag, torch::Tensor u,
+            int64_t r, int64_t ksplit);
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
 void adam8bit_step(torch::Tensor p, torch::Tensor g, torch::Tensor m_q,
                    torch::Tensor v_q, torch::Tensor m_absmax,
                    torch::Tensor v_absmax, double lr, double b1, double b2,
@@ -63,4 +71,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logprob_loss_bwd", &logprob_loss_bwd,
         "dlogits for the masked PG/GRPO loss");
   m.def("adam8bit_step", &adam8bit_step, "fused blockwise 8-bit Adam step");
+  m.def("nf4_gemm", &nf4_gemm,
+        "fused nf4-dequant MFMA GEMM with bias + LoRA-B epilogue");
+  m.def("lora_u", &lora_u, "split-K LoRA A projection (u = x @ A^T)");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA mapping probe");
 }

@@ -1,0 +1,338 @@
+#include "hip/hip_runtime.h"
+// Fused nf4-dequant + MFMA GEMM — the 4-bit weight path of both the
+// generation engine and the learner base projections (SURVEY.md §2.4:
+// "hand-written fused nf4-dequant+MFMA GEMM" north star; replaces
+// bitsandbytes kDequantizeBlockwise + cuBLAS in the reference's stack).
+//
+//   y[M,N] = x[M,K] @ dequant(W4)[N,K]^T (+ bias) (+ u[M,r] @ Bs[N,r]^T)
+//
+// Weights are PREPACKED into MFMA B-fragment order at load time (we own
+// the format): one dword per (n-tile, k-step, lane) holding the lane's 8
+// nf4 nibbles, so the weight stream is perfectly coalesced 256 B per wave
+// instruction and dequant is 8 LUT-mul-cvt ops straight into the MFMA
+// B fragment. absmax is fragment-ordered fp32. The LoRA correction rides
+// the same accumulators: u = x@A^T (computed by the split-K lora_u kernel
+// below) enters as ONE extra MFMA k-step per rank-32 block against the
+// bf16-prepacked, scale-folded B matrix — the adapter stays exact bf16
+// (never quantized), matching the reference's nf4-base + bf16-LoRA
+// semantics (distributed_actor.py:58-66 + helper.py:25-46).
+//
+// Fragment mapping (mfma_f32_16x16x32_bf16, verified by mfma_probe test):
+//   A: lane l holds A[row = l&15][k = (l>>4)*8 + j], j = 0..7
+//   B: lane l holds B[col = l&15][k = (l>>4)*8 + j]   (B consumed as N x K)
+//   D: lane l, reg r -> row = (l>>4)*4 + r, col = l&15
+//
+// Geometry: 256 threads = 4 waves; block tile = BN 256 cols (4 waves x 4
+// n-tiles) x BM = 16*MT rows; K loop in 64-deep chunks with the x tile
+// staged in LDS behind an XOR swizzle (byte ^= (row&15)<<4) so the
+// ds_read_b128 A-fragment reads are conflict-free (guide G4/T2).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16v8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+namespace {
+
+__constant__ float NF4_LUT[16] = {
+    -1.0f, -0.6961928009986877f, -0.5250730514526367f, -0.39491748809814453f,
+    -0.28444138169288635f, -0.18477343022823334f, -0.09105003625154495f, 0.0f,
+    0.07958029955625534f, 0.16093020141124725f, 0.24611230194568634f,
+    0.33791524171829224f, 0.44070982933044434f, 0.5626170039176941f,
+    0.7229568362236023f, 1.0f};
+
+DEV_INLINE bf16v8 lds_read_frag(const char* base, int byte_off) {
+  return *reinterpret_cast<const bf16v8*>(base + byte_off);
+}
+
+template <int MT>
+__global__ __launch_bounds__(256)
+void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
+                     const uint32_t* __restrict__ w4f,       // frag-packed
+                     const float* __restrict__ amaxf,        // frag-ordered
+                     const __hip_bfloat16* __restrict__ bias,  // (N) | null
+                     const float* __restrict__ u,            // (M, r) | null
+                     const uint32_t* __restrict__ bfrag,     // B frag | null
+                     __hip_bfloat16* __restrict__ y,         // (M, N)
+                     int M, int N, int K, int r) {
+  constexpr int BM = 16 * MT;
+  const int tid = threadIdx.x;
+  const int l = tid & 63;
+  const int wave = tid >> 6;
+  const int lrow = l & 15;          // fragment row/col index
+  const int lk = l >> 4;            // fragment k-group
+  // grid: x = m-block (fast dim), y = n-block — adjacent linear block ids
+  // share the same weight panel, so the XCD L2 absorbs re-reads when the
+  // m dimension is tiled (T1 locality without an explicit remap)
+  const int mbase = blockIdx.x * BM;
+  const int ntile0 = blockIdx.y * 16 + wave * 4;  // this wave's 4 n-tiles
+  const int ksteps = K / 32;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* x_lds = smem;                         // BM * 128 bytes (64 bf16/row)
+  char* u_lds = smem + BM * 128;              // BM * r * 2 bytes
+
+  f32x4 acc[MT][4];
+  #pragma unroll
+  for (int mt = 0; mt < MT; ++mt)
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) acc[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // stage u (fp32 -> bf16) once; it is tiny (BM x r)
+  if (u != nullptr) {
+    for (int i = tid; i < BM * (r / 8); i += 256) {
+      const int row = i / (r / 8);
+      const int unit = i % (r / 8);
+      bf16x8 s;
+      if (mbase + row < M) {
+        const float* up = u + (int64_t)(mbase + row) * r + unit * 8;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) s.v[j] = f2bf(up[j]);
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) s.v[j] = f2bf(0.f);
+      }
+      *reinterpret_cast<bf16x8*>(u_lds + (int64_t)row * r * 2 + unit * 16) = s;
+    }
+  }
+
+  // ---- main K loop: 64-deep chunks ----
+  for (int kb = 0; kb < K / 64; ++kb) {
+    __syncthreads();
+    // stage x chunk (BM x 64 bf16), XOR-swizzled
+    for (int i = tid; i < BM * 8; i += 256) {
+      const int row = i >> 3;
+      const int unit = i & 7;
+      bf16x8 s;
+      if (mbase + row < M) {
+        s = *reinterpret_cast<const bf16x8*>(
+            x + (int64_t)(mbase + row) * K + kb * 64 + unit * 8);
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) s.v[j] = f2bf(0.f);
+      }
+      const int off = (row * 128 + unit * 16) ^ ((row & 15) << 4);
+      *reinterpret_cast<bf16x8*>(x_lds + off) = s;
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kstep = kb * 2 + ks;
+      // A fragments for every m-tile
+      bf16v8 afrag[MT];
+      #pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+        const int row = mt * 16 + lrow;
+        const int off = (row * 128 + (ks * 4 + lk) * 16) ^ ((row & 15) << 4);
+        afrag[mt] = lds_read_frag(x_lds, off);
+      }
+      // 4 n-tiles: dequant B fragment + MFMA
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int ntg = ntile0 + nt;
+        const uint32_t wbits = w4f[((int64_t)ntg * ksteps + kstep) * 64 + l];
+        const float am = amaxf[((int64_t)ntg * (K / 64) + (kstep >> 1)) * 16
+                               + lrow];
+        bf16v8 bfr;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bfr[j] = (__bf16)(NF4_LUT[(wbits >> (4 * j)) & 0xF] * am);
+        #pragma unroll
+        for (int mt = 0; mt < MT; ++mt)
+          acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mt], bfr, acc[mt][nt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- LoRA epilogue: one extra MFMA k-step per rank-32 block ----
+  if (u != nullptr) {
+    __syncthreads();
+    for (int rs = 0; rs < r / 32; ++rs) {
+      bf16v8 afrag[MT];
+      #pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+        const int row = mt * 16 + lrow;
+        afrag[mt] = *reinterpret_cast<const bf16v8*>(
+            u_lds + (int64_t)row * r * 2 + (rs * 32 + lk * 8) * 2);
+      }
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int ntg = ntile0 + nt;
+        const uint4 bw = *reinterpret_cast<const uint4*>(
+            bfrag + ((int64_t)ntg * (r / 32) + rs) * 64 * 4 + l * 4);
+        const bf16v8 bfr = *reinterpret_cast<const bf16v8*>(&bw);
+        #pragma unroll
+        for (int mt = 0; mt < MT; ++mt)
+          acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mt], bfr, acc[mt][nt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: bias + store ----
+  #pragma unroll
+  for (int nt = 0; nt < 4; ++nt) {
+    const int n = (ntile0 + nt) * 16 + lrow;
+    const float bv = (bias != nullptr) ? bf2f(bias[n]) : 0.f;
+    #pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+      #pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int m = mbase + mt * 16 + lk * 4 + rr;
+        if (m < M)
+          y[(int64_t)m * N + n] = f2bf(acc[mt][nt][rr] + bv);
+      }
+    }
+  }
+}
+
+// ---- split-K LoRA A kernel: u[M,r] = x[M,K] @ A^T, fp32 atomic combine.
+// A is prepacked into B-fragment order (bf16, 16 B per (r-tile, k-step,
+// lane)); x A-fragments are read straight from global (x is L2-resident
+// at decode batch sizes). One wave per (m-tile, r-tile, k-split) block.
+__global__ __launch_bounds__(64)
+void lora_u_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
+                   const uint32_t* __restrict__ afrag,     // (r/16, K/32, 64, 4)
+                   float* __restrict__ u,                  // (M, r) zeroed
+                   int M, int K, int r, int ksplit) {
+  const int l = threadIdx.x;
+  const int lrow = l & 15;
+  const int lk = l >> 4;
+  const int mt = blockIdx.x;
+  const int rt = blockIdx.y;
+  const int ks_id = blockIdx.z;
+  const int ksteps = K / 32;
+  const int per = (ksteps + ksplit - 1) / ksplit;
+  const int k0 = ks_id * per;
+  const int k1 = min(ksteps, k0 + per);
+
+  const int m = mt * 16 + lrow;
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int kstep = k0; kstep < k1; ++kstep) {
+    bf16v8 af;
+    if (m < M) {
+      af = *reinterpret_cast<const bf16v8*>(
+          x + (int64_t)m * K + kstep * 32 + lk * 8);
+    } else {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) af[j] = (__bf16)0.f;
+    }
+    const uint4 bw = *reinterpret_cast<const uint4*>(
+        afrag + ((int64_t)rt * ksteps + kstep) * 64 * 4 + l * 4);
+    const bf16v8 bf = *reinterpret_cast<const bf16v8*>(&bw);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+  }
+  #pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+    const int mrow = mt * 16 + lk * 4 + rr;
+    if (mrow < M)
+      atomicAdd(&u[(int64_t)mrow * r + rt * 16 + lrow], acc[rr]);
+  }
+}
+
+// ---- single-tile MFMA mapping probe (numerics test support) ----
+__global__ void mfma_probe_kernel(const __hip_bfloat16* a,  // (16, 32)
+                                  const __hip_bfloat16* b,  // (16, 32)
+                                  float* c) {               // (16, 16)
+  const int l = threadIdx.x;
+  bf16v8 av, bv;
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    av[j] = (__bf16)bf2f(a[(l & 15) * 32 + (l >> 4) * 8 + j]);
+    bv[j] = (__bf16)bf2f(b[(l & 15) * 32 + (l >> 4) * 8 + j]);
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc, 0, 0, 0);
+  #pragma unroll
+  for (int rr = 0; rr < 4; ++rr)
+    c[((l >> 4) * 4 + rr) * 16 + (l & 15)] = acc[rr];
+}
+
+}  // namespace
+
+torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
+                       c10::optional<torch::Tensor> bias,
+                       c10::optional<torch::Tensor> u,
+                       c10::optional<torch::Tensor> bfrag,
+                       int64_t N, int64_t K, int64_t r) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(w4f.scalar_type() == at::kInt || w4f.scalar_type() == at::kUInt32);
+  const int M = x.size(0);
+  TORCH_CHECK(x.size(1) == K);
+  TORCH_CHECK(N % 256 == 0, "nf4_gemm: N must be a multiple of 256, got ", N);
+  TORCH_CHECK(K % 64 == 0);
+  auto y = torch::empty({(int64_t)M, N}, x.options());
+  if (M == 0) return y;
+
+  const bool has_lora = u.has_value() && r > 0;
+  if (has_lora) {
+    TORCH_CHECK(r % 32 == 0 && bfrag.has_value());
+    TORCH_CHECK(u->scalar_type() == at::kFloat && u->is_contiguous());
+  }
+
+  // pick MT: largest tile with a reasonably filled grid
+  int mt = std::min<int>((M + 15) / 16, 5);
+  while (mt > 1 && (N / 256) * ((M + 16 * mt - 1) / (16 * mt)) < 160) --mt;
+  const int BM = 16 * mt;
+  dim3 grid((M + BM - 1) / BM, N / 256), block(256);
+  size_t smem = (size_t)BM * 128 + (has_lora ? (size_t)BM * r * 2 : 0);
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const __hip_bfloat16* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0)
+    bias_p = reinterpret_cast<const __hip_bfloat16*>(bias->data_ptr());
+  const float* u_p = has_lora ? u->data_ptr<float>() : nullptr;
+  const uint32_t* bf_p = has_lora
+      ? reinterpret_cast<const uint32_t*>(bfrag->data_ptr()) : nullptr;
+
+  #define LAUNCH(MTV) \
+    hipLaunchKernelGGL(nf4_gemm_kernel<MTV>, grid, block, smem, stream, \
+        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()), \
+        reinterpret_cast<const uint32_t*>(w4f.data_ptr()), \
+        amaxf.data_ptr<float>(), bias_p, u_p, bf_p, \
+        reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), \
+        M, (int)N, (int)K, (int)r)
+  switch (mt) {
+    case 1: LAUNCH(1); break;
+    case 2: LAUNCH(2); break;
+    case 3: LAUNCH(3); break;
+    case 4: LAUNCH(4); break;
+    default: LAUNCH(5); break;
+  }
+  #undef LAUNCH
+  HIP_CHECK_LAST();
+  return y;
+}
+
+void lora_u(torch::Tensor x, torch::Tensor afrag, torch::Tensor u,
+            int64_t r, int64_t ksplit) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(u.scalar_type() == at::kFloat && u.is_contiguous());
+  const int M = x.size(0), K = x.size(1);
+  TORCH_CHECK(r % 16 == 0 && K % 32 == 0);
+  u.zero_();
+  dim3 grid((M + 15) / 16, r / 16, ksplit), block(64);
+  hipLaunchKernelGGL(lora_u_kernel, grid, block, 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                     reinterpret_cast<const uint32_t*>(afrag.data_ptr()),
+                     u.data_ptr<float>(), M, K, (int)r, (int)ksplit);
+  HIP_CHECK_LAST();
+}
+
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16);
+  auto c = torch::empty({16, 16}, a.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     reinterpret_cast<const __hip_bfloat16*>(a.contiguous().data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(b.contiguous().data_ptr()),
+                     c.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return c;
+}

@@ -240,3 +240,102 @@ def test_adam8bit_kernel_matches_cpu_path():
         o_cpu.step()
     torch.testing.assert_close(p_gpu.cpu(), p_cpu.detach(), rtol=1e-3,
                                atol=1e-3)
+
+
+# ----------------------------------------------------- nf4 MFMA GEMM
+
+def test_mfma_mapping_probe(ext):
+    """Verify the assumed 16x16x32 bf16 fragment mapping: probe computes
+    A @ B^T under the documented lane layout. Asymmetric operands so a
+    transposed mapping cannot pass (guide G9)."""
+    torch.manual_seed(10)
+    a = torch.randn(16, 32, device=_dev(), dtype=torch.bfloat16)
+    b = torch.randn(16, 32, device=_dev(), dtype=torch.bfloat16) * 0.5
+    b[3] += 2.0  # extra asymmetry
+    c = ext.mfma_probe(a, b)
+    ref = a.float() @ b.float().t()
+    torch.testing.assert_close(c, ref, rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("M,N,K,r,bias", [
+    (160, 512, 512, 0, False),
+    (7, 256, 64, 0, True),
+    (160, 4608, 3584, 96, True),
+    (33, 2048, 1024, 64, False),
+])
+def test_nf4_gemm(ext, M, N, K, r, bias):
+    from distrl_llm_amd.models.quant import (prepack_bf16_fragments,
+                                             prepack_nf4_fragments)
+    from distrl_llm_amd.ops import reference as R
+    torch.manual_seed(11)
+    dev = _dev()
+    w = torch.randn(N, K, device=dev) * 0.05
+    packed, absmax = R.quantize_nf4(w, 64)
+    w4f, amaxf = prepack_nf4_fragments(packed, absmax, N, K)
+    wdq = R.dequantize_nf4(packed, absmax, (N, K), 64).to(torch.bfloat16)
+    x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+    b = (torch.randn(N, device=dev, dtype=torch.bfloat16)
+         if bias else None)
+
+    ref = x.float() @ wdq.float().t()
+    if bias:
+        ref = ref + b.float()
+    u = bfr = None
+    if r > 0:
+        u32 = torch.randn(M, r, device=dev) * 0.3
+        B = (torch.randn(N, r, device=dev) * 0.05).to(torch.bfloat16)
+        bfr = prepack_bf16_fragments(B)
+        u = u32.contiguous()
+        ref = ref + u32.to(torch.bfloat16).float() @ B.float().t()
+
+    y = ext.nf4_gemm(x, w4f, amaxf, b, u, bfr, N, K, r)
+    torch.testing.assert_close(y.float(), ref, rtol=3e-2, atol=3e-1)
+
+
+def test_lora_u(ext):
+    from distrl_llm_amd.models.quant import prepack_bf16_fragments
+    torch.manual_seed(12)
+    dev = _dev()
+    M, K, r = 160, 3584, 96
+    x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+    A = (torch.randn(r, K, device=dev) * 0.05).to(torch.bfloat16)
+    afrag = prepack_bf16_fragments(A)
+    u = torch.empty(M, r, device=dev, dtype=torch.float32)
+    ext.lora_u(x, afrag, u, r, 4)
+    ref = x.float() @ A.float().t()
+    torch.testing.assert_close(u, ref, rtol=2e-2, atol=5e-1)
+
+
+def test_engine_nf4_matches_merged():
+    """The nf4 fused-GEMM decode path must agree with the merged-bf16 path
+    up to nf4 quantization of the base weights (same quantized image both
+    sides, adapters exact): greedy outputs identical."""
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    spec = get_spec("small-qwen2")
+    dev = torch.device("cuda:0")
+    model = CausalLM(spec, lora_r=8, lora_alpha=16, dtype=torch.bfloat16,
+                     device=dev).random_init(21)
+    model.quantize_nf4_()
+    # give LoRA a nonzero contribution
+    with torch.no_grad():
+        for p in model.parameters():
+            if p.requires_grad:
+                p.add_(torch.randn_like(p) * 0.02)
+    cfg = EngineConfig(max_seq_length=256, kv_block_size=16,
+                       num_kv_blocks=512, max_num_seqs=64)
+    prompts = [list(range(1, 40)), [5, 7, 11, 13]]
+    sp = SamplingParams(max_tokens=8, temperature=0.0, n=1)
+
+    eng_nf4 = Engine(model, cfg, device=dev, seed=0)
+    assert eng_nf4.fused.nf4, "nf4 path must be active for this spec"
+    out_nf4 = eng_nf4.generate(prompts, sp, eos_token_id=None)
+
+    eng_bf16 = Engine(model, cfg, device=dev, seed=0)
+    eng_bf16.fused.nf4 = False
+    out_bf16 = eng_bf16.generate(prompts, sp, eos_token_id=None)
+
+    for a, b in zip(out_nf4, out_bf16):
+        agree = sum(x == y for x, y in zip(a[0], b[0]))
+        assert agree >= 7, (a[0], b[0])
