@@ -244,7 +244,21 @@ class Engine:
         pos32 = positions.to(torch.int32)
         N_batch = input_ids.shape[0]
 
-        def proj(x, lw, site, N_out, K_in, r_mult):
+        # per-step LoRA-u pool (zeroed ONCE per step; the split-K lora_u
+        # kernels accumulate into per-(layer, site) slices)
+        u_pool = None
+        site_off = {"qkv": (0, 3), "o": (3, 1), "gateup": (4, 2), "down": (6, 1)}
+        if nf4 and r > 0:
+            key = (N_batch, r)
+            if getattr(self, "_u_pool_key", None) != key:
+                self._u_pool = torch.zeros(
+                    len(lws), N_batch, 7 * r, dtype=torch.float32,
+                    device=input_ids.device)
+                self._u_pool_key = key
+            u_pool = self._u_pool
+            u_pool.zero_()
+
+        def proj(x, lw, li, site, N_out, K_in):
             """One fused projection: nf4 GEMM + LoRA (or merged bf16)."""
             if not nf4:
                 if site == "qkv":
@@ -252,9 +266,9 @@ class Engine:
                 return F.linear(x, getattr(lw, f"{site}_w"))
             bias = lw.qkv_b if site == "qkv" else None
             if r > 0:
-                rt = r * r_mult
-                u = torch.empty(N_batch, rt, dtype=torch.float32,
-                                device=x.device)
+                o0, mult = site_off[site]
+                rt = r * mult
+                u = u_pool[li, :, o0 * r:o0 * r + rt]
                 ksplit = max(1, min(8, 512 // max(1, ((N_batch + 15) // 16)
                                                  * (rt // 16))))
                 ext.lora_u(x, getattr(lw, f"{site}_afrag"), u, rt, ksplit)
@@ -269,7 +283,7 @@ class Engine:
         res = m.model.embed_tokens(input_ids)
         h = ext.rmsnorm_fwd(res, lws[0].in_norm, eps)
         for li, lw in enumerate(lws):
-            qkv = proj(h, lw, "qkv", qs + 2 * kvs, s.hidden_size, 3)
+            qkv = proj(h, lw, li, "qkv", qs + 2 * kvs, s.hidden_size)
             ext.rope_scatter_qkv(qkv, pos32, slot_mapping, self._inv_freq,
                                  self.pool.key[li], self.pool.value[li],
                                  s.num_heads, s.num_kv_heads, s.head_dim)
@@ -277,12 +291,12 @@ class Engine:
                 qkv, s.num_heads, s.head_dim, qs + 2 * kvs,
                 self.pool.key[li], self.pool.value[li], block_tables,
                 context_lens, self.scale)
-            o = proj(attn.view(-1, qs), lw, "o", s.hidden_size, qs, 1)
+            o = proj(attn.view(-1, qs), lw, li, "o", s.hidden_size, qs)
             h, res = ext.add_rmsnorm_fwd(res, o, lw.post_norm, eps)
-            gu = proj(h, lw, "gateup", 2 * s.intermediate_size,
-                      s.hidden_size, 2)
+            gu = proj(h, lw, li, "gateup", 2 * s.intermediate_size,
+                      s.hidden_size)
             act = ext.silu_mul_packed(gu)
-            d = proj(act, lw, "down", s.hidden_size, s.intermediate_size, 1)
+            d = proj(act, lw, li, "down", s.hidden_size, s.intermediate_size)
             next_w = (lws[li + 1].in_norm if li + 1 < len(lws)
                       else m.model.norm.weight)
             h, res = ext.add_rmsnorm_fwd(res, d, next_w, eps)

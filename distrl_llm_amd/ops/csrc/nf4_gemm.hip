@@ -34,7 +34,11 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 namespace {
 
-__constant__ float NF4_LUT[16] = {
+// nf4 codebook. NOTE: indexed per-lane with a divergent nibble, so it must
+// live in LDS at run time — a divergent index into __constant__/global
+// memory scalarizes into a waterfall on CDNA (measured 8x whole-kernel
+// slowdown). Each kernel copies it to LDS once.
+__device__ const float NF4_LUT[16] = {
     -1.0f, -0.6961928009986877f, -0.5250730514526367f, -0.39491748809814453f,
     -0.28444138169288635f, -0.18477343022823334f, -0.09105003625154495f, 0.0f,
     0.07958029955625534f, 0.16093020141124725f, 0.24611230194568634f,
@@ -54,7 +58,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
                      const float* __restrict__ u,            // (M, r) | null
                      const uint32_t* __restrict__ bfrag,     // B frag | null
                      __hip_bfloat16* __restrict__ y,         // (M, N)
-                     int M, int N, int K, int r) {
+                     int M, int N, int K, int r, int u_stride) {
   constexpr int BM = 16 * MT;
   const int tid = threadIdx.x;
   const int l = tid & 63;
@@ -69,8 +73,10 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   const int ksteps = K / 32;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* x_lds = smem;                         // BM * 128 bytes (64 bf16/row)
-  char* u_lds = smem + BM * 128;              // BM * r * 2 bytes
+  float* lut = reinterpret_cast<float*>(smem);        // 16 floats
+  char* x_lds = smem + 64;                    // BM * 128 bytes (64 bf16/row)
+  char* u_lds = x_lds + BM * 128;             // BM * r * 2 bytes
+  if (tid < 16) lut[tid] = NF4_LUT[tid];
 
   f32x4 acc[MT][4];
   #pragma unroll
@@ -85,7 +91,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
       const int unit = i % (r / 8);
       bf16x8 s;
       if (mbase + row < M) {
-        const float* up = u + (int64_t)(mbase + row) * r + unit * 8;
+        const float* up = u + (int64_t)(mbase + row) * u_stride + unit * 8;
         #pragma unroll
         for (int j = 0; j < 8; ++j) s.v[j] = f2bf(up[j]);
       } else {
@@ -137,7 +143,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
         bf16v8 bfr;
         #pragma unroll
         for (int j = 0; j < 8; ++j)
-          bfr[j] = (__bf16)(NF4_LUT[(wbits >> (4 * j)) & 0xF] * am);
+          bfr[j] = (__bf16)(lut[(wbits >> (4 * j)) & 0xF] * am);
         #pragma unroll
         for (int mt = 0; mt < MT; ++mt)
           acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -196,7 +202,7 @@ __global__ __launch_bounds__(64)
 void lora_u_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
                    const uint32_t* __restrict__ afrag,     // (r/16, K/32, 64, 4)
                    float* __restrict__ u,                  // (M, r) zeroed
-                   int M, int K, int r, int ksplit) {
+                   int M, int K, int r, int ksplit, int u_stride) {
   const int l = threadIdx.x;
   const int lrow = l & 15;
   const int lk = l >> 4;
@@ -228,7 +234,7 @@ void lora_u_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   for (int rr = 0; rr < 4; ++rr) {
     const int mrow = mt * 16 + lk * 4 + rr;
     if (mrow < M)
-      atomicAdd(&u[(int64_t)mrow * r + rt * 16 + lrow], acc[rr]);
+      atomicAdd(&u[(int64_t)mrow * u_stride + rt * 16 + lrow], acc[rr]);
   }
 }
 
@@ -268,9 +274,11 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   if (M == 0) return y;
 
   const bool has_lora = u.has_value() && r > 0;
+  int u_stride = 0;
   if (has_lora) {
     TORCH_CHECK(r % 32 == 0 && bfrag.has_value());
-    TORCH_CHECK(u->scalar_type() == at::kFloat && u->is_contiguous());
+    TORCH_CHECK(u->scalar_type() == at::kFloat && u->stride(1) == 1);
+    u_stride = u->stride(0);
   }
 
   // pick MT: largest tile with a reasonably filled grid
@@ -278,7 +286,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   while (mt > 1 && (N / 256) * ((M + 16 * mt - 1) / (16 * mt)) < 160) --mt;
   const int BM = 16 * mt;
   dim3 grid((M + BM - 1) / BM, N / 256), block(256);
-  size_t smem = (size_t)BM * 128 + (has_lora ? (size_t)BM * r * 2 : 0);
+  size_t smem = 64 + (size_t)BM * 128 + (has_lora ? (size_t)BM * r * 2 : 0);
 
   auto stream = at::cuda::getCurrentCUDAStream();
   const __hip_bfloat16* bias_p = nullptr;
@@ -294,7 +302,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
         reinterpret_cast<const uint32_t*>(w4f.data_ptr()), \
         amaxf.data_ptr<float>(), bias_p, u_p, bf_p, \
         reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), \
-        M, (int)N, (int)K, (int)r)
+        M, (int)N, (int)K, (int)r, u_stride)
   switch (mt) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -311,16 +319,17 @@ void lora_u(torch::Tensor x, torch::Tensor afrag, torch::Tensor u,
             int64_t r, int64_t ksplit) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
               x.scalar_type() == at::kBFloat16);
-  TORCH_CHECK(u.scalar_type() == at::kFloat && u.is_contiguous());
+  TORCH_CHECK(u.scalar_type() == at::kFloat && u.stride(1) == 1);
   const int M = x.size(0), K = x.size(1);
   TORCH_CHECK(r % 16 == 0 && K % 32 == 0);
-  u.zero_();
+  // caller zeroes u (batched across all sites/layers per step)
   dim3 grid((M + 15) / 16, r / 16, ksplit), block(64);
   hipLaunchKernelGGL(lora_u_kernel, grid, block, 0,
                      at::cuda::getCurrentCUDAStream(),
                      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
                      reinterpret_cast<const uint32_t*>(afrag.data_ptr()),
-                     u.data_ptr<float>(), M, K, (int)r, (int)ksplit);
+                     u.data_ptr<float>(), M, K, (int)r, (int)ksplit,
+                     (int)u.stride(0));
   HIP_CHECK_LAST();
 }
 

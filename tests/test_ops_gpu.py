@@ -300,7 +300,7 @@ def test_lora_u(ext):
     x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
     A = (torch.randn(r, K, device=dev) * 0.05).to(torch.bfloat16)
     afrag = prepack_bf16_fragments(A)
-    u = torch.empty(M, r, device=dev, dtype=torch.float32)
+    u = torch.zeros(M, r, device=dev, dtype=torch.float32)
     ext.lora_u(x, afrag, u, r, 4)
     ref = x.float() @ A.float().t()
     torch.testing.assert_close(u, ref, rtol=2e-2, atol=5e-1)
