@@ -247,12 +247,17 @@ class Engine:
         # per-step LoRA-u pool (zeroed ONCE per step; the split-K lora_u
         # kernels accumulate into per-(layer, site) slices)
         u_pool = None
-        site_off = {"qkv": (0, 3), "o": (3, 1), "gateup": (4, 2), "down": (6, 1)}
+        site_off = {}
         if nf4 and r > 0:
-            key = (N_batch, r)
+            off = 0
+            for site in ("qkv", "o", "gateup", "down"):
+                rp = getattr(lws[0], f"{site}_r")
+                site_off[site] = (off, rp)
+                off += rp
+            key = (N_batch, off)
             if getattr(self, "_u_pool_key", None) != key:
                 self._u_pool = torch.zeros(
-                    len(lws), N_batch, 7 * r, dtype=torch.float32,
+                    len(lws), N_batch, off, dtype=torch.float32,
                     device=input_ids.device)
                 self._u_pool_key = key
             u_pool = self._u_pool
@@ -266,9 +271,8 @@ class Engine:
                 return F.linear(x, getattr(lw, f"{site}_w"))
             bias = lw.qkv_b if site == "qkv" else None
             if r > 0:
-                o0, mult = site_off[site]
-                rt = r * mult
-                u = u_pool[li, :, o0 * r:o0 * r + rt]
+                o0, rt = site_off[site]
+                u = u_pool[li, :, o0:o0 + rt]
                 ksplit = max(1, min(8, 512 // max(1, ((N_batch + 15) // 16)
                                                  * (rt // 16))))
                 ext.lora_u(x, getattr(lw, f"{site}_afrag"), u, rt, ksplit)

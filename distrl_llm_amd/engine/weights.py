@@ -28,7 +28,8 @@ class LayerWeights:
                  "qkv_w4", "qkv_amax", "o_w4", "o_amax",
                  "gateup_w4", "gateup_amax", "down_w4", "down_amax",
                  "qkv_afrag", "qkv_bfrag", "o_afrag", "o_bfrag",
-                 "gateup_afrag", "gateup_bfrag", "down_afrag", "down_bfrag")
+                 "gateup_afrag", "gateup_bfrag", "down_afrag", "down_bfrag",
+                 "qkv_r", "o_r", "gateup_r", "down_r")
 
 
 def _effective(mod) -> torch.Tensor:
@@ -91,8 +92,11 @@ class FusedWeights:
                            ("down", [mlp.down_proj])):
             A = torch.cat([m.lora_A.detach() for m in mods], dim=0)  # (r_tot, K)
             r_tot = A.shape[0]
+            r_pad = (r_tot + 31) // 32 * 32  # kernel needs rank % 32 == 0
             N = sum(m.out_features for m in mods)
-            B = A.new_zeros(N, r_tot)
+            if r_pad != r_tot:
+                A = torch.cat([A, A.new_zeros(r_pad - r_tot, A.shape[1])], 0)
+            B = A.new_zeros(N, r_pad)
             n0, r0 = 0, 0
             for m in mods:
                 B[n0:n0 + m.out_features, r0:r0 + m.r] = (
@@ -103,6 +107,7 @@ class FusedWeights:
                     prepack_bf16_fragments(A.to(torch.bfloat16)))
             setattr(lw, f"{name}_bfrag",
                     prepack_bf16_fragments(B.to(torch.bfloat16)))
+            setattr(lw, f"{name}_r", r_pad)
 
     @torch.no_grad()
     def refresh(self):

@@ -73,10 +73,13 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   const int ksteps = K / 32;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* lut = reinterpret_cast<float*>(smem);        // 16 floats
-  char* x_lds = smem + 64;                    // BM * 128 bytes (64 bf16/row)
+  // 256-entry paired LUT: entry b = (code[b&15], code[b>>4]) — one
+  // ds_read_b64 + one v_pk_mul dequantizes a whole packed BYTE
+  float2* lut2 = reinterpret_cast<float2*>(smem);     // 256 * 8 B
+  char* x_lds = smem + 2048;                  // BM * 128 bytes (64 bf16/row)
   char* u_lds = x_lds + BM * 128;             // BM * r * 2 bytes
-  if (tid < 16) lut[tid] = NF4_LUT[tid];
+  if (tid < 256)
+    lut2[tid] = make_float2(NF4_LUT[tid & 15], NF4_LUT[tid >> 4]);
 
   f32x4 acc[MT][4];
   #pragma unroll
@@ -102,8 +105,30 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
     }
   }
 
-  // ---- main K loop: 64-deep chunks ----
-  for (int kb = 0; kb < K / 64; ++kb) {
+  // ---- main K loop: 64-deep chunks with one-chunk-ahead weight
+  // prefetch. Without it the compiler issues each weight dword right
+  // before its use with an immediate vmcnt wait — every fragment pays
+  // full HBM latency serially (measured 8-20x slowdown).
+  const int nkb = K / 64;
+  uint32_t wb_cur[8], wb_nxt[8];
+  float am_cur[8], am_nxt[8];
+
+  #define LOAD_WCHUNK(KB, WB, AM)                                         \
+    _Pragma("unroll")                                                     \
+    for (int ks = 0; ks < 2; ++ks) {                                      \
+      _Pragma("unroll")                                                   \
+      for (int nt = 0; nt < 4; ++nt) {                                    \
+        const int kstep = (KB) * 2 + ks;                                  \
+        const int ntg = ntile0 + nt;                                      \
+        WB[ks * 4 + nt] = w4f[((int64_t)ntg * ksteps + kstep) * 64 + l];  \
+        AM[ks * 4 + nt] = amaxf[((int64_t)ntg * (K / 64)                  \
+                                 + (kstep >> 1)) * 16 + lrow];            \
+      }                                                                   \
+    }
+
+  LOAD_WCHUNK(0, wb_cur, am_cur);
+
+  for (int kb = 0; kb < nkb; ++kb) {
     __syncthreads();
     // stage x chunk (BM x 64 bf16), XOR-swizzled
     for (int i = tid; i < BM * 8; i += 256) {
@@ -120,11 +145,13 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
       const int off = (row * 128 + unit * 16) ^ ((row & 15) << 4);
       *reinterpret_cast<bf16x8*>(x_lds + off) = s;
     }
+    if (kb + 1 < nkb) {
+      LOAD_WCHUNK(kb + 1, wb_nxt, am_nxt);
+    }
     __syncthreads();
 
     #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      const int kstep = kb * 2 + ks;
       // A fragments for every m-tile
       bf16v8 afrag[MT];
       #pragma unroll
@@ -136,21 +163,28 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
       // 4 n-tiles: dequant B fragment + MFMA
       #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        const int ntg = ntile0 + nt;
-        const uint32_t wbits = w4f[((int64_t)ntg * ksteps + kstep) * 64 + l];
-        const float am = amaxf[((int64_t)ntg * (K / 64) + (kstep >> 1)) * 16
-                               + lrow];
+        const uint32_t wbits = wb_cur[ks * 4 + nt];
+        const float am = am_cur[ks * 4 + nt];
         bf16v8 bfr;
         #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          bfr[j] = (__bf16)(lut[(wbits >> (4 * j)) & 0xF] * am);
+        for (int b = 0; b < 4; ++b) {
+          const float2 pair = lut2[(wbits >> (8 * b)) & 0xFF];
+          bfr[2 * b] = (__bf16)(pair.x * am);
+          bfr[2 * b + 1] = (__bf16)(pair.y * am);
+        }
         #pragma unroll
         for (int mt = 0; mt < MT; ++mt)
           acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[mt], bfr, acc[mt][nt], 0, 0, 0);
       }
     }
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      wb_cur[i] = wb_nxt[i];
+      am_cur[i] = am_nxt[i];
+    }
   }
+  #undef LOAD_WCHUNK
 
   // ---- LoRA epilogue: one extra MFMA k-step per rank-32 block ----
   if (u != nullptr) {
@@ -286,7 +320,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   while (mt > 1 && (N / 256) * ((M + 16 * mt - 1) / (16 * mt)) < 160) --mt;
   const int BM = 16 * mt;
   dim3 grid((M + BM - 1) / BM, N / 256), block(256);
-  size_t smem = 64 + (size_t)BM * 128 + (has_lora ? (size_t)BM * r * 2 : 0);
+  size_t smem = 2048 + (size_t)BM * 128 + (has_lora ? (size_t)BM * r * 2 : 0);
 
   auto stream = at::cuda::getCurrentCUDAStream();
   const __hip_bfloat16* bias_p = nullptr;
