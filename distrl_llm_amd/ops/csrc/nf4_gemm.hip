@@ -128,60 +128,84 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
 
   LOAD_WCHUNK(0, wb_cur, am_cur);
 
-  for (int kb = 0; kb < nkb; ++kb) {
-    __syncthreads();
-    // stage x chunk (BM x 64 bf16), XOR-swizzled
-    for (int i = tid; i < BM * 8; i += 256) {
-      const int row = i >> 3;
-      const int unit = i & 7;
-      bf16x8 s;
-      if (mbase + row < M) {
-        s = *reinterpret_cast<const bf16x8*>(
-            x + (int64_t)(mbase + row) * K + kb * 64 + unit * 8);
-      } else {
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) s.v[j] = f2bf(0.f);
+  // x is staged in SUPER-panels of SK columns: one barrier pair per
+  // SK/64 weight chunks, so the chunk loop in between runs barrier-free
+  // and the one-chunk-ahead weight prefetch is never drained (hipcc puts
+  // a vmcnt(0) on any in-loop ds_write/barrier path — guide §5 traps).
+  constexpr int SK = (MT <= 2) ? 512 : 256;   // host asserts K % SK == 0
+  constexpr int SKU = SK / 8;                 // bf16x8 units per row
+  constexpr int XIT = (BM * SKU + 255) / 256; // staging iters (VGPR-resident)
+
+  for (int sk = 0; sk < K; sk += SK) {
+    // issue-early x loads for the whole super-panel
+    bf16x8 xs[XIT];
+    #pragma unroll
+    for (int it = 0; it < XIT; ++it) {
+      const int i = tid + it * 256;
+      if (i < BM * SKU) {
+        const int row = i / SKU;
+        if (mbase + row < M) {
+          xs[it] = *reinterpret_cast<const bf16x8*>(
+              x + (int64_t)(mbase + row) * K + sk + (i % SKU) * 8);
+        } else {
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) xs[it].v[j] = f2bf(0.f);
+        }
       }
-      const int off = (row * 128 + unit * 16) ^ ((row & 15) << 4);
-      *reinterpret_cast<bf16x8*>(x_lds + off) = s;
     }
-    if (kb + 1 < nkb) {
-      LOAD_WCHUNK(kb + 1, wb_nxt, am_nxt);
+    __syncthreads();  // previous panel fully consumed
+    #pragma unroll
+    for (int it = 0; it < XIT; ++it) {
+      const int i = tid + it * 256;
+      if (i < BM * SKU) {
+        const int row = i / SKU;
+        const int off = (row * (SK * 2) + (i % SKU) * 16)
+                        ^ ((row & 15) << 4);
+        *reinterpret_cast<bf16x8*>(x_lds + off) = xs[it];
+      }
     }
     __syncthreads();
 
-    #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      // A fragments for every m-tile
-      bf16v8 afrag[MT];
-      #pragma unroll
-      for (int mt = 0; mt < MT; ++mt) {
-        const int row = mt * 16 + lrow;
-        const int off = (row * 128 + (ks * 4 + lk) * 16) ^ ((row & 15) << 4);
-        afrag[mt] = lds_read_frag(x_lds, off);
+    for (int kc = 0; kc < SK / 64; ++kc) {
+      const int kb = (sk + kc * 64) / 64;
+      if (kb + 1 < nkb) {
+        LOAD_WCHUNK(kb + 1, wb_nxt, am_nxt);
       }
-      // 4 n-tiles: dequant B fragment + MFMA
       #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-        const uint32_t wbits = wb_cur[ks * 4 + nt];
-        const float am = am_cur[ks * 4 + nt];
-        bf16v8 bfr;
+      for (int ks = 0; ks < 2; ++ks) {
+        // A fragments for every m-tile
+        bf16v8 afrag[MT];
         #pragma unroll
-        for (int b = 0; b < 4; ++b) {
-          const float2 pair = lut2[(wbits >> (8 * b)) & 0xFF];
-          bfr[2 * b] = (__bf16)(pair.x * am);
-          bfr[2 * b + 1] = (__bf16)(pair.y * am);
+        for (int mt = 0; mt < MT; ++mt) {
+          const int row = mt * 16 + lrow;
+          const int off = (row * (SK * 2)
+                           + (kc * 8 + ks * 4 + lk) * 16)
+                          ^ ((row & 15) << 4);
+          afrag[mt] = lds_read_frag(x_lds, off);
         }
+        // 4 n-tiles: dequant B fragment + MFMA
         #pragma unroll
-        for (int mt = 0; mt < MT; ++mt)
-          acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag[mt], bfr, acc[mt][nt], 0, 0, 0);
+        for (int nt = 0; nt < 4; ++nt) {
+          const uint32_t wbits = wb_cur[ks * 4 + nt];
+          const float am = am_cur[ks * 4 + nt];
+          bf16v8 bfr;
+          #pragma unroll
+          for (int b = 0; b < 4; ++b) {
+            const float2 pair = lut2[(wbits >> (8 * b)) & 0xFF];
+            bfr[2 * b] = (__bf16)(pair.x * am);
+            bfr[2 * b + 1] = (__bf16)(pair.y * am);
+          }
+          #pragma unroll
+          for (int mt = 0; mt < MT; ++mt)
+            acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[mt], bfr, acc[mt][nt], 0, 0, 0);
+        }
       }
-    }
-    #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      wb_cur[i] = wb_nxt[i];
-      am_cur[i] = am_nxt[i];
+      #pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        wb_cur[i] = wb_nxt[i];
+        am_cur[i] = am_nxt[i];
+      }
     }
   }
   #undef LOAD_WCHUNK
@@ -319,8 +343,11 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   int mt = std::min<int>((M + 15) / 16, 5);
   while (mt > 1 && (N / 256) * ((M + 16 * mt - 1) / (16 * mt)) < 160) --mt;
   const int BM = 16 * mt;
+  const int SK = (mt <= 2) ? 512 : 256;  // must mirror the kernel constexpr
+  TORCH_CHECK(K % SK == 0, "nf4_gemm: K (", K, ") % ", SK, " != 0");
   dim3 grid((M + BM - 1) / BM, N / 256), block(256);
-  size_t smem = 2048 + (size_t)BM * 128 + (has_lora ? (size_t)BM * r * 2 : 0);
+  size_t smem = 2048 + (size_t)BM * SK * 2
+                + (has_lora ? (size_t)BM * r * 2 : 0);
 
   auto stream = at::cuda::getCurrentCUDAStream();
   const __hip_bfloat16* bias_p = nullptr;
