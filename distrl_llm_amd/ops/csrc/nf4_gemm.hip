@@ -75,9 +75,10 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // 256-entry paired LUT: entry b = (code[b&15], code[b>>4]) — one
   // ds_read_b64 + one v_pk_mul dequantizes a whole packed BYTE
+  constexpr int SK_ = (MT <= 2) ? 512 : 256;  // x super-panel columns
   float2* lut2 = reinterpret_cast<float2*>(smem);     // 256 * 8 B
-  char* x_lds = smem + 2048;                  // BM * 128 bytes (64 bf16/row)
-  char* u_lds = x_lds + BM * 128;             // BM * r * 2 bytes
+  char* x_lds = smem + 2048;                  // BM * SK * 2 bytes
+  char* u_lds = x_lds + BM * SK_ * 2;         // BM * r * 2 bytes
   if (tid < 256)
     lut2[tid] = make_float2(NF4_LUT[tid & 15], NF4_LUT[tid >> 4]);
 
@@ -132,7 +133,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   // SK/64 weight chunks, so the chunk loop in between runs barrier-free
   // and the one-chunk-ahead weight prefetch is never drained (hipcc puts
   // a vmcnt(0) on any in-loop ds_write/barrier path — guide §5 traps).
-  constexpr int SK = (MT <= 2) ? 512 : 256;   // host asserts K % SK == 0
+  constexpr int SK = SK_;                     // host asserts K % SK == 0
   constexpr int SKU = SK / 8;                 // bf16x8 units per row
   constexpr int XIT = (BM * SKU + 255) / 256; // staging iters (VGPR-resident)
 
