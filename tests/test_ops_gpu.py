@@ -259,7 +259,7 @@ def test_mfma_mapping_probe(ext):
 
 @pytest.mark.parametrize("M,N,K,r,bias", [
     (160, 512, 512, 0, False),
-    (7, 256, 64, 0, True),
+    (7, 256, 512, 0, True),
     (160, 4608, 3584, 96, True),
     (33, 2048, 1024, 64, False),
 ])
