@@ -112,8 +112,8 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   // before its use with an immediate vmcnt wait — every fragment pays
   // full HBM latency serially (measured 8-20x slowdown).
   const int nkb = K / 64;
-  uint32_t wb_cur[8], wb_nxt[8];
-  float am_cur[8], am_nxt[8];
+  uint32_t wb_cur[8], wb_nxt[8], wb_nx2[8];
+  float am_cur[8], am_nxt[8], am_nx2[8];
 
   #define LOAD_WCHUNK(KB, WB, AM)                                         \
     _Pragma("unroll")                                                     \
@@ -129,6 +129,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
     }
 
   LOAD_WCHUNK(0, wb_cur, am_cur);
+  if (nkb > 1) LOAD_WCHUNK(1, wb_nxt, am_nxt);
 
   // x is staged in SUPER-panels of SK columns: one barrier pair per
   // SK/64 weight chunks, so the chunk loop in between runs barrier-free
@@ -170,8 +171,8 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
 
     for (int kc = 0; kc < SK / 64; ++kc) {
       const int kb = (sk + kc * 64) / 64;
-      if (kb + 1 < nkb) {
-        LOAD_WCHUNK(kb + 1, wb_nxt, am_nxt);
+      if (kb + 2 < nkb) {
+        LOAD_WCHUNK(kb + 2, wb_nx2, am_nx2);
       }
       #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
@@ -207,6 +208,8 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
       for (int i = 0; i < 8; ++i) {
         wb_cur[i] = wb_nxt[i];
         am_cur[i] = am_nxt[i];
+        wb_nxt[i] = wb_nx2[i];
+        am_nxt[i] = am_nx2[i];
       }
     }
   }
@@ -276,8 +279,9 @@ void lora_u_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
 
   const int m = mt * 16 + lrow;
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int kstep = k0; kstep < k1; ++kstep) {
-    bf16v8 af;
+  // software-pipelined 2-deep: issue next k-steps' fragments while the
+  // current MFMA runs (the bare loop serializes on load latency)
+  auto load_af = [&](int kstep, bf16v8& af, uint4& bw) {
     if (m < M) {
       af = *reinterpret_cast<const bf16v8*>(
           x + (int64_t)m * K + kstep * 32 + lk * 8);
@@ -285,10 +289,21 @@ void lora_u_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
       #pragma unroll
       for (int j = 0; j < 8; ++j) af[j] = (__bf16)0.f;
     }
-    const uint4 bw = *reinterpret_cast<const uint4*>(
+    bw = *reinterpret_cast<const uint4*>(
         afrag + ((int64_t)rt * ksteps + kstep) * 64 * 4 + l * 4);
-    const bf16v8 bf = *reinterpret_cast<const bf16v8*>(&bw);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+  };
+  bf16v8 af_c, af_n;
+  uint4 bw_c, bw_n;
+  if (k0 < k1) load_af(k0, af_c, bw_c);
+  if (k0 + 1 < k1) load_af(k0 + 1, af_n, bw_n);
+  for (int kstep = k0; kstep < k1; ++kstep) {
+    bf16v8 af_f;
+    uint4 bw_f;
+    if (kstep + 2 < k1) load_af(kstep + 2, af_f, bw_f);
+    const bf16v8 bf = *reinterpret_cast<const bf16v8*>(&bw_c);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af_c, bf, acc, 0, 0, 0);
+    af_c = af_n; bw_c = bw_n;
+    af_n = af_f; bw_n = bw_f;
   }
   #pragma unroll
   for (int rr = 0; rr < 4; ++rr) {
