@@ -58,7 +58,9 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
                      const float* __restrict__ u,            // (M, r) | null
                      const uint32_t* __restrict__ bfrag,     // B frag | null
                      __hip_bfloat16* __restrict__ y,         // (M, N)
-                     int M, int N, int K, int r, int u_stride) {
+                     float* __restrict__ ws,                 // split-K | null
+                     int M, int N, int K, int r, int u_stride,
+                     int nsp_per) {
   constexpr int BM = 16 * MT;
   const int tid = threadIdx.x;
   const int l = tid & 63;
@@ -70,6 +72,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   // m dimension is tiled (T1 locality without an explicit remap)
   const int mbase = blockIdx.x * BM;
   const int ntile0 = blockIdx.y * 16 + wave * 4;  // this wave's 4 n-tiles
+  const int zid = blockIdx.z;                     // split-K slice
   const int ksteps = K / 32;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -127,8 +130,15 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
       }                                                                   \
     }
 
-  LOAD_WCHUNK(0, wb_cur, am_cur);
-  if (nkb > 1) LOAD_WCHUNK(1, wb_nxt, am_nxt);
+  // this slice's super-panel range (split-K over whole panels)
+  constexpr int SKC = ((MT <= 2) ? 512 : 256) / 64;  // chunks per panel
+  const int npanels = K / ((MT <= 2) ? 512 : 256);
+  const int sp0 = zid * nsp_per;
+  const int sp1 = min(npanels, sp0 + nsp_per);
+  const int kb0 = sp0 * SKC;
+  const int kb_end = sp1 * SKC;
+  if (kb0 < kb_end) LOAD_WCHUNK(kb0, wb_cur, am_cur);
+  if (kb0 + 1 < kb_end) LOAD_WCHUNK(kb0 + 1, wb_nxt, am_nxt);
 
   // x is staged in SUPER-panels of SK columns: one barrier pair per
   // SK/64 weight chunks, so the chunk loop in between runs barrier-free
@@ -138,7 +148,8 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   constexpr int SKU = SK / 8;                 // bf16x8 units per row
   constexpr int XIT = (BM * SKU + 255) / 256; // staging iters (VGPR-resident)
 
-  for (int sk = 0; sk < K; sk += SK) {
+  for (int sp = sp0; sp < sp1; ++sp) {
+    const int sk = sp * SK;
     // issue-early x loads for the whole super-panel
     bf16x8 xs[XIT];
     #pragma unroll
@@ -170,7 +181,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
 
     for (int kc = 0; kc < SK / 64; ++kc) {
       const int kb = (sk + kc * 64) / 64;
-      if (kb + 2 < nkb) {
+      if (kb + 2 < kb_end) {
         LOAD_WCHUNK(kb + 2, wb_nx2, am_nx2);
       }
       #pragma unroll
@@ -215,7 +226,8 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   #undef LOAD_WCHUNK
 
   // ---- LoRA epilogue: one extra MFMA k-step per rank-32 block ----
-  if (u != nullptr) {
+  // (slice 0 only under split-K; slices contribute additively)
+  if (u != nullptr && zid == 0) {
     __syncthreads();
     for (int rs = 0; rs < r / 32; ++rs) {
       bf16v8 afrag[MT];
@@ -239,21 +251,50 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
     }
   }
 
-  // ---- epilogue: bias + store ----
+  // ---- epilogue: bias + store (fp32 slab under split-K) ----
   #pragma unroll
   for (int nt = 0; nt < 4; ++nt) {
     const int n = (ntile0 + nt) * 16 + lrow;
-    const float bv = (bias != nullptr) ? bf2f(bias[n]) : 0.f;
+    const float bv = (bias != nullptr && zid == 0) ? bf2f(bias[n]) : 0.f;
     #pragma unroll
     for (int mt = 0; mt < MT; ++mt) {
       #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         const int m = mbase + mt * 16 + lk * 4 + rr;
-        if (m < M)
-          y[(int64_t)m * N + n] = f2bf(acc[mt][nt][rr] + bv);
+        if (m < M) {
+          if (ws != nullptr)
+            ws[((int64_t)zid * M + m) * N + n] = acc[mt][nt][rr] + bv;
+          else
+            y[(int64_t)m * N + n] = f2bf(acc[mt][nt][rr] + bv);
+        }
       }
     }
   }
+}
+
+// split-K combine: y = bf16(sum_z ws[z])
+__global__ __launch_bounds__(256)
+void splitk_reduce_kernel(const float* __restrict__ ws,
+                          __hip_bfloat16* __restrict__ y,
+                          int64_t mn, int ksplit) {
+  const int64_t i0 = ((int64_t)blockIdx.x * 256 + threadIdx.x) * 4;
+  if (i0 + 3 >= mn) {
+    for (int64_t i = i0; i < mn; ++i) {
+      float a = 0.f;
+      for (int z = 0; z < ksplit; ++z) a += ws[(int64_t)z * mn + i];
+      y[i] = f2bf(a);
+    }
+    return;
+  }
+  float4 a = *reinterpret_cast<const float4*>(ws + i0);
+  for (int z = 1; z < ksplit; ++z) {
+    const float4 b = *reinterpret_cast<const float4*>(ws + (int64_t)z * mn + i0);
+    a.x += b.x; a.y += b.y; a.z += b.z; a.w += b.w;
+  }
+  bf16x8* dummy;
+  __hip_bfloat16 o[4] = {f2bf(a.x), f2bf(a.y), f2bf(a.z), f2bf(a.w)};
+  *reinterpret_cast<uint2*>(y + i0) = *reinterpret_cast<uint2*>(o);
+  (void)dummy;
 }
 
 // ---- split-K LoRA A kernel: u[M,r] = x[M,K] @ A^T, fp32 atomic combine.
@@ -361,9 +402,24 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   const int BM = 16 * mt;
   const int SK = (mt <= 2) ? 512 : 256;  // must mirror the kernel constexpr
   TORCH_CHECK(K % SK == 0, "nf4_gemm: K (", K, ") % ", SK, " != 0");
-  dim3 grid((M + BM - 1) / BM, N / 256), block(256);
+  // split-K over super-panels until the grid fills the chip (the skinny
+  // decode shapes otherwise run at ~1 wave/SIMD, fully latency-exposed)
+  const int npanels = (int)K / SK;
+  const int base_blocks = ((M + BM - 1) / BM) * ((int)N / 256);
+  int ksplit = 1;
+  while (ksplit * 2 <= npanels && base_blocks * ksplit < 512 && ksplit < 8)
+    ksplit *= 2;
+  const int nsp_per = (npanels + ksplit - 1) / ksplit;
+  ksplit = (npanels + nsp_per - 1) / nsp_per;
+  dim3 grid((M + BM - 1) / BM, N / 256, ksplit), block(256);
   size_t smem = 2048 + (size_t)BM * SK * 2
                 + (has_lora ? (size_t)BM * r * 2 : 0);
+  torch::Tensor ws;
+  float* ws_p = nullptr;
+  if (ksplit > 1) {
+    ws = torch::empty({ksplit, (int64_t)M, N}, x.options().dtype(at::kFloat));
+    ws_p = ws.data_ptr<float>();
+  }
 
   auto stream = at::cuda::getCurrentCUDAStream();
   const __hip_bfloat16* bias_p = nullptr;
@@ -378,8 +434,8 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
         reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()), \
         reinterpret_cast<const uint32_t*>(w4f.data_ptr()), \
         amaxf.data_ptr<float>(), bias_p, u_p, bf_p, \
-        reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), \
-        M, (int)N, (int)K, (int)r, u_stride)
+        reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), ws_p, \
+        M, (int)N, (int)K, (int)r, u_stride, nsp_per)
   switch (mt) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -388,6 +444,13 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
     default: LAUNCH(5); break;
   }
   #undef LAUNCH
+  if (ksplit > 1) {
+    const int64_t mn = (int64_t)M * N;
+    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(CDIV(mn, 1024)), dim3(256),
+                       0, stream, ws_p,
+                       reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                       mn, ksplit);
+  }
   HIP_CHECK_LAST();
   return y;
 }
