@@ -50,7 +50,7 @@ DEV_INLINE bf16v8 lds_read_frag(const char* base, int byte_off) {
   return *reinterpret_cast<const bf16v8*>(base + byte_off);
 }
 
-template <int MT>
+template <int MT, int DBG = 0>  // DBG: 1 = skip LUT dequant, 2 = skip x LDS
 __global__ __launch_bounds__(256)
 void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
                      const uint32_t* __restrict__ w4f,       // frag-packed
@@ -85,6 +85,8 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   char* u_lds = x_lds + BM * SK_ * 2;         // BM * r * 2 bytes
   if (tid < 256)
     lut2[tid] = make_float2(NF4_LUT[tid & 15], NF4_LUT[tid >> 4]);
+  // per-lane register copy of the codebook for the shuffle-LUT dequant
+  const float lut_reg = NF4_LUT[tid & 15];
 
   f32x4 acc[MT][4];
   #pragma unroll
@@ -115,8 +117,8 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   // before its use with an immediate vmcnt wait — every fragment pays
   // full HBM latency serially (measured 8-20x slowdown).
   const int nkb = K / 64;
-  uint32_t wb_cur[8], wb_nxt[8], wb_nx2[8];
-  float am_cur[8], am_nxt[8], am_nx2[8];
+  uint32_t wb_cur[8] = {}, wb_nxt[8] = {}, wb_nx2[8] = {};
+  float am_cur[8] = {}, am_nxt[8] = {}, am_nx2[8] = {};
 
   #define LOAD_WCHUNK(KB, WB, AM)                                         \
     _Pragma("unroll")                                                     \
@@ -191,11 +193,16 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
         bf16v8 afrag[MT];
         #pragma unroll
         for (int mt = 0; mt < MT; ++mt) {
-          const int row = mt * 16 + lrow;
-          const int off = (row * (SK * 2)
-                           + (kc * 8 + ks * 4 + lk) * 16)
-                          ^ ((row & 15) << 4);
-          afrag[mt] = lds_read_frag(x_lds, off);
+          if constexpr (DBG == 2) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) afrag[mt][j] = (__bf16)1.0f;
+          } else {
+            const int row = mt * 16 + lrow;
+            const int off = (row * (SK * 2)
+                             + (kc * 8 + ks * 4 + lk) * 16)
+                            ^ ((row & 15) << 4);
+            afrag[mt] = lds_read_frag(x_lds, off);
+          }
         }
         // 4 n-tiles: dequant B fragment + MFMA
         #pragma unroll
@@ -203,11 +210,19 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
           const uint32_t wbits = wb_cur[ks * 4 + nt];
           const float am = am_cur[ks * 4 + nt];
           bf16v8 bfr;
-          #pragma unroll
-          for (int b = 0; b < 4; ++b) {
-            const float2 pair = lut2[(wbits >> (8 * b)) & 0xFF];
-            bfr[2 * b] = (__bf16)(pair.x * am);
-            bfr[2 * b + 1] = (__bf16)(pair.y * am);
+          if constexpr (DBG == 1) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+              bfr[j] = (__bf16)(am + (float)(wbits & 1));
+          } else {
+            // register LUT via cross-lane shuffle (ds_bpermute):
+            // lanes 0-15 hold the 16 nf4 codes; no LDS traffic, no
+            // divergent-LDS hazards
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              const int nib = (wbits >> (4 * j)) & 0xF;
+              bfr[j] = (__bf16)(__shfl(lut_reg, nib, WAVE) * am);
+            }
           }
           #pragma unroll
           for (int mt = 0; mt < MT; ++mt)
@@ -397,9 +412,8 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
     u_stride = u->stride(0);
   }
 
-  // pick MT: largest tile with a reasonably filled grid
+  // largest m-tile (short per-block K chains; occupancy comes from split-K)
   int mt = std::min<int>((M + 15) / 16, 5);
-  while (mt > 1 && (N / 256) * ((M + 16 * mt - 1) / (16 * mt)) < 160) --mt;
   const int BM = 16 * mt;
   const int SK = (mt <= 2) ? 512 : 256;  // must mirror the kernel constexpr
   TORCH_CHECK(K % SK == 0, "nf4_gemm: K (", K, ") % ", SK, " != 0");
@@ -408,17 +422,21 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   const int npanels = (int)K / SK;
   const int base_blocks = ((M + BM - 1) / BM) * ((int)N / 256);
   int ksplit = 1;
-  while (ksplit * 2 <= npanels && base_blocks * ksplit < 512 && ksplit < 8)
+  while (ksplit * 2 <= npanels && base_blocks * ksplit < 512 && ksplit < 16)
     ksplit *= 2;
+  if (const char* e = getenv("DISTRL_NF4_KSPLIT")) ksplit = atoi(e);
+  ksplit = std::max(1, std::min(ksplit, npanels));
   const int nsp_per = (npanels + ksplit - 1) / ksplit;
   ksplit = (npanels + nsp_per - 1) / nsp_per;
   dim3 grid((M + BM - 1) / BM, N / 256, ksplit), block(256);
   size_t smem = 2048 + (size_t)BM * SK * 2
                 + (has_lora ? (size_t)BM * r * 2 : 0);
+  if (getenv("DISTRL_NF4_LDSPAD")) smem += 8192;  // debug: OOB guard
   torch::Tensor ws;
   float* ws_p = nullptr;
   if (ksplit > 1) {
     ws = torch::empty({ksplit, (int64_t)M, N}, x.options().dtype(at::kFloat));
+    if (getenv("DISTRL_NF4_WSZERO")) ws.zero_();
     ws_p = ws.data_ptr<float>();
   }
 
@@ -430,8 +448,25 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   const uint32_t* bf_p = has_lora
       ? reinterpret_cast<const uint32_t*>(bfrag->data_ptr()) : nullptr;
 
+  int dbg = 0;
+  if (const char* e = getenv("DISTRL_NF4_DBG")) dbg = atoi(e);
+  if (dbg == 1 && mt == 1) {
+    hipLaunchKernelGGL((nf4_gemm_kernel<1, 1>), grid, block, smem, stream,
+        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+        reinterpret_cast<const uint32_t*>(w4f.data_ptr()),
+        amaxf.data_ptr<float>(), bias_p, u_p, bf_p,
+        reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), ws_p,
+        M, (int)N, (int)K, (int)r, u_stride, nsp_per);
+  } else if (dbg == 2 && mt == 1) {
+    hipLaunchKernelGGL((nf4_gemm_kernel<1, 2>), grid, block, smem, stream,
+        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+        reinterpret_cast<const uint32_t*>(w4f.data_ptr()),
+        amaxf.data_ptr<float>(), bias_p, u_p, bf_p,
+        reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), ws_p,
+        M, (int)N, (int)K, (int)r, u_stride, nsp_per);
+  } else
   #define LAUNCH(MTV) \
-    hipLaunchKernelGGL(nf4_gemm_kernel<MTV>, grid, block, smem, stream, \
+    hipLaunchKernelGGL((nf4_gemm_kernel<MTV>), grid, block, smem, stream, \
         reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()), \
         reinterpret_cast<const uint32_t*>(w4f.data_ptr()), \
         amaxf.data_ptr<float>(), bias_p, u_p, bf_p, \
@@ -445,6 +480,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
     default: LAUNCH(5); break;
   }
   #undef LAUNCH
+  if (ksplit > 1 && getenv("DISTRL_NF4_RETWS")) return ws;  // debug
   if (ksplit > 1) {
     const int64_t mn = (int64_t)M * N;
     hipLaunchKernelGGL(splitk_reduce_kernel, dim3(CDIV(mn, 1024)), dim3(256),
