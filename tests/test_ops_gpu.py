@@ -328,9 +328,14 @@ def test_engine_nf4_matches_merged():
     prompts = [list(range(1, 40)), [5, 7, 11, 13]]
     sp = SamplingParams(max_tokens=8, temperature=0.0, n=1)
 
-    eng_nf4 = Engine(model, cfg, device=dev, seed=0)
-    assert eng_nf4.fused.nf4, "nf4 path must be active for this spec"
-    out_nf4 = eng_nf4.generate(prompts, sp, eos_token_id=None)
+    import os
+    os.environ["DISTRL_DECODE_NF4"] = "1"
+    try:
+        eng_nf4 = Engine(model, cfg, device=dev, seed=0)
+        assert eng_nf4.fused.nf4, "nf4 path must be active for this spec"
+        out_nf4 = eng_nf4.generate(prompts, sp, eos_token_id=None)
+    finally:
+        del os.environ["DISTRL_DECODE_NF4"]
 
     eng_bf16 = Engine(model, cfg, device=dev, seed=0)
     eng_bf16.fused.nf4 = False
