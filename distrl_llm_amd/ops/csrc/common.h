@@ -56,7 +56,7 @@ DEV_INLINE float block_max(float x, float* red) {
   x = wave_max(x);
   if (lane == 0) red[wid] = x;
   __syncthreads();
-  float r = (threadIdx.x < nw) ? red[threadIdx.x] : -INFINITY;
+  float r = (threadIdx.x < nw) ? red[threadIdx.x] : -1e30f;
   if (wid == 0) {
     #pragma unroll
     for (int off = 32; off > 0; off >>= 1) r = fmaxf(r, __shfl_xor(r, off, WAVE));

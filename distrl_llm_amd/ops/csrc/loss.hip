@@ -28,7 +28,7 @@ void logprob_lse_kernel(const T* __restrict__ logits,
   const T* lr = logits + row * V;
   const int64_t tgt = targets[row];
 
-  float m = -INFINITY, s = 0.f;
+  float m = -1e30f, s = 0.f;
   if constexpr (std::is_same<T, __hip_bfloat16>::value) {
     const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
     const int nvec = V / 8;

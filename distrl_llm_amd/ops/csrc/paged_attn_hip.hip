@@ -34,17 +34,24 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
                          const int* __restrict__ block_tables,      // (N,max_nb)
                          const int* __restrict__ ctx_lens,          // (N,)
                          __hip_bfloat16* __restrict__ out,          // (N,H,D)
+                         float* __restrict__ o_part,  // (CS,N,H,D) | null
+                         float* __restrict__ ml_part, // (CS,N,H,2) | null
                          int H, int KV, int max_nb, int block_size,
-                         int Lpad, float scale, int64_t q_row_stride) {
+                         int Lpad, float scale, int64_t q_row_stride,
+                         int slice_len) {
   const int seq = blockIdx.x;
   const int kv = blockIdx.y;
-  const int L = ctx_lens[seq];
+  const int zid = blockIdx.z;     // context slice (flash-decoding split)
+  const int Lall = ctx_lens[seq];
+  const int c0 = zid * slice_len;
+  const int L = min(Lall, c0 + slice_len);  // this slice: [c0, L)
   const int tid = threadIdx.x;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* q_lds = reinterpret_cast<float*>(smem_raw);          // GROUP*D
   float* denom = q_lds + GROUP * D;                           // padded to 4
-  float* scores = denom + ((GROUP + 3) & ~3);                 // GROUP*Lpad
+  float* gmax = denom + ((GROUP + 3) & ~3);                   // slice maxes
+  float* scores = gmax + ((GROUP + 3) & ~3);                  // GROUP*Lpad
   __hip_bfloat16* v_lds = reinterpret_cast<__hip_bfloat16*>(
       scores + (size_t)GROUP * Lpad);                         // VTILE*D
 
@@ -59,7 +66,7 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   __syncthreads();
 
   // ---- phase 1: one token per lane, per-head partials in VGPRs ----
-  for (int t = tid; t < L; t += blockDim.x) {
+  for (int t = c0 + tid; t < L; t += blockDim.x) {
     const int64_t row = (int64_t)bt[t / block_size] * block_size
                         + t % block_size;
     const bf16x8* kp = reinterpret_cast<const bf16x8*>(
@@ -81,7 +88,7 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
       }
     }
     #pragma unroll
-    for (int h = 0; h < GROUP; ++h) scores[(size_t)h * Lpad + t] = part[h];
+    for (int h = 0; h < GROUP; ++h) scores[(size_t)h * Lpad + (t - c0)] = part[h];
   }
   __syncthreads();
 
@@ -89,19 +96,23 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   const int wid = tid / WAVE;
   const int wlane = tid % WAVE;
   const int nw = blockDim.x / WAVE;
+  const int Ls = L - c0;  // tokens in this slice (may be <= 0)
   for (int h = wid; h < GROUP; h += nw) {
     float* s = scores + (size_t)h * Lpad;
-    float m = -INFINITY;
-    for (int t = wlane; t < L; t += WAVE) m = fmaxf(m, s[t]);
+    float m = -1e30f;
+    for (int t = wlane; t < Ls; t += WAVE) m = fmaxf(m, s[t]);
     m = wave_max(m);
     float d = 0.f;
-    for (int t = wlane; t < L; t += WAVE) {
+    for (int t = wlane; t < Ls; t += WAVE) {
       float e = __expf(s[t] - m);
       s[t] = e;
       d += e;
     }
     d = wave_sum(d);
-    if (wlane == 0) denom[h] = d;
+    if (wlane == 0) {
+      denom[h] = d;
+      gmax[h] = m;
+    }
   }
   __syncthreads();
 
@@ -112,11 +123,11 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   const int u = tid;                    // one unit per thread (tid < UNITS)
   const int uh = u / DV, ud = u % DV;
 
-  for (int base = 0; base < L; base += VTILE) {
-    const int tile = min(VTILE, L - base);
+  for (int base = 0; base < Ls; base += VTILE) {
+    const int tile = min(VTILE, Ls - base);
     // stage V rows [base, base+tile) for this kv head
     for (int i = tid; i < tile * DV; i += blockDim.x) {
-      const int tt = base + i / DV;
+      const int tt = c0 + base + i / DV;
       const int64_t row = (int64_t)bt[tt / block_size] * block_size
                           + tt % block_size;
       reinterpret_cast<bf16x8*>(v_lds)[i] =
@@ -138,13 +149,59 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   }
 
   if (u < UNITS) {
-    const float inv = 1.f / denom[uh];
     const int h = kv * GROUP + uh;
-    bf16x8 o;
-    #pragma unroll
-    for (int e = 0; e < 8; ++e) o.v[e] = f2bf(acc[e] * inv);
-    *reinterpret_cast<bf16x8*>(out + ((int64_t)seq * H + h) * D + ud * 8) = o;
+    if (o_part == nullptr) {
+      const float inv = 1.f / denom[uh];
+      bf16x8 o;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) o.v[e] = f2bf(acc[e] * inv);
+      *reinterpret_cast<bf16x8*>(out + ((int64_t)seq * H + h) * D + ud * 8) = o;
+    } else {
+      // flash-decoding partial: unnormalized o + (m, l) per slice
+      const int64_t oz = (((int64_t)zid * gridDim.x + seq) * H + h) * D + ud * 8;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) o_part[oz + e] = acc[e];
+      if (ud == 0) {
+        const int64_t mz = (((int64_t)zid * gridDim.x + seq) * H + h) * 2;
+        ml_part[mz] = (Ls > 0) ? gmax[uh] : -1e30f;
+        ml_part[mz + 1] = (Ls > 0) ? denom[uh] : 0.f;
+      }
+    }
   }
+}
+
+// flash-decoding combine: out[seq,h] = sum_z w_z*o_z / sum_z w_z*l_z,
+// w_z = exp(m_z - max_z m_z). grid (N, H), D/64 elems per lane.
+template <int D>
+__global__ __launch_bounds__(64)
+void paged_decode_combine_kernel(const float* __restrict__ o_part,
+                                 const float* __restrict__ ml_part,
+                                 __hip_bfloat16* __restrict__ out,
+                                 int H, int cs) {
+  const int seq = blockIdx.x;
+  const int h = blockIdx.y;
+  const int N = gridDim.x;
+  const int lane = threadIdx.x;
+  float M = -1e30f;
+  for (int z = 0; z < cs; ++z)
+    M = fmaxf(M, ml_part[(((int64_t)z * N + seq) * H + h) * 2]);
+  float Lsum = 0.f;
+  float o[D / 64];
+  #pragma unroll
+  for (int e = 0; e < D / 64; ++e) o[e] = 0.f;
+  for (int z = 0; z < cs; ++z) {
+    const int64_t mz = (((int64_t)z * N + seq) * H + h) * 2;
+    const float w = __expf(ml_part[mz] - M);
+    Lsum += w * ml_part[mz + 1];
+    const int64_t oz = (((int64_t)z * N + seq) * H + h) * D;
+    #pragma unroll
+    for (int e = 0; e < D / 64; ++e)
+      o[e] += w * o_part[oz + lane + e * 64];
+  }
+  const float inv = 1.f / Lsum;
+  #pragma unroll
+  for (int e = 0; e < D / 64; ++e)
+    out[((int64_t)seq * H + h) * D + lane + e * 64] = f2bf(o[e] * inv);
 }
 
 template <int D, int GROUP>
@@ -152,16 +209,36 @@ void launch(const torch::Tensor& q, const torch::Tensor& kcache,
             const torch::Tensor& vcache, const torch::Tensor& bt,
             const torch::Tensor& ctx, torch::Tensor& out, int H, int KV,
             int max_nb, int block_size, int Lpad, float scale, size_t smem,
-            int64_t q_row_stride) {
-  dim3 grid(q.size(0), KV), block(256);
+            int64_t q_row_stride, int cs, int slice_len) {
+  const int N = q.size(0);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid(N, KV, cs), block(256);
+  float* o_p = nullptr;
+  float* ml_p = nullptr;
+  torch::Tensor o_part, ml_part;
+  if (cs > 1) {
+    auto opts = out.options().dtype(at::kFloat);
+    o_part = torch::empty({cs, N, H, D}, opts);
+    ml_part = torch::empty({cs, N, H, 2}, opts);
+    o_p = o_part.data_ptr<float>();
+    ml_p = ml_part.data_ptr<float>();
+  }
   hipLaunchKernelGGL((paged_decode_kernel<D, GROUP>), grid, block, smem,
-                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     stream,
                      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(kcache.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(vcache.data_ptr()),
                      bt.data_ptr<int>(), ctx.data_ptr<int>(),
                      reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
-                     H, KV, max_nb, block_size, Lpad, scale, q_row_stride);
+                     o_p, ml_p,
+                     H, KV, max_nb, block_size, Lpad, scale, q_row_stride,
+                     slice_len);
+  if (cs > 1) {
+    hipLaunchKernelGGL((paged_decode_combine_kernel<D>), dim3(N, H), dim3(64),
+                       0, stream, o_p, ml_p,
+                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                       H, cs);
+  }
 }
 
 }  // namespace
@@ -187,8 +264,15 @@ torch::Tensor paged_attention_decode_strided(
   if (N == 0) return out;
 
   const int max_ctx = max_nb * block_size;
-  const int Lpad = max_ctx + 4;  // +pad to stagger LDS banks across heads
-  size_t smem = (size_t)group * D * 4 + ((group + 3) & ~3) * 4
+  // flash-decoding context split: fill the chip (N*KV blocks alone leave
+  // most CUs idle at decode batch sizes) and shorten the per-block chain
+  int cs = 1;
+  while (cs * 2 <= 8 && N * KV * cs < 1024
+         && (max_ctx + cs * 2 - 1) / (cs * 2) >= 128)
+    cs *= 2;
+  const int slice_len = (max_ctx + cs - 1) / cs;
+  const int Lpad = slice_len + 4;  // +pad to stagger LDS banks across heads
+  size_t smem = (size_t)group * D * 4 + 2 * ((group + 3) & ~3) * 4
                 + (size_t)group * Lpad * 4 + (size_t)VTILE * D * 2;
   TORCH_CHECK(smem <= 160 * 1024,
               "context too long for single-pass decode kernel: ", max_ctx);
@@ -199,7 +283,8 @@ torch::Tensor paged_attention_decode_strided(
   #define CASE(DD, GG) \
     if (D == DD && group == GG) { \
       launch<DD, GG>(q, kcache, vcache, bt, ctx, out, H, KV, max_nb, \
-                     block_size, Lpad, sc, smem, q_row_stride); \
+                     block_size, Lpad, sc, smem, q_row_stride, cs, \
+                     slice_len); \
       HIP_CHECK_LAST(); return out; }
   CASE(128, 7) CASE(128, 5) CASE(128, 4) CASE(128, 8) CASE(128, 6)
   CASE(128, 2) CASE(128, 1) CASE(64, 7) CASE(64, 4) CASE(64, 2) CASE(64, 1)

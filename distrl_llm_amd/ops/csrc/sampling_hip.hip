@@ -52,7 +52,7 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
   const int nvec = BF16 ? V / 8 : 0;
 
   // ---- pass A: max (bf16: 16 B/lane vector loads) ----
-  float m = -INFINITY;
+  float m = -1e30f;
   if constexpr (BF16) {
     const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
     for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
@@ -133,7 +133,7 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
   const float z_min = -(thr_bin + 1) * (ZRANGE / NBINS);
 
   // ---- pass C: Gumbel-argmax over the kept set ----
-  float best = -INFINITY;
+  float best = -1e30f;
   int64_t best_i = 0;
   if constexpr (BF16) {
     const bf16x8* l8 = reinterpret_cast<const bf16x8*>(lr);
