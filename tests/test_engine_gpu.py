@@ -87,3 +87,38 @@ def test_native_extension_is_loaded():
     ext = get_extension()
     assert ext is not None, "gfx950 extension must load on the GPU box"
     assert "_build" in ext.__file__  # in-tree .so
+
+
+def test_llama3_8b_bf16_path():
+    """BASELINE config 4 readiness: Llama-3-8B bf16 (non-quantized) decode
+    + PG learner micro-step on the same engine machinery (no qkv bias,
+    GQA 4:1, vocab 128256)."""
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    from distrl_llm_amd.train.learner import Learner
+    from distrl_llm_amd.utils.tokenizer import ByteTokenizer
+
+    dev = torch.device("cuda:0")
+    spec = get_spec("meta-llama/Meta-Llama-3-8B")
+    model = CausalLM(spec, lora_r=32, lora_alpha=16, dtype=torch.bfloat16,
+                     device=dev).random_init(3407)
+    tok = ByteTokenizer(vocab_size=spec.vocab_size)
+    engine = Engine(model, EngineConfig(max_seq_length=512,
+                                        gpu_memory_utilization=0.2),
+                    device=dev, seed=0)
+    prompts = [tok.encode("Compute 6*7."), tok.encode("What is 10-3?")]
+    sp = SamplingParams(max_tokens=12, temperature=1.2, n=2, top_p=0.95)
+    outs = engine.generate(prompts, sp, eos_token_id=tok.eos_token_id)
+    assert len(outs) == 2 and all(len(o) == 2 for o in outs)
+    for o in outs:
+        for ids in o:
+            assert 1 <= len(ids) <= 12
+
+    learner = Learner(model, tok, lr=2e-5, max_prompt_tokens=64,
+                      max_new_tokens=32, train_batch_size=2)
+    answers = [tok.decode(ids) for o in outs for ids in o]
+    probs = ["Compute 6*7."] * 2 + ["What is 10-3?"] * 2
+    loss = learner.accumulate_gradients(probs, answers, [1.0, -0.2, 0.4, -0.6])
+    learner.step()
+    assert torch.isfinite(torch.tensor(loss))
