@@ -75,7 +75,12 @@ class LoRALinear(nn.Module):
         y = F.linear(x, self.weight, self.bias)
         if self.r > 0:
             xd = self.lora_dropout(x)
-            y = y + F.linear(F.linear(xd, self.lora_A), self.lora_B) * self.scale
+            u = F.linear(xd, self.lora_A)
+            # addmm folds the scale and the accumulation into the adapter
+            # GEMM itself (no separate mul/add kernels in the hot path)
+            y2 = y.reshape(-1, self.out_features)
+            y = torch.addmm(y2, u.reshape(-1, self.r), self.lora_B.t(),
+                            beta=1.0, alpha=self.scale).view_as(y)
         return y
 
     def extra_repr(self) -> str:
