@@ -62,7 +62,15 @@ class FusedWeights:
         # GEMM path streams 4.3x fewer weight bytes and is selected with
         # DISTRL_DECODE_NF4=1 (kept fully tested).
         import os
-        want_nf4 = use_nf4 and os.environ.get("DISTRL_DECODE_NF4", "0") == "1"
+        env = os.environ.get("DISTRL_DECODE_NF4")
+        if env is not None:
+            want_nf4 = use_nf4 and env == "1"
+        else:
+            # default: merged cache for models whose bf16 copy is cheap;
+            # fused nf4 path for very large models (e.g. 32B: a second
+            # 65 GB merged copy is not worth the HBM)
+            n_params = sum(p.numel() for p in model.parameters())
+            want_nf4 = use_nf4 and n_params > 16e9
         self.nf4 = bool(want_nf4 and q0.weight_nf4 is not None
                         and q0.weight.is_cuda and shapes_ok)
         self.lora_r = q0.r
