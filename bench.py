@@ -86,7 +86,9 @@ def main():
         "episodes": 1,
         "num_candidates": args.num_candidates,
         "batch_size": batch_size,
-        "train_batch_size": 8,
+        # hardware-tuned micro-batch (288 GB HBM; reference's 8 was a
+        # 24 GB OOM bound): same objective math, larger GEMMs (+12%)
+        "train_batch_size": 16,
         "temperature": 1.2,
         "save_every": 10**9,
         "eval_every": 0,
@@ -193,6 +195,7 @@ def main():
             "global_batch": batch_size,
             "num_candidates": args.num_candidates,
             "samples_per_step": batch_size * args.num_candidates,
+            "train_micro_batch": 16,
             "seq_len": max_prompt + max_new,
             "max_new_tokens": max_new,
             "parallelism": f"dp{world_size} ({num_actors} actors + "

@@ -138,22 +138,30 @@ class Engine:
         return self.model.logits(x[last_idx])
 
     def _prefill_attention(self, q, k, v, lens: List[int]) -> torch.Tensor:
-        """Causal attention over concatenated prompts. GPU: torch SDPA per
-        prompt (library path; custom flash prefill kernel is an upgrade
-        path — decode is the hot loop); CPU: reference varlen."""
+        """Causal attention over concatenated prompts. GPU: one batched
+        SDPA call over right-padded prompts (library path; custom flash
+        prefill kernel is an upgrade path — decode is the hot loop);
+        CPU: reference varlen."""
         if q.is_cuda:
             group = self.spec.num_heads // self.spec.num_kv_heads
-            outs = []
+            B, Lmax = len(lens), max(lens)
+            H, KV, D = self.spec.num_heads, self.spec.num_kv_heads, self.spec.head_dim
+            qp = q.new_zeros(B, Lmax, H, D)
+            kp = k.new_zeros(B, Lmax, KV, D)
+            vp = v.new_zeros(B, Lmax, KV, D)
             start = 0
-            for L in lens:
-                qs = q[start:start + L].transpose(0, 1).unsqueeze(0)
-                ks = k[start:start + L].repeat_interleave(group, 1).transpose(0, 1).unsqueeze(0)
-                vs = v[start:start + L].repeat_interleave(group, 1).transpose(0, 1).unsqueeze(0)
-                o = F.scaled_dot_product_attention(qs, ks, vs, is_causal=True,
-                                                   scale=self.scale)
-                outs.append(o.squeeze(0).transpose(0, 1))
+            for i, L in enumerate(lens):
+                qp[i, :L] = q[start:start + L]
+                kp[i, :L] = k[start:start + L]
+                vp[i, :L] = v[start:start + L]
                 start += L
-            return torch.cat(outs, 0)
+            kp = kp.repeat_interleave(group, 2)
+            vp = vp.repeat_interleave(group, 2)
+            o = F.scaled_dot_product_attention(
+                qp.transpose(1, 2), kp.transpose(1, 2), vp.transpose(1, 2),
+                is_causal=True, scale=self.scale).transpose(1, 2)
+            # trailing pad rows never influence real rows under causal
+            return torch.cat([o[i, :L] for i, L in enumerate(lens)], 0)
         return R.varlen_prefill_attention(q, k, v, lens, self.scale)
 
     # ------------------------------------------------------------ decode
