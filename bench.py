@@ -204,7 +204,8 @@ def main():
     }
     print(json.dumps(result), flush=True)
 
-    trainer.fabric.broadcast_obj(("stop", None), src=0)
+    trainer._cmd_seq = getattr(trainer, "_cmd_seq", 0) + 1
+    trainer.fabric.broadcast_obj((trainer._cmd_seq, "stop", None), src=0)
     trainer.fabric.close()
 
 

@@ -69,15 +69,25 @@ class Trainer:
     # ----------------------------------------------------------- plumbing
 
     def _cmd(self, name: str, payload=None):
-        """Rank 0: broadcast a command then execute it locally."""
+        """Rank 0: broadcast a command then execute it locally. Each
+        command carries a monotonically increasing sequence id — the
+        explicit round handshake SURVEY.md §5.2 asks for in place of the
+        reference's implicit adapter-file ordering."""
         assert self.fabric.rank == 0
-        self.fabric.broadcast_obj((name, payload), src=0)
+        self._cmd_seq = getattr(self, "_cmd_seq", 0) + 1
+        self.fabric.broadcast_obj((self._cmd_seq, name, payload), src=0)
         return self._handle(name, payload)
 
     def follower_loop(self):
         """Ranks != 0: execute the command stream until stop."""
+        expected = 0
         while True:
-            name, payload = self.fabric.broadcast_obj(src=0)
+            seq, name, payload = self.fabric.broadcast_obj(src=0)
+            expected += 1
+            if seq != expected:
+                raise RuntimeError(
+                    f"rank {self.fabric.rank}: command stream out of sync "
+                    f"(got seq {seq}, expected {expected}, cmd {name!r})")
             if name == "stop":
                 break
             self._handle(name, payload)
@@ -205,7 +215,8 @@ class Trainer:
         try:
             self._train_rank0()
         finally:
-            self.fabric.broadcast_obj(("stop", None), src=0)
+            self._cmd_seq = getattr(self, "_cmd_seq", 0) + 1
+            self.fabric.broadcast_obj((self._cmd_seq, "stop", None), src=0)
 
     def rl_round(self, batch: Dict, sp_dict: Optional[Dict] = None) -> Dict:
         """Rank 0: one full RL round — generate fan-out, rewards,
