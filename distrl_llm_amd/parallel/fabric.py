@@ -60,6 +60,16 @@ class Fabric:
                 device_id=device if device.type == "cuda" else None)
         self._owns_pg = True
 
+        # explicit gloo group for the CPU control plane: object collectives
+        # (prompt dicts, generations) and barriers are pinned here so their
+        # routing never depends on the combined backend's default device
+        if device.type == "cuda":
+            self.cpu_group = dist.new_group(
+                backend="gloo",
+                timeout=datetime.timedelta(seconds=timeout_s))
+        else:
+            self.cpu_group = None  # default group already is gloo
+
         self.learner_group = dist.new_group(self.learner_ranks,
                                             timeout=datetime.timedelta(seconds=timeout_s))
 
@@ -67,16 +77,17 @@ class Fabric:
 
     def broadcast_obj(self, obj=None, src: int = 0):
         box = [obj]
-        dist.broadcast_object_list(box, src=src)
+        dist.broadcast_object_list(box, src=src, group=self.cpu_group,
+                                   device=torch.device("cpu"))
         return box[0]
 
     def gather_obj(self, obj, dst: int = 0) -> Optional[List]:
         out = [None] * self.world_size if self.rank == dst else None
-        dist.gather_object(obj, out, dst=dst)
+        dist.gather_object(obj, out, dst=dst, group=self.cpu_group)
         return out
 
     def barrier(self):
-        dist.barrier()
+        dist.barrier(group=self.cpu_group)
 
     # ------------------------------------------------------- data plane
 
