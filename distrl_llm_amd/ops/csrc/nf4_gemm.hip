@@ -411,8 +411,10 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
     u_stride = u->stride(0);
   }
 
-  // largest m-tile (short per-block K chains; occupancy comes from split-K)
-  int mt = std::min<int>((M + 15) / 16, 5);
+  // m-tile size: MT5 costs 232 regs (2 waves/SIMD); MT3/4 fit 3 waves/SIMD
+  int mtcap = 5;
+  if (const char* e = getenv("DISTRL_NF4_MTCAP")) mtcap = atoi(e);
+  int mt = std::min<int>((M + 15) / 16, std::max(1, mtcap));
   const int BM = 16 * mt;
   const int SK = (mt <= 2) ? 512 : 256;  // must mirror the kernel constexpr
   TORCH_CHECK(K % SK == 0, "nf4_gemm: K (", K, ") % ", SK, " != 0");
@@ -420,8 +422,11 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   // decode shapes otherwise run at ~1 wave/SIMD, fully latency-exposed)
   const int npanels = (int)K / SK;
   const int base_blocks = ((M + BM - 1) / BM) * ((int)N / 256);
+  int blk_target = 512;
+  if (const char* e = getenv("DISTRL_NF4_BLKTGT")) blk_target = atoi(e);
   int ksplit = 1;
-  while (ksplit * 2 <= npanels && base_blocks * ksplit < 512 && ksplit < 16)
+  while (ksplit * 2 <= npanels && base_blocks * ksplit < blk_target
+         && ksplit < 16)
     ksplit *= 2;
   if (const char* e = getenv("DISTRL_NF4_KSPLIT")) ksplit = atoi(e);
   ksplit = std::max(1, std::min(ksplit, npanels));
