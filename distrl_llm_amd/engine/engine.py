@@ -17,7 +17,7 @@ broadcast into these tensors.
 from __future__ import annotations
 
 import time
-from typing import Dict, List, Optional, Sequence as Seq
+from typing import List, Optional
 
 import torch
 import torch.nn.functional as F
@@ -26,7 +26,7 @@ from ..config import EngineConfig, SamplingParams
 from ..models.model import CausalLM
 from ..ops import functional as OF
 from ..ops import reference as R
-from .kvcache import BlockAllocator, KVCachePool, Sequence
+from .kvcache import KVCachePool, Sequence
 
 
 class Engine:

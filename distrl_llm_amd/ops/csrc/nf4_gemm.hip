@@ -422,7 +422,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   // decode shapes otherwise run at ~1 wave/SIMD, fully latency-exposed)
   const int npanels = (int)K / SK;
   const int base_blocks = ((M + BM - 1) / BM) * ((int)N / 256);
-  int blk_target = 512;
+  int blk_target = 1024;
   if (const char* e = getenv("DISTRL_NF4_BLKTGT")) blk_target = atoi(e);
   int ksplit = 1;
   while (ksplit * 2 <= npanels && base_blocks * ksplit < blk_target

@@ -13,7 +13,6 @@ quant/dequant, and the fused log-prob + advantage loss.
 
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch

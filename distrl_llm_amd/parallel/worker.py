@@ -10,7 +10,7 @@ construction instead of Ray placement groups.
 from __future__ import annotations
 
 import os
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import torch
 

@@ -12,7 +12,7 @@ in the reference but is never wired into ``reward_function``
 from __future__ import annotations
 
 import re
-from typing import List, Sequence
+from typing import Sequence
 
 import numpy as np
 

@@ -12,7 +12,7 @@ printable escape so reward regexes operate on real strings.
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 
 class ByteTokenizer:
