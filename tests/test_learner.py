@@ -104,3 +104,19 @@ def test_adam8bit_tracks_fp32_adam():
     # 8-bit state quantization: small tracking error, same trajectory scale
     rel = (p1 - p2).abs().max() / p2.abs().max()
     assert rel < 0.05, rel.item()
+
+
+def test_micro_batch_invariance():
+    """The accumulated gradient must not depend on the micro-batch split
+    (justifies hardware-tuned train_batch_size: same objective math)."""
+    grads = {}
+    for mb in (2, 4):
+        torch.manual_seed(0)
+        learner = _make_learner()
+        learner.train_batch_size = mb
+        problems = [f"problem {i}" for i in range(4)]
+        answers = [f"<answer>{i}</answer>" for i in range(4)]
+        learner.accumulate_gradients(problems, answers, [0.5, -1.0, 0.25, 0.8])
+        grads[mb] = [p.grad.clone() for p in learner.params]
+    for g2, g4 in zip(grads[2], grads[4]):
+        torch.testing.assert_close(g2, g4, rtol=1e-4, atol=1e-6)
