@@ -243,6 +243,9 @@ class Trainer:
         self._cmd("sync_weights")
         update_dur = time.time() - t0
 
+        gen_tokens = sum(
+            int(t) for cand in candidates for group in cand["token_lengths"]
+            for t in group)
         stats = dict(stats)
         stats.update({
             "loss": loss,
@@ -250,6 +253,10 @@ class Trainer:
             "timing/update_duration": update_dur,
             "timing/reward_duration": reward_dur,
             "timing/generation_duration": gen_dur,
+            "timing/generation_tokens_per_sec":
+                gen_tokens / gen_dur if gen_dur > 0 else 0.0,
+            "timing/samples_per_sec": (
+                len(problems) / (gen_dur + reward_dur + update_dur)),
         })
         return stats
 
@@ -283,6 +290,9 @@ class Trainer:
                         "timing/update_duration": stats["timing/update_duration"],
                         "timing/reward_duration": stats["timing/reward_duration"],
                         "timing/generation_duration": stats["timing/generation_duration"],
+                        "timing/generation_tokens_per_sec":
+                            stats["timing/generation_tokens_per_sec"],
+                        "timing/samples_per_sec": stats["timing/samples_per_sec"],
                     }, step=total_batch_steps)
 
                 if self.eval_every > 0 and total_batch_steps % self.eval_every == 0:
