@@ -138,10 +138,30 @@ class Engine:
         return self.model.logits(x[last_idx])
 
     def _prefill_attention(self, q, k, v, lens: List[int]) -> torch.Tensor:
-        """Causal attention over concatenated prompts. GPU: one batched
-        SDPA call over right-padded prompts (library path; custom flash
-        prefill kernel is an upgrade path — decode is the hot loop);
-        CPU: reference varlen."""
+        """Causal attention over concatenated prompts. GPU: the varlen
+        causal prefill HIP kernel (q-row tiles, lane-per-kv scoring —
+        ops/csrc/paged_attn.hip); batched-SDPA fallback only for prompts
+        too long for its LDS score tiles. CPU: reference varlen."""
+        if q.is_cuda and max(lens) <= 2200:
+            from ..ops.build import get_extension
+            ext = get_extension()
+            if ext is None:
+                raise RuntimeError("prefill requires the gfx950 extension")
+            tq0, trows, tkv0 = [], [], []
+            start = 0
+            for L in lens:
+                for r0 in range(0, L, 16):
+                    tq0.append(start + r0)
+                    trows.append(min(16, L - r0))
+                    tkv0.append(start)
+                start += L
+            dev = q.device
+            return ext.prefill_attention(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                torch.tensor(tq0, dtype=torch.int32, device=dev),
+                torch.tensor(trows, dtype=torch.int32, device=dev),
+                torch.tensor(tkv0, dtype=torch.int32, device=dev),
+                max(lens), self.scale)
         if q.is_cuda:
             group = self.spec.num_heads // self.spec.num_kv_heads
             B, Lmax = len(lens), max(lens)

@@ -17,6 +17,10 @@ void rope_scatter_qkv(torch::Tensor qkv, torch::Tensor positions,
                       torch::Tensor key_cache, torch::Tensor value_cache,
                       int64_t H, int64_t KV, int64_t D);
 torch::Tensor silu_mul_packed(torch::Tensor gu);
+torch::Tensor prefill_attention(torch::Tensor q, torch::Tensor k,
+                                torch::Tensor v, torch::Tensor tile_q0,
+                                torch::Tensor tile_rows, torch::Tensor tile_kv0,
+                                int64_t max_len, double scale);
 torch::Tensor paged_attention_decode_strided(
     torch::Tensor q, int64_t n_heads, int64_t head_dim, int64_t q_row_stride,
     torch::Tensor kcache, torch::Tensor vcache, torch::Tensor block_tables,
@@ -61,6 +65,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_mul_packed", &silu_mul_packed, "SiLU*mul on packed [gate|up]");
   m.def("paged_attention_decode_strided", &paged_attention_decode_strided,
         "paged GQA decode attention (packed-q rows)");
+  m.def("prefill_attention", &prefill_attention,
+        "varlen causal prefill attention (prompt phase)");
   m.def("kv_cache_scatter", &kv_cache_scatter, "paged KV cache scatter");
   m.def("paged_attention_decode", &paged_attention_decode,
         "paged GQA decode attention");

@@ -302,3 +302,181 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kcache,
       q.view({q.size(0), q.size(1) * q.size(2)}), q.size(1), q.size(2),
       q.size(1) * q.size(2), kcache, vcache, block_tables, ctx_lens, scale);
 }
+
+// ---- varlen causal prefill attention (prompt phase) ----
+// One workgroup per (q-row-tile of 16, q-head): same three-phase structure
+// as the decode kernel with the GQA group slot re-used for the 16 q rows
+// of the tile — lane-per-kv-token scoring against the LDS-staged q tile
+// (with causal masking), per-row softmax, PV through LDS V tiles.
+// K/V are the flat pre-RoPE'd prefill tensors (no paging needed here).
+// Completes the "prefill + decode as HIP kernels" surface (BASELINE.json
+// north star); replaces the torch-SDPA library call in the engine.
+namespace {
+
+template <int D>
+__global__ __launch_bounds__(256)
+void prefill_attn_kernel(const __hip_bfloat16* __restrict__ q,   // (T,H,D)
+                         const __hip_bfloat16* __restrict__ k,   // (T,KV,D)
+                         const __hip_bfloat16* __restrict__ v,   // (T,KV,D)
+                         const int* __restrict__ tile_q0,    // global q row
+                         const int* __restrict__ tile_rows,  // rows in tile
+                         const int* __restrict__ tile_kv0,   // prompt start
+                         __hip_bfloat16* __restrict__ out,       // (T,H,D)
+                         int H, int KV, int Lpad, float scale) {
+  constexpr int QT = 16;
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int kvh = h / (H / KV);
+  const int q0 = tile_q0[tile];
+  const int rows = tile_rows[tile];
+  const int kv0 = tile_kv0[tile];
+  // causal: tile rows attend to kv positions [kv0, q0 + rows)
+  const int Lmax = q0 - kv0 + rows;
+  const int tid = threadIdx.x;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* q_lds = reinterpret_cast<float*>(smem_raw);          // QT*D
+  float* denom = q_lds + QT * D;                              // QT
+  float* scores = denom + QT;                                 // QT*Lpad
+  __hip_bfloat16* v_lds = reinterpret_cast<__hip_bfloat16*>(
+      scores + (size_t)QT * Lpad);                            // VTILE*D
+
+  for (int i = tid; i < QT * D; i += blockDim.x) {
+    const int r = i / D;
+    q_lds[i] = (r < rows)
+        ? bf2f(q[((int64_t)(q0 + r) * H + h) * D + i % D]) * scale
+        : 0.f;
+  }
+  __syncthreads();
+
+  // phase 1: lane-per-kv-token scoring with causal mask
+  for (int t = tid; t < Lmax; t += blockDim.x) {
+    const __hip_bfloat16* kp = k + ((int64_t)(kv0 + t) * KV + kvh) * D;
+    float part[QT];
+    #pragma unroll
+    for (int r = 0; r < QT; ++r) part[r] = 0.f;
+    #pragma unroll
+    for (int c = 0; c < D / 8; ++c) {
+      const bf16x8 kvec = *reinterpret_cast<const bf16x8*>(kp + c * 8);
+      float kf[8];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) kf[j] = bf2f(kvec.v[j]);
+      #pragma unroll
+      for (int r = 0; r < QT; ++r) {
+        const float* qr = q_lds + r * D + c * 8;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) part[r] += qr[j] * kf[j];
+      }
+    }
+    #pragma unroll
+    for (int r = 0; r < QT; ++r) {
+      // causal: kv position kv0+t visible to q row q0+r iff kv0+t <= q0+r
+      const bool vis = (kv0 + t) <= (q0 + r);
+      scores[(size_t)r * Lpad + t] = vis ? part[r] : -1e30f;
+    }
+  }
+  __syncthreads();
+
+  // phase 2: per-row softmax
+  const int wid = tid / WAVE;
+  const int wlane = tid % WAVE;
+  const int nw = blockDim.x / WAVE;
+  for (int r = wid; r < QT; r += nw) {
+    float* s = scores + (size_t)r * Lpad;
+    float m = -1e30f;
+    for (int t = wlane; t < Lmax; t += WAVE) m = fmaxf(m, s[t]);
+    m = wave_max(m);
+    float d = 0.f;
+    for (int t = wlane; t < Lmax; t += WAVE) {
+      float e = __expf(s[t] - m);
+      s[t] = e;
+      d += e;
+    }
+    d = wave_sum(d);
+    if (wlane == 0) denom[r] = d;
+  }
+  __syncthreads();
+
+  // phase 3: PV through LDS V tiles
+  constexpr int DV = D / 8;
+  constexpr int UNITS = QT * DV;  // 128 (D=64) or 256 (D=128)
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  const int u = tid;
+  const int ur = u / DV, ud = u % DV;
+  for (int base = 0; base < Lmax; base += VTILE) {
+    const int tl = min(VTILE, Lmax - base);
+    for (int i = tid; i < tl * DV; i += blockDim.x) {
+      reinterpret_cast<bf16x8*>(v_lds)[i] =
+          *reinterpret_cast<const bf16x8*>(
+              v + ((int64_t)(kv0 + base + i / DV) * KV + kvh) * D
+              + (i % DV) * 8);
+    }
+    __syncthreads();
+    if (u < UNITS) {
+      const float* ps = scores + (size_t)ur * Lpad + base;
+      #pragma unroll 4
+      for (int j = 0; j < tl; ++j) {
+        const float p = ps[j];
+        const bf16x8 vv = reinterpret_cast<const bf16x8*>(v_lds)[j * DV + ud];
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) acc[e] += p * bf2f(vv.v[e]);
+      }
+    }
+    __syncthreads();
+  }
+
+  if (u < UNITS && ur < rows) {
+    const float inv = 1.f / denom[ur];
+    bf16x8 o;
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) o.v[e] = f2bf(acc[e] * inv);
+    *reinterpret_cast<bf16x8*>(
+        out + ((int64_t)(q0 + ur) * H + h) * D + ud * 8) = o;
+  }
+}
+
+}  // namespace
+
+torch::Tensor prefill_attention(torch::Tensor q, torch::Tensor k,
+                                torch::Tensor v, torch::Tensor tile_q0,
+                                torch::Tensor tile_rows, torch::Tensor tile_kv0,
+                                int64_t max_len, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous()
+              && v.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(tile_q0.scalar_type() == at::kInt);
+  const int H = q.size(1), D = q.size(2), KV = k.size(1);
+  const int n_tiles = tile_q0.size(0);
+  auto out = torch::empty_like(q);
+  if (n_tiles == 0) return out;
+  const int Lpad = (int)max_len + 4;
+  size_t smem = 16 * D * 4 + 16 * 4 + (size_t)16 * Lpad * 4
+                + (size_t)VTILE * D * 2;
+  TORCH_CHECK(smem <= 160 * 1024, "prompt too long for prefill kernel: ",
+              max_len);
+  dim3 grid(n_tiles, H), block(256);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (D == 128) {
+    hipLaunchKernelGGL(prefill_attn_kernel<128>, grid, block, smem, stream,
+        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+        tile_q0.data_ptr<int>(), tile_rows.data_ptr<int>(),
+        tile_kv0.data_ptr<int>(),
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+        H, KV, Lpad, (float)scale);
+  } else if (D == 64) {
+    hipLaunchKernelGGL(prefill_attn_kernel<64>, grid, block, smem, stream,
+        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+        tile_q0.data_ptr<int>(), tile_rows.data_ptr<int>(),
+        tile_kv0.data_ptr<int>(),
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+        H, KV, Lpad, (float)scale);
+  } else {
+    TORCH_CHECK(false, "prefill attention: head_dim 64 or 128 only");
+  }
+  HIP_CHECK_LAST();
+  return out;
+}

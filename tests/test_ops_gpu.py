@@ -344,3 +344,33 @@ def test_engine_nf4_matches_merged():
     for a, b in zip(out_nf4, out_bf16):
         agree = sum(x == y for x, y in zip(a[0], b[0]))
         assert agree >= 7, (a[0], b[0])
+
+
+@pytest.mark.parametrize("D,H,KV,lens", [
+    (128, 28, 4, [210, 17, 333]),
+    (64, 8, 2, [100, 5, 64]),
+])
+def test_prefill_attention(ext, D, H, KV, lens):
+    torch.manual_seed(13)
+    from distrl_llm_amd.ops import reference as R
+    total = sum(lens)
+    q = torch.randn(total, H, D, device=_dev(), dtype=torch.bfloat16)
+    k = torch.randn(total, KV, D, device=_dev(), dtype=torch.bfloat16)
+    v = torch.randn(total, KV, D, device=_dev(), dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(D)
+    tq0, trows, tkv0 = [], [], []
+    start = 0
+    for L in lens:
+        for r0 in range(0, L, 16):
+            tq0.append(start + r0)
+            trows.append(min(16, L - r0))
+            tkv0.append(start)
+        start += L
+    dev = _dev()
+    out = ext.prefill_attention(
+        q, k, v, torch.tensor(tq0, dtype=torch.int32, device=dev),
+        torch.tensor(trows, dtype=torch.int32, device=dev),
+        torch.tensor(tkv0, dtype=torch.int32, device=dev), max(lens), scale)
+    ref = R.varlen_prefill_attention(q.float(), k.float(), v.float(), lens,
+                                     scale)
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
