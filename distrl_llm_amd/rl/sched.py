@@ -87,6 +87,17 @@ def worker_chunk_sizes(batch_size: int, num_actors: int, num_learners: int = 1,
     for i in range(eff_learners):
         learner_chunks[i] = eff_lcs
 
+    # With no actors to absorb it, the remainder (batch not divisible by
+    # the learner chunks — e.g. a dataset's last batch) goes +1 to the
+    # first learners. The reference drops it and then crashes in
+    # split_dict_lists (sum mismatch) — found by property test.
+    leftover = batch_size - sum(actor_chunks) - sum(learner_chunks)
+    i = 0
+    while leftover > 0 and num_learners > 0:
+        learner_chunks[i % num_learners] += 1
+        leftover -= 1
+        i += 1
+
     out = actor_chunks + learner_chunks
     assert sum(out) == batch_size, (out, batch_size)
     return out
