@@ -52,7 +52,6 @@ class Engine:
         self.pool = KVCachePool(self.spec.num_layers, num_blocks,
                                 cfg.kv_block_size, self.spec.num_kv_heads,
                                 self.spec.head_dim, self.dtype, self.device)
-        self._graph_runner = None
         if self.device.type == "cuda":
             from .weights import FusedWeights
             self.fused = FusedWeights(model)
