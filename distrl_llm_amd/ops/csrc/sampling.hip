@@ -30,7 +30,7 @@ template <> DEV_INLINE float ld<__hip_bfloat16>(const __hip_bfloat16* p, int64_t
 template <> DEV_INLINE float ld<float>(const float* p, int64_t i) { return p[i]; }
 
 template <typename T>
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(1024)
 void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
                    float top_p, int top_k, const int64_t* __restrict__ seeds,
                    const int64_t* __restrict__ step,
@@ -39,8 +39,8 @@ void sample_kernel(const T* __restrict__ logits, int V, float inv_temp,
   __shared__ float bin_p[NBINS];
   __shared__ int bin_c[NBINS];
   __shared__ int thr_bin_s;
-  __shared__ float best_val_s[4];
-  __shared__ int64_t best_idx_s[4];
+  __shared__ float best_val_s[16];
+  __shared__ int64_t best_idx_s[16];
 
   const int row = blockIdx.x;
   const T* lr = logits + (int64_t)row * V;
@@ -201,7 +201,7 @@ torch::Tensor sample_tokens(torch::Tensor logits, double temperature,
   const int B = logits.size(0), V = logits.size(1);
   auto out = torch::empty({B}, logits.options().dtype(at::kLong));
   if (B == 0) return out;
-  dim3 grid(B), block(256);
+  dim3 grid(B), block(1024);
   auto stream = at::cuda::getCurrentCUDAStream();
   const float inv_t = 1.f / (float)temperature;
   if (logits.scalar_type() == at::kBFloat16) {
