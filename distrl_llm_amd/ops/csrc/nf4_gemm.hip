@@ -85,6 +85,9 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   if (tid < 256)
     lut2[tid] = make_float2(NF4_LUT[tid & 15], NF4_LUT[tid >> 4]);
   // per-lane register copy of the codebook for the shuffle-LUT dequant
+  // (a pure-VALU cndmask-tree variant measured 1.5x SLOWER: ~9 VALU/value
+  // beats ds_bpermute's latency only on paper — the compiler pipelines
+  // the shuffles ~2-deep and the VALU tree just adds issue pressure)
   const float lut_reg = NF4_LUT[tid & 15];
 
   f32x4 acc[MT][4];
