@@ -76,6 +76,12 @@ def main():
     num_actors = world_size - num_learners
     batch_size = args.prompts_per_gpu * world_size
 
+    # hardware-tuned micro-batch (288 GB HBM; the reference's 8 was a 24 GB
+    # OOM bound): same objective math (tests/test_learner.py micro-batch
+    # invariance), sized so resident activations fit — ~2x layers/hidden of
+    # 32B needs the smaller step
+    micro = 4 if "32b" in model_name.lower() else 16
+
     config = {
         "run_name": "bench",
         "project_name": "bench",
@@ -86,9 +92,7 @@ def main():
         "episodes": 1,
         "num_candidates": args.num_candidates,
         "batch_size": batch_size,
-        # hardware-tuned micro-batch (288 GB HBM; reference's 8 was a
-        # 24 GB OOM bound): same objective math, larger GEMMs (+12%)
-        "train_batch_size": 16,
+        "train_batch_size": micro,
         "temperature": 1.2,
         "save_every": 10**9,
         "eval_every": 0,
@@ -195,7 +199,7 @@ def main():
             "global_batch": batch_size,
             "num_candidates": args.num_candidates,
             "samples_per_step": batch_size * args.num_candidates,
-            "train_micro_batch": 16,
+            "train_micro_batch": micro,
             "seq_len": max_prompt + max_new,
             "max_new_tokens": max_new,
             "parallelism": f"dp{world_size} ({num_actors} actors + "
