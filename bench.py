@@ -80,7 +80,8 @@ def main():
     # OOM bound): same objective math (tests/test_learner.py micro-batch
     # invariance), sized so resident activations fit — ~2x layers/hidden of
     # 32B needs the smaller step
-    micro = 4 if "32b" in model_name.lower() else 16
+    big = "32b" in model_name.lower()
+    micro = 2 if big else 16
 
     config = {
         "run_name": "bench",
@@ -105,8 +106,8 @@ def main():
         "max_lora_rank": 32,
         "topk": args.num_candidates,
         "learner_chunk_size": max(1, batch_size // world_size),
-        "actor_gpu_usage": 0.91,
-        "learner_gpu_usage": 0.35,
+        "actor_gpu_usage": 0.5 if big else 0.91,
+        "learner_gpu_usage": 0.2 if big else 0.35,
         "lora_alpha": 16,
         "lora_dropout": 0.0,
         "seed": 3407,
