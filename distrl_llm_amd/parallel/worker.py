@@ -65,6 +65,12 @@ def build_worker(rank: int, world_size: int, config: Dict,
     if load_4bit:
         model.quantize_nf4_()
 
+    if config.get("load_adapter"):
+        # warm-start from a PEFT adapter directory (every rank loads the
+        # same file, so replicas stay identical)
+        from ..models.lora import load_adapter
+        load_adapter(model, config["load_adapter"])
+
     tokenizer = load_tokenizer(config["model"], spec.vocab_size)
 
     is_learner = rank >= num_actors

@@ -120,3 +120,24 @@ def test_quantize_model_attaches_sidecar():
     # weight replaced by its quantized image: close but not identical
     assert not torch.equal(before, mod.weight)
     assert (before - mod.weight).abs().max() < 0.1
+
+
+def test_cli_warm_start_adapter(tmp_path):
+    """--load_adapter loads a saved PEFT adapter into a fresh worker's
+    model (warm-start extension; the reference has no resume path,
+    SURVEY §5.4)."""
+    from distrl_llm_amd.models.lora import load_adapter, lora_state_dict, save_adapter
+    spec = get_spec("tiny-qwen2")
+    m1 = CausalLM(spec, lora_r=4, lora_alpha=8, dtype=torch.float32).random_init(0)
+    with torch.no_grad():
+        for p in m1.parameters():
+            if p.requires_grad:
+                p.add_(torch.randn_like(p) * 0.05)
+    path = str(tmp_path / "warm")
+    save_adapter(m1, path, "tiny-qwen2", r=4, alpha=8)
+
+    m2 = CausalLM(spec, lora_r=4, lora_alpha=8, dtype=torch.float32).random_init(1)
+    load_adapter(m2, path)
+    a, b = lora_state_dict(m1), lora_state_dict(m2)
+    for k in a:
+        torch.testing.assert_close(a[k], b[k])

@@ -51,6 +51,9 @@ def parse_args():
     # native-framework extensions (not in the reference CLI)
     args.add_argument("--synthetic_dataset", type=int, default=0,
                       help="Use N synthetic MATH-shaped prompts instead of --dataset")
+    args.add_argument("--load_adapter", type=str, default=None,
+                      help="Warm-start: PEFT adapter directory to load into "
+                           "every worker before training")
     args.add_argument("--seed", type=int, default=3407)
     args.add_argument("--backend_device", type=str, default="auto",
                       choices=["auto", "cuda", "cpu"])
@@ -86,6 +89,7 @@ def build_config(args) -> dict:
         "lora_alpha": args.lora_alpha,
         "lora_dropout": args.lora_dropout,
         "seed": args.seed,
+        "load_adapter": args.load_adapter,
     }
 
 
