@@ -246,7 +246,14 @@ class Trainer:
         gen_tokens = sum(
             int(t) for cand in candidates for group in cand["token_lengths"]
             for t in group)
+        sample = {
+            "sample_problem": candidates[0]["problem"][0][0] if candidates else "",
+            "sample_answer": candidates[0]["answers"][0][0] if candidates else "",
+            "sample_reward": (float(candidates[0]["rewards"][0][0])
+                              if candidates else 0.0),
+        }
         stats = dict(stats)
+        stats.update(sample)
         stats.update({
             "loss": loss,
             "num_samples": len(problems),
@@ -274,6 +281,11 @@ class Trainer:
                 total_samples += len(batch["problem"])
 
                 stats = self.rl_round(batch)
+                # per-round sample dump (reference
+                # distributed_trainer.py:297-299)
+                print(f"Sample problem: {stats['sample_problem'][:200]!r}")
+                print(f"Sample answer: {stats['sample_answer'][:200]!r}")
+                print(f"Sample reward: {stats['sample_reward']}")
                 self._cmd("save_adapter", self.lora_save_path)
 
                 if self.logger is not None:
