@@ -2,9 +2,12 @@
 // hot loop of the generation engine (reference counterpart: vLLM
 // paged_attention_v1/v2), written CDNA4-first:
 //
-//   grid = (num_seqs, n_kv_heads); one workgroup owns one (sequence,
-//   kv-head) pair and its whole group of query heads (GQA 7:1 on
-//   Qwen2.5-7B). 4 waves / 256 threads.
+//   grid = (num_seqs, n_kv_heads, context_slices); one workgroup owns one
+//   (sequence, kv-head, context-slice) triple and the kv-head's whole
+//   group of query heads (GQA 7:1 on Qwen2.5-7B). 4 waves / 256 threads.
+//   The flash-decoding context split (chosen to fill the chip at decode
+//   batch sizes) emits unnormalized fp32 partials + (m, l) per slice,
+//   combined by paged_decode_combine_kernel.
 //   Phase 1 (scores): each LANE owns one context token — it streams that
 //   token's K row with bf16x8 loads and keeps one fp32 partial per query
 //   head in registers (GROUP is a template param so the per-head array

@@ -89,3 +89,34 @@ def test_continuous_batching_admission(setup):
     expected0 = _naive_greedy(model, prompts[0], 6)
     assert results[0][0] == expected0
     assert small.pool.allocator.num_free == 40
+
+
+def test_prefill_token_budget_chunks(setup):
+    """Admission splits prefill into batches bounded by the token budget
+    without losing prompts."""
+    model, _ = setup
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    cfg = EngineConfig(max_seq_length=128, kv_block_size=8, num_kv_blocks=512,
+                       max_num_seqs=64)
+    engine = Engine(model, cfg, device=torch.device("cpu"), seed=0)
+    prompts = [[i + 1] * 20 for i in range(6)]
+    sp = SamplingParams(max_tokens=4, temperature=0.0, n=1)
+    results = engine.generate(prompts, sp, eos_token_id=None,
+                              prefill_token_budget=25)  # forces 1/batch
+    assert [len(r) for r in results] == [1] * 6
+    for p, r in zip(prompts, results):
+        assert r[0] == _naive_greedy(model, p, 4)
+
+
+def test_sampling_params_validation():
+    import pytest as _pytest
+    from distrl_llm_amd.config import SamplingParams
+    with _pytest.raises(ValueError):
+        SamplingParams(max_tokens=0)
+    with _pytest.raises(ValueError):
+        SamplingParams(top_p=0.0)
+    with _pytest.raises(ValueError):
+        SamplingParams(temperature=-1.0)
+    with _pytest.raises(ValueError):
+        SamplingParams(n=0)
