@@ -65,11 +65,12 @@ def build_worker(rank: int, world_size: int, config: Dict,
     if load_4bit:
         model.quantize_nf4_()
 
-    if config.get("load_adapter"):
+    if config.get("load_adapter") or config.get("resume"):
         # warm-start from a PEFT adapter directory (every rank loads the
-        # same file, so replicas stay identical)
+        # same file, so replicas stay identical); --resume implies loading
+        # the checkpoint's adapter
         from ..models.lora import load_adapter
-        load_adapter(model, config["load_adapter"])
+        load_adapter(model, config.get("resume") or config["load_adapter"])
 
     tokenizer = load_tokenizer(config["model"], spec.vocab_size)
 
@@ -83,6 +84,17 @@ def build_worker(rank: int, world_size: int, config: Dict,
         setattr(eng_cfg, k, v)
     engine = Engine(model, eng_cfg, device=device, seed=seed + 1000 * rank)
 
+    resume = config.get("resume")
+    if resume:
+        # restore this rank's sampling-RNG stream (bit-identical resume
+        # when the world size matches the checkpointing run; absent file
+        # = topology changed, continue with the fresh seeded stream)
+        es = os.path.join(resume, f"engine_state_rank{rank}.pt")
+        if os.path.exists(es):
+            st = torch.load(es, map_location="cpu", weights_only=True)
+            engine.generator.set_state(st["generator"])
+            engine._seq_counter = int(st["seq_counter"])
+
     learner = None
     if is_learner:
         learner = Learner(model, tokenizer, lr=config["lr"],
@@ -90,6 +102,12 @@ def build_worker(rank: int, world_size: int, config: Dict,
                           max_new_tokens=config["max_new_tokens"],
                           train_batch_size=config["train_batch_size"],
                           use_8bit_adam=config.get("use_8bit_adam", True))
+        if resume:
+            opt_path = os.path.join(resume, "optimizer_state.pt")
+            if os.path.exists(opt_path):
+                learner.load_state_dict(
+                    torch.load(opt_path, map_location="cpu",
+                               weights_only=True))
 
     fabric = Fabric(rank, world_size, num_actors, num_learners, device)
 

@@ -65,6 +65,14 @@ class Trainer:
         self.eval_sampling_params = SamplingParams(
             max_tokens=c["max_new_tokens"], temperature=0.6, top_p=0.95, n=8)
         self.eos_token_id = getattr(tokenizer, "eos_token_id", None)
+        # --resume <dir>: rank 0 restores counters + dataset RNG; the
+        # adapter/optimizer/engine states are restored in build_worker
+        self._resume_state = None
+        if config.get("resume") and fabric.rank == 0:
+            import torch
+            st = os.path.join(config["resume"], "trainer_state.pt")
+            if os.path.exists(st):
+                self._resume_state = torch.load(st, weights_only=False)
 
     # ----------------------------------------------------------- plumbing
 
@@ -114,10 +122,33 @@ class Trainer:
                 self._save_adapter(payload)
             return None
         if name == "save_checkpoint":
-            if self.fabric.rank == 0:
-                self._save_adapter(payload)
-            return None
+            return self._save_checkpoint_handler(payload)
         raise ValueError(f"unknown command {name!r}")
+
+    def _save_checkpoint_handler(self, payload):
+        """Full resumable checkpoint (extension over the reference's
+        adapter-only save, distributed_trainer.py:377-380 / SURVEY §5.4):
+        PEFT adapter + rank-0 trainer counters/dataset-RNG + lead-learner
+        optimizer state + per-rank engine RNG stream. With the same world
+        size, ``--resume <dir>`` continues the run bit-identically."""
+        import torch
+        path, meta = (payload if isinstance(payload, (tuple, list))
+                      else (payload, None))
+        os.makedirs(path, exist_ok=True)
+        if meta is not None:
+            torch.save({"generator": self.engine.generator.get_state(),
+                        "seq_counter": self.engine._seq_counter},
+                       os.path.join(path,
+                                    f"engine_state_rank{self.fabric.rank}.pt"))
+            if (self.fabric.is_learner
+                    and self.fabric.rank == self.fabric.learner_ranks[0]):
+                torch.save(self.learner.state_dict(),
+                           os.path.join(path, "optimizer_state.pt"))
+        if self.fabric.rank == 0:
+            self._save_adapter(path)
+            if meta is not None:
+                torch.save(meta, os.path.join(path, "trainer_state.pt"))
+        return None
 
     def _lora_params(self):
         model = self.engine.model
@@ -268,15 +299,25 @@ class Trainer:
         return stats
 
     def _train_rank0(self):
-        total_batch_steps = 0
-        total_samples = 0
+        rs = self._resume_state
+        total_batch_steps = rs["total_batch_steps"] if rs else 0
+        total_samples = rs["total_samples"] if rs else 0
+        start_episode = rs["episode"] if rs else 0
+        skip_batches = rs["batch_in_episode"] if rs else 0
+        if rs is not None:
+            # restore the dataset RNG to the start of the checkpointed
+            # episode so the re-shuffle reproduces the same batch order
+            self.train_dataset._rng.setstate(rs["dataset_rng"])
 
-        if self.eval_every > 0:
+        if self.eval_every > 0 and rs is None:
             self.evaluate(total_batch_steps)
 
-        for episode in range(self.episodes):
+        for episode in range(start_episode, self.episodes):
+            ep_rng_state = self.train_dataset._rng.getstate()
             dataset = self.train_dataset.shuffle()
-            for batch in dataset.iter(batch_size=self.batch_size):
+            for bi, batch in enumerate(dataset.iter(batch_size=self.batch_size)):
+                if bi < skip_batches:
+                    continue  # replayed prefix of a mid-episode resume
                 total_batch_steps += 1
                 total_samples += len(batch["problem"])
 
@@ -310,12 +351,22 @@ class Trainer:
                 if self.eval_every > 0 and total_batch_steps % self.eval_every == 0:
                     self.evaluate(total_batch_steps)
                 if self.save_every > 0 and total_batch_steps % self.save_every == 0:
+                    meta = {"episode": episode, "batch_in_episode": bi + 1,
+                            "total_batch_steps": total_batch_steps,
+                            "total_samples": total_samples,
+                            "dataset_rng": ep_rng_state}
                     self._cmd("save_checkpoint",
-                              os.path.join(self.run_directory,
-                                           f"model_{total_batch_steps}"))
+                              (os.path.join(self.run_directory,
+                                            f"model_{total_batch_steps}"),
+                               meta))
+            skip_batches = 0
+            meta = {"episode": episode + 1, "batch_in_episode": 0,
+                    "total_batch_steps": total_batch_steps,
+                    "total_samples": total_samples,
+                    "dataset_rng": self.train_dataset._rng.getstate()}
             self._cmd("save_checkpoint",
-                      os.path.join(self.run_directory,
-                                   f"model_{total_batch_steps}"))
+                      (os.path.join(self.run_directory,
+                                    f"model_{total_batch_steps}"), meta))
 
     # --------------------------------------------------------------- eval
 

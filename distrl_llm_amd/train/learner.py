@@ -49,6 +49,14 @@ class Learner:
         self.pad_token_id = (pad_token_id if pad_token_id is not None
                              else getattr(tokenizer, "pad_token_id", 0) or 0)
 
+    # -------------------------------------------------- checkpoint state
+
+    def state_dict(self):
+        return {"optimizer": self.optimizer.state_dict()}
+
+    def load_state_dict(self, sd):
+        self.optimizer.load_state_dict(sd["optimizer"])
+
     # ------------------------------------------------------- tokenization
 
     def _encode_batch(self, problems: Sequence[str], answers: Sequence[str]):

@@ -61,6 +61,31 @@ class Adam8bit(torch.optim.Optimizer):
                 self._step_torch(p, state, lr, b1, b2, eps, wd)
         return loss
 
+    def load_state_dict(self, state_dict):
+        """Dtype-preserving load. ``torch.optim.Optimizer.load_state_dict``
+        casts floating state tensors to the param dtype, which would
+        corrupt the fp32 block absmaxes when params are bf16 — map the
+        saved state by position instead and only move devices."""
+        saved_groups = state_dict["param_groups"]
+        groups = self.param_groups
+        if len(saved_groups) != len(groups) or any(
+                len(sg["params"]) != len(g["params"])
+                for sg, g in zip(saved_groups, groups)):
+            raise ValueError("loaded state dict has a different parameter layout")
+        id_map = {}
+        for sg, g in zip(saved_groups, groups):
+            for old_id, p in zip(sg["params"], g["params"]):
+                id_map[old_id] = p
+            for k, v in sg.items():
+                if k != "params":
+                    g[k] = v
+        self.state.clear()
+        for old_id, s in state_dict["state"].items():
+            p = id_map[old_id]
+            self.state[p] = {
+                k: (v.to(p.device) if torch.is_tensor(v) else v)
+                for k, v in s.items()}
+
     def _step_torch(self, p, state, lr, b1, b2, eps, wd):
         """Reference implementation of the quantized-state update (same
         math as the HIP kernel: dequant states -> Adam -> requant)."""

@@ -1,0 +1,179 @@
+"""Checkpoint/resume: full trainer-state save (adapter + optimizer +
+counters + dataset RNG + per-rank engine RNG) and bit-identical
+continuation — an extension over the reference's adapter-only
+save_checkpoint (reference distributed_trainer.py:377-380).
+"""
+
+import os
+import zlib
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+# --------------------------------------------------------- optimizer state
+
+
+def _fit_steps(opt, params, grads_per_step):
+    for grads in grads_per_step:
+        for p, g in zip(params, grads):
+            p.grad = g.clone()
+        opt.step()
+        for p in params:
+            p.grad = None
+
+
+def test_adam8bit_state_roundtrip():
+    """Save/load of Adam8bit state continues bit-identically and keeps
+    the fp32 block absmaxes fp32 (no dtype cast through load)."""
+    from distrl_llm_amd.train.optim import Adam8bit
+
+    torch.manual_seed(0)
+    pa = [torch.randn(300, requires_grad=True), torch.randn(7, 33, requires_grad=True)]
+    pb = [p.detach().clone().requires_grad_(True) for p in pa]
+    ga = [[torch.randn_like(p) for p in pa] for _ in range(5)]
+
+    oa = Adam8bit(pa, lr=1e-2)
+    ob = Adam8bit(pb, lr=1e-2)
+    _fit_steps(oa, pa, ga[:3])
+    _fit_steps(ob, pb, ga[:3])
+
+    sd = torch.load_state = oa.state_dict()
+    # round-trip through serialized bytes like a real checkpoint
+    import io
+    buf = io.BytesIO()
+    torch.save(sd, buf)
+    buf.seek(0)
+    ob.load_state_dict(torch.load(buf, weights_only=True))
+    for p in pb:
+        st = ob.state[p]
+        assert st["m_absmax"].dtype == torch.float32
+        assert st["v_absmax"].dtype == torch.float32
+        assert st["m_q"].dtype == torch.int8
+
+    _fit_steps(oa, pa, ga[3:])
+    _fit_steps(ob, pb, ga[3:])
+    for x, y in zip(pa, pb):
+        assert torch.equal(x, y)
+
+
+def test_adam8bit_load_rejects_mismatched_layout():
+    from distrl_llm_amd.train.optim import Adam8bit
+    pa = [torch.randn(10, requires_grad=True)]
+    pb = [torch.randn(10, requires_grad=True), torch.randn(4, requires_grad=True)]
+    oa, ob = Adam8bit(pa), Adam8bit(pb)
+    with pytest.raises(ValueError):
+        ob.load_state_dict(oa.state_dict())
+
+
+# -------------------------------------------------------- e2e gloo resume
+
+
+def _jitter_reward(answers, solutions):
+    """Reference rewards + deterministic per-answer jitter so random-init
+    training produces nonzero advantages (same trick as bench.py)."""
+    from distrl_llm_amd.rl.rewards import reward_function
+    r = reward_function(answers, solutions)
+    for i, a in enumerate(answers):
+        r[i, 0] += (zlib.crc32(a.encode()) % 1000) / 1000.0
+    return r
+
+
+def _ckpt_worker(rank, world_size, tmpdir, port, run_name, resume):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.chdir(tmpdir)
+
+    from distrl_llm_amd.parallel.worker import build_worker
+    from distrl_llm_amd.rl.data import (ListDataset, process_dataset,
+                                        r1_preprompt, synthetic_math_dataset)
+
+    config = {
+        "run_name": run_name,
+        "project_name": "test",
+        "lora_save_path": os.path.join(tmpdir, f"adapter_{run_name}"),
+        "lr": 1e-3,
+        "max_prompt_tokens": 48,
+        "max_new_tokens": 8,
+        "episodes": 1,
+        "num_candidates": 2,
+        "batch_size": 2,
+        "train_batch_size": 2,
+        "temperature": 1.0,
+        "save_every": 1,     # checkpoint after every round
+        "eval_every": 0,
+        "model": "tiny-qwen2",
+        "dataset": "synthetic",
+        "number_of_actors": 1,
+        "number_of_learners": 1,
+        "learner": "grpo",
+        "max_lora_rank": 4,
+        "topk": 2,
+        "learner_chunk_size": 2,
+        "actor_gpu_usage": 0.91,
+        "learner_gpu_usage": 0.35,
+        "lora_alpha": 8,
+        "lora_dropout": 0.0,
+        "seed": 11,
+        "use_8bit_adam": True,
+        "resume": resume,
+    }
+    rows = synthetic_math_dataset(8, seed=1)
+    rows = process_dataset(None, rows, r1_preprompt)
+    train = ListDataset(rows[:6], seed=1)  # 3 batches of 2 per episode
+    test = ListDataset(rows[6:], seed=1)
+    trainer = build_worker(
+        rank, world_size, config, train_dataset=train, test_dataset=test,
+        reward_function=_jitter_reward, device=torch.device("cpu"),
+        engine_overrides={"num_kv_blocks": 512, "kv_block_size": 8,
+                          "max_seq_length": 512})
+    trainer.train()
+    trainer.fabric.close()
+
+
+def _load_adapter_tensors(path):
+    from safetensors.torch import load_file
+    return load_file(os.path.join(path, "adapter_model.safetensors"))
+
+
+@pytest.mark.timeout(600)
+def test_mid_episode_resume_bit_identical(tmp_path):
+    """Run A: 3 rounds straight through. Run B: resume from run A's
+    round-1 checkpoint and finish rounds 2-3. Final adapters must be
+    bit-identical (optimizer, dataset order, counters and the per-rank
+    sampling-RNG streams all restored)."""
+    tmpdir = str(tmp_path)
+    port = 28500 + os.getpid() % 500
+    mp.spawn(_ckpt_worker, nprocs=2, args=(2, tmpdir, port, "a", None),
+             join=True)
+
+    run_a = os.path.join(tmpdir, "run_a")
+    ckpts = sorted(os.listdir(run_a), key=lambda d: int(d.split("_")[1]))
+    assert ckpts[0] == "model_1"
+    c1 = os.path.join(run_a, "model_1")
+    # full checkpoint contents
+    for f in ("adapter_model.safetensors", "adapter_config.json",
+              "trainer_state.pt", "optimizer_state.pt",
+              "engine_state_rank0.pt", "engine_state_rank1.pt"):
+        assert os.path.exists(os.path.join(c1, f)), f
+    meta = torch.load(os.path.join(c1, "trainer_state.pt"),
+                      weights_only=False)
+    assert meta["episode"] == 0 and meta["batch_in_episode"] == 1
+    assert meta["total_batch_steps"] == 1
+
+    mp.spawn(_ckpt_worker, nprocs=2, args=(2, tmpdir, port + 7, "b", c1),
+             join=True)
+
+    final_a = _load_adapter_tensors(os.path.join(run_a, "model_3"))
+    final_b = _load_adapter_tensors(
+        os.path.join(tmpdir, "run_b", "model_3"))
+    assert final_a.keys() == final_b.keys()
+    moved = False
+    for k in final_a:
+        assert torch.equal(final_a[k], final_b[k]), k
+        if "lora_B" in k and final_a[k].abs().sum() > 0:
+            moved = True
+    # guard against the degenerate all-zero-advantage case: training must
+    # actually have updated the adapter for the comparison to mean anything
+    assert moved

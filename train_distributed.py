@@ -54,6 +54,11 @@ def parse_args():
     args.add_argument("--load_adapter", type=str, default=None,
                       help="Warm-start: PEFT adapter directory to load into "
                            "every worker before training")
+    args.add_argument("--resume", type=str, default=None,
+                      help="Resume from a checkpoint directory written by the "
+                           "save cadence (run_<name>/model_<step>): restores "
+                           "adapter, optimizer, trainer counters, dataset "
+                           "order and per-rank RNG streams")
     args.add_argument("--seed", type=int, default=3407)
     args.add_argument("--backend_device", type=str, default="auto",
                       choices=["auto", "cuda", "cpu"])
@@ -90,6 +95,7 @@ def build_config(args) -> dict:
         "lora_dropout": args.lora_dropout,
         "seed": args.seed,
         "load_adapter": args.load_adapter,
+        "resume": args.resume,
     }
 
 
