@@ -1,0 +1,153 @@
+"""Minimal OpenAI-style serving surface over the generation engine.
+
+The reference has no serving API (Engine use is in-process only —
+SURVEY.md §2.3 / docs/ROADMAP.md #10); this is a native extension so the
+paged-KV engine can also be used as a standalone completion server:
+
+    python -m distrl_llm_amd.serve --model qwen2.5-7b --port 8000
+
+Endpoints (OpenAI completions-compatible subset):
+    GET  /health           liveness
+    GET  /v1/models        the single served model
+    POST /v1/completions   prompt(s) -> n sampled completions
+
+One engine, one request at a time (the engine batches *within* a request
+via its continuous-batching admission loop; cross-request batching is a
+scheduler, not a serving-surface, concern). Handlers are sync ``def`` so
+FastAPI runs them in its threadpool; an explicit lock serializes engine
+access.
+"""
+
+# NOTE: no `from __future__ import annotations` here — FastAPI resolves
+# endpoint annotations by name at request time, and CompletionRequest is
+# local to create_app (postponed annotations would break body binding).
+import threading
+import time
+from typing import List, Optional, Union
+
+from .config import SamplingParams
+
+
+def create_app(engine, tokenizer, model_name: str):
+    from fastapi import FastAPI, HTTPException
+    from pydantic import BaseModel
+
+    class CompletionRequest(BaseModel):
+        prompt: Union[str, List[str]]
+        model: Optional[str] = None
+        max_tokens: int = 16
+        temperature: float = 1.0
+        top_p: float = 1.0
+        top_k: int = 0
+        n: int = 1
+        echo: bool = False
+        stop: Optional[Union[str, List[str]]] = None
+
+    app = FastAPI(title="distrl-mi355x", version="0.1")
+    lock = threading.Lock()
+    created = int(time.time())
+
+    @app.get("/health")
+    def health():
+        return {"status": "ok"}
+
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model",
+                          "created": created, "owned_by": "distrl-mi355x"}]}
+
+    @app.post("/v1/completions")
+    def completions(req: CompletionRequest):
+        prompts = [req.prompt] if isinstance(req.prompt, str) else list(req.prompt)
+        if not prompts:
+            raise HTTPException(status_code=400, detail="empty prompt")
+        stops = ([req.stop] if isinstance(req.stop, str) else req.stop) or []
+        try:
+            sp = SamplingParams(max_tokens=req.max_tokens,
+                                temperature=req.temperature,
+                                top_p=req.top_p, top_k=req.top_k, n=req.n)
+        except ValueError as e:
+            raise HTTPException(status_code=400, detail=str(e))
+        prompt_ids = [tokenizer.encode(p) for p in prompts]
+        eos = getattr(tokenizer, "eos_token_id", None)
+        with lock:
+            outs = engine.generate(prompt_ids, sp, eos_token_id=eos)
+
+        choices, completion_tokens = [], 0
+        for pi, (p, ids, per_prompt) in enumerate(zip(prompts, prompt_ids, outs)):
+            for ci, out_ids in enumerate(per_prompt):
+                completion_tokens += len(out_ids)
+                finish = "stop" if (eos is not None and out_ids
+                                    and out_ids[-1] == eos) else "length"
+                vocab = getattr(tokenizer, "vocab_size", None)
+                keep = [t for t in out_ids
+                        if vocab is None or t < vocab or t > 260]
+                text = tokenizer.decode(keep, skip_special_tokens=True)
+                for s in stops:
+                    cut = text.find(s)
+                    if cut >= 0:
+                        text, finish = text[:cut], "stop"
+                choices.append({"index": pi * sp.n + ci,
+                                "text": (p + text) if req.echo else text,
+                                "finish_reason": finish,
+                                "logprobs": None})
+        prompt_tokens = sum(len(ids) for ids in prompt_ids)
+        return {"id": f"cmpl-{created}-{int(time.time() * 1e6) & 0xFFFFFF:x}",
+                "object": "text_completion",
+                "created": int(time.time()),
+                "model": model_name,
+                "choices": choices,
+                "usage": {"prompt_tokens": prompt_tokens,
+                          "completion_tokens": completion_tokens,
+                          "total_tokens": prompt_tokens + completion_tokens}}
+
+    return app
+
+
+def main():
+    import argparse
+
+    import torch
+
+    from .config import EngineConfig
+    from .engine.engine import Engine
+    from .models.model import CausalLM
+    from .models.spec import get_spec, is_4bit_model_name
+    from .utils.tokenizer import load_tokenizer
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", type=str, default="qwen2.5-7b")
+    ap.add_argument("--host", type=str, default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--max-seq-length", type=int, default=4096)
+    ap.add_argument("--gpu-memory-utilization", type=float, default=0.9)
+    ap.add_argument("--adapter", type=str, default=None,
+                    help="PEFT adapter directory to apply before serving")
+    args = ap.parse_args()
+
+    device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
+    dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    spec = get_spec(args.model)
+    model = CausalLM(spec, lora_r=32 if args.adapter else 0, lora_alpha=16,
+                     dtype=dtype, device=device)
+    model.random_init(args.seed)
+    if is_4bit_model_name(args.model):
+        model.quantize_nf4_()
+    if args.adapter:
+        from .models.lora import load_adapter
+        load_adapter(model, args.adapter)
+    tokenizer = load_tokenizer(args.model, spec.vocab_size)
+    engine = Engine(model, EngineConfig(
+        max_seq_length=args.max_seq_length,
+        gpu_memory_utilization=args.gpu_memory_utilization),
+        device=device, seed=args.seed)
+
+    import uvicorn
+    uvicorn.run(create_app(engine, tokenizer, args.model),
+                host=args.host, port=args.port)
+
+
+if __name__ == "__main__":
+    main()

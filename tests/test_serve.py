@@ -1,0 +1,69 @@
+"""Serving surface: OpenAI-style /v1/completions over the engine (native
+extension — the reference is in-process only, docs/ROADMAP.md #10)."""
+
+import pytest
+import torch
+
+
+@pytest.fixture(scope="module")
+def client():
+    from fastapi.testclient import TestClient
+
+    from distrl_llm_amd.config import EngineConfig
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    from distrl_llm_amd.serve import create_app
+    from distrl_llm_amd.utils.tokenizer import ByteTokenizer
+
+    spec = get_spec("tiny-qwen2")
+    model = CausalLM(spec, lora_r=0, dtype=torch.float32).random_init(5)
+    engine = Engine(model, EngineConfig(max_seq_length=128, kv_block_size=8,
+                                        num_kv_blocks=256, max_num_seqs=32),
+                    device=torch.device("cpu"), seed=0)
+    tok = ByteTokenizer(vocab_size=spec.vocab_size)
+    return TestClient(create_app(engine, tok, "tiny-qwen2"))
+
+
+def test_health_and_models(client):
+    assert client.get("/health").json() == {"status": "ok"}
+    models = client.get("/v1/models").json()
+    assert models["data"][0]["id"] == "tiny-qwen2"
+
+
+def test_completion_single(client):
+    r = client.post("/v1/completions", json={
+        "prompt": "2+2=", "max_tokens": 6, "temperature": 0.0})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "text_completion"
+    assert len(body["choices"]) == 1
+    c = body["choices"][0]
+    assert isinstance(c["text"], str) and c["finish_reason"] in ("stop", "length")
+    assert body["usage"]["prompt_tokens"] == 4  # small vocab -> byte mode
+    assert body["usage"]["total_tokens"] == (body["usage"]["prompt_tokens"]
+                                             + body["usage"]["completion_tokens"])
+    assert body["usage"]["completion_tokens"] <= 6
+    # greedy decoding is deterministic
+    r2 = client.post("/v1/completions", json={
+        "prompt": "2+2=", "max_tokens": 6, "temperature": 0.0})
+    assert r2.json()["choices"][0]["text"] == c["text"]
+
+
+def test_completion_batch_and_n(client):
+    r = client.post("/v1/completions", json={
+        "prompt": ["abcd", "wxyz"], "max_tokens": 4, "temperature": 1.0,
+        "top_p": 0.9, "n": 3})
+    body = r.json()
+    assert len(body["choices"]) == 6
+    assert [c["index"] for c in body["choices"]] == list(range(6))
+
+
+def test_completion_echo_and_validation(client):
+    r = client.post("/v1/completions", json={
+        "prompt": "hi", "max_tokens": 2, "temperature": 0.0, "echo": True})
+    assert r.json()["choices"][0]["text"].startswith("hi")
+    # invalid sampling params -> 400, not a crash
+    r = client.post("/v1/completions", json={"prompt": "x", "max_tokens": 0})
+    assert r.status_code == 400
+    r = client.post("/v1/completions", json={"prompt": [], "max_tokens": 2})
+    assert r.status_code == 400
