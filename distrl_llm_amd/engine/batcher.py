@@ -1,0 +1,122 @@
+"""Cross-request dynamic batching for the serving surface.
+
+The engine batches *within* one ``generate`` call (continuous-batching
+admission, n-candidate fan-out). For serving, concurrent requests should
+share one decode wave too: each ``generate`` call refreshes the fused
+decode weights and (on GPU) captures/replays a hipGraph sized to the
+batch, so 16 one-prompt calls cost ~16x one 16-prompt call. The reference
+has no serving surface at all (docs/ROADMAP.md #10); this is native.
+
+One background thread owns the engine. ``submit`` blocks the calling
+(FastAPI threadpool) thread until its slice of the merged batch is done.
+Requests are grouped by identical (SamplingParams, eos) since the engine
+takes one SamplingParams per call; different groups run back-to-back in
+arrival order.
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+import time
+from typing import List, Optional
+
+from ..config import SamplingParams
+
+_SHUTDOWN = object()
+
+
+class _Request:
+    __slots__ = ("prompts", "sp", "eos", "event", "result", "error")
+
+    def __init__(self, prompts, sp, eos):
+        self.prompts = prompts
+        self.sp = sp
+        self.eos = eos
+        self.event = threading.Event()
+        self.result = None
+        self.error = None
+
+
+class DynamicBatcher:
+    """Thread-owning wrapper that merges concurrent generate calls.
+
+    max_wait_ms: after the first waiting request is picked up, how long
+    to keep draining the queue for co-batchable requests before running.
+    Zero still batches whatever has already queued up while the previous
+    wave was running (the common steady-state case).
+    """
+
+    def __init__(self, engine, max_wait_ms: float = 2.0):
+        self.engine = engine
+        self.max_wait_ms = max_wait_ms
+        self.calls = 0  # engine.generate invocations (for tests/metrics)
+        self._q: "queue.Queue" = queue.Queue()
+        self._thread = threading.Thread(target=self._loop, daemon=True,
+                                        name="distrl-batcher")
+        self._thread.start()
+
+    # ---------------------------------------------------------------- API
+
+    def submit(self, prompts: List[List[int]], sp: SamplingParams,
+               eos_token_id: Optional[int] = None) -> List[List[List[int]]]:
+        """Blocking: returns the engine.generate result for ``prompts``."""
+        req = _Request(prompts, sp, eos_token_id)
+        self._q.put(req)
+        req.event.wait()
+        if req.error is not None:
+            raise req.error
+        return req.result
+
+    def close(self):
+        self._q.put(_SHUTDOWN)
+        self._thread.join(timeout=10)
+
+    # --------------------------------------------------------------- loop
+
+    @staticmethod
+    def _key(req: _Request):
+        sp = req.sp
+        return (sp.max_tokens, sp.temperature, sp.top_p, sp.top_k, sp.n,
+                req.eos)
+
+    def _loop(self):
+        shutdown = False
+        while not shutdown:
+            first = self._q.get()
+            if first is _SHUTDOWN:
+                return
+            batch = [first]
+            deadline = time.monotonic() + self.max_wait_ms / 1e3
+            while True:
+                timeout = deadline - time.monotonic()
+                try:
+                    nxt = self._q.get(timeout=max(timeout, 0.0))
+                except queue.Empty:
+                    break
+                if nxt is _SHUTDOWN:
+                    shutdown = True
+                    break
+                batch.append(nxt)
+            groups: dict = {}
+            for req in batch:
+                groups.setdefault(self._key(req), []).append(req)
+            for reqs in groups.values():
+                self._run_group(reqs)
+
+    def _run_group(self, reqs: List[_Request]):
+        merged = [p for r in reqs for p in r.prompts]
+        try:
+            self.calls += 1
+            outs = self.engine.generate(merged, reqs[0].sp,
+                                        eos_token_id=reqs[0].eos)
+            off = 0
+            for r in reqs:
+                r.result = outs[off:off + len(r.prompts)]
+                off += len(r.prompts)
+        except Exception as e:  # deliver the failure to every waiter
+            for r in reqs:
+                r.error = e
+        finally:
+            for r in reqs:
+                r.event.set()

@@ -391,6 +391,13 @@ class Engine:
         try:
             return self._generate_inner(prompts, sp, eos_token_id,
                                         prefill_token_budget)
+        except Exception:
+            # a failure mid-generation strands this call's in-flight
+            # sequences' KV blocks; generate calls are serialized, so no
+            # other sequence is live — reset the pool instead of leaking
+            # (long-lived serving processes would otherwise exhaust it)
+            self.pool.allocator.reset()
+            raise
         finally:
             if was_training:
                 self.model.train()

@@ -47,6 +47,14 @@ class BlockAllocator:
     def refcount(self, block: int) -> int:
         return self._ref[block]
 
+    def reset(self) -> None:
+        """Return every block to the free list, dropping all refcounts.
+        Only valid when no live sequence remains (error recovery after a
+        failed generation — generate calls are serialized, so a failure
+        strands every in-flight sequence of that call and nothing else)."""
+        self._free = list(range(self.num_blocks - 1, -1, -1))
+        self._ref = [0] * self.num_blocks
+
 
 class KVCachePool:
     """Per-layer paged K/V tensors.

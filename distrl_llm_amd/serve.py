@@ -11,24 +11,25 @@ Endpoints (OpenAI completions-compatible subset):
     GET  /v1/models        the single served model
     POST /v1/completions   prompt(s) -> n sampled completions
 
-One engine, one request at a time (the engine batches *within* a request
-via its continuous-batching admission loop; cross-request batching is a
-scheduler, not a serving-surface, concern). Handlers are sync ``def`` so
-FastAPI runs them in its threadpool; an explicit lock serializes engine
-access.
+Handlers are sync ``def`` so FastAPI runs them in its threadpool;
+concurrent requests are merged into shared decode waves by
+``engine.batcher.DynamicBatcher`` (requests with identical sampling
+params co-batch into one ``engine.generate`` call — one fused-weight
+refresh and one hipGraph wave instead of one per request).
 """
 
 # NOTE: no `from __future__ import annotations` here — FastAPI resolves
 # endpoint annotations by name at request time, and CompletionRequest is
 # local to create_app (postponed annotations would break body binding).
-import threading
 import time
 from typing import List, Optional, Union
 
 from .config import SamplingParams
+from .engine.batcher import DynamicBatcher
 
 
-def create_app(engine, tokenizer, model_name: str):
+def create_app(engine, tokenizer, model_name: str,
+               batch_wait_ms: float = 2.0):
     from fastapi import FastAPI, HTTPException
     from pydantic import BaseModel
 
@@ -44,7 +45,8 @@ def create_app(engine, tokenizer, model_name: str):
         stop: Optional[Union[str, List[str]]] = None
 
     app = FastAPI(title="distrl-mi355x", version="0.1")
-    lock = threading.Lock()
+    batcher = DynamicBatcher(engine, max_wait_ms=batch_wait_ms)
+    app.state.batcher = batcher
     created = int(time.time())
 
     @app.get("/health")
@@ -71,8 +73,7 @@ def create_app(engine, tokenizer, model_name: str):
             raise HTTPException(status_code=400, detail=str(e))
         prompt_ids = [tokenizer.encode(p) for p in prompts]
         eos = getattr(tokenizer, "eos_token_id", None)
-        with lock:
-            outs = engine.generate(prompt_ids, sp, eos_token_id=eos)
+        outs = batcher.submit(prompt_ids, sp, eos_token_id=eos)
 
         choices, completion_tokens = [], 0
         for pi, (p, ids, per_prompt) in enumerate(zip(prompts, prompt_ids, outs)):
