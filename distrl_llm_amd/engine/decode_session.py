@@ -87,12 +87,23 @@ class DecodeSession:
         logits = e._decode_forward(self.tokens, pos, slot, self.block_tables,
                                    self.ctx_lens)
         if self.sp.temperature > 0.0:
-            from ..ops.build import get_extension
-            ext = get_extension()
-            sampled = ext.sample_tokens(logits, float(self.sp.temperature),
-                                        float(self.sp.top_p),
-                                        int(self.sp.top_k), self.seeds,
-                                        self.step_idx)
+            if logits.is_cuda:
+                from ..ops.build import get_extension
+                ext = get_extension()
+                sampled = ext.sample_tokens(logits,
+                                            float(self.sp.temperature),
+                                            float(self.sp.top_p),
+                                            int(self.sp.top_k), self.seeds,
+                                            self.step_idx)
+            else:
+                # CPU session (DISTRL_FORCE_SESSION=1 CI coverage of the
+                # state machine): reference sampler on the engine stream
+                from ..ops import functional as OF
+                sampled = OF.sample_tokens(logits,
+                                           float(self.sp.temperature),
+                                           float(self.sp.top_p),
+                                           int(self.sp.top_k),
+                                           generator=e.generator)
         else:
             sampled = logits.argmax(-1)
         self.out_buf.index_copy_(0, self.step_idx, sampled.unsqueeze(0))

@@ -16,6 +16,7 @@ broadcast into these tensors.
 
 from __future__ import annotations
 
+import os
 import time
 from typing import List, Optional
 
@@ -469,8 +470,12 @@ class Engine:
 
         try_admit()
 
-        if self.device.type == "cuda" and not self.cfg.enforce_eager:
-            # wave-based decode sessions (device state + hipGraph replay)
+        force_session = os.environ.get("DISTRL_FORCE_SESSION") == "1"
+        if ((self.device.type == "cuda" and not self.cfg.enforce_eager)
+                or force_session):
+            # wave-based decode sessions (device state + hipGraph replay;
+            # DISTRL_FORCE_SESSION=1 runs the same state machine on CPU
+            # so CI covers the graph step body without a GPU)
             from .decode_session import DecodeSession
             while running or waiting:
                 if not running:
