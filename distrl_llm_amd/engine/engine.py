@@ -28,6 +28,7 @@ from ..models.model import CausalLM
 from ..ops import functional as OF
 from ..ops import reference as R
 from .kvcache import KVCachePool, Sequence
+from ..utils.trace import trace_range
 
 
 class Engine:
@@ -451,7 +452,8 @@ class Engine:
                 batch_tokens += L
             if not batch:
                 return
-            logits = self._prefill_batch(batch)
+            with trace_range("engine/prefill"):
+                logits = self._prefill_batch(batch)
             for i, parent in enumerate(batch):
                 lg = logits[i:i + 1].expand(sp.n, -1).contiguous()
                 first = OF.sample_tokens(lg, sp.temperature, sp.top_p, sp.top_k,
@@ -486,7 +488,8 @@ class Engine:
                                 "KV pool too small to admit any waiting prompt")
                         break
                 session = DecodeSession(self, running, sp, eos_token_id)
-                outs = session.run()
+                with trace_range(f"engine/decode_wave[{len(running)}]"):
+                    outs = session.run()
                 for q, ids in zip(running, outs):
                     results[q.parent_prompt].append(ids)
                     self._finish(q)

@@ -29,6 +29,7 @@ import numpy as np
 from ..config import SamplingParams
 from ..models import lora as lora_io
 from ..parallel.fabric import Fabric
+from ..utils.trace import trace_range
 from .advantage import even_chunk_sizes, merge_candidates, process_candidates
 from .sched import split_dict_lists, worker_chunk_sizes
 
@@ -107,7 +108,8 @@ class Trainer:
             return self._update_handler(payload)
         if name == "sync_weights":
             params = self._lora_params()
-            self.fabric.broadcast_lora(params)
+            with trace_range("rl/sync_weights"):
+                self.fabric.broadcast_lora(params)
             return None
         if name == "barrier":
             import torch
@@ -181,7 +183,8 @@ class Trainer:
         my_task = chunks[self.fabric.rank]
         result = None
         if len(my_task["problem"]) > 0:
-            result = self._generate_task(my_task, sp)
+            with trace_range("rl/generate"):
+                result = self._generate_task(my_task, sp)
         gathered = self.fabric.gather_obj(result, dst=0)
         if self.fabric.rank != 0:
             return None
@@ -207,11 +210,12 @@ class Trainer:
         """Rank 0: per-group reward arrays (reference
         distributed_trainer.py:205-219)."""
         t0 = time.time()
-        for cand in candidates:
-            rewards = []
-            for answers, solutions in zip(cand["answers"], cand["solution"]):
-                rewards.append(self.reward_function(answers, solutions))
-            cand["rewards"] = rewards
+        with trace_range("rl/reward"):
+            for cand in candidates:
+                rewards = []
+                for answers, solutions in zip(cand["answers"], cand["solution"]):
+                    rewards.append(self.reward_function(answers, solutions))
+                cand["rewards"] = rewards
         return time.time() - t0
 
     # ------------------------------------------------------------- update
@@ -223,13 +227,14 @@ class Trainer:
         loss = None
         if self.fabric.is_learner:
             problems, answers, rewards = chunks[self.fabric.learner_index]
-            if len(problems) > 0:
-                loss = self.learner.accumulate_gradients(problems, answers,
-                                                         rewards)
-            else:
-                loss = 0.0
-            self.fabric.allreduce_mean_grads(self.learner.params)
-            self.learner.step()
+            with trace_range("rl/update"):
+                if len(problems) > 0:
+                    loss = self.learner.accumulate_gradients(problems, answers,
+                                                             rewards)
+                else:
+                    loss = 0.0
+                self.fabric.allreduce_mean_grads(self.learner.params)
+                self.learner.step()
             loss = self.fabric.allreduce_mean_scalar(loss)
         gathered = self.fabric.gather_obj(loss, dst=0)
         if self.fabric.rank != 0:
