@@ -19,9 +19,19 @@ _ENABLED = None
 
 
 def _enabled() -> bool:
+    """Markers are pure observability — if roctx is unavailable in this
+    torch build they self-disable rather than fail the compute path."""
     global _ENABLED
     if _ENABLED is None:
-        _ENABLED = torch.cuda.is_available()
+        if not torch.cuda.is_available():
+            _ENABLED = False
+        else:
+            try:
+                torch.cuda.nvtx.range_push("distrl/trace_probe")
+                torch.cuda.nvtx.range_pop()
+                _ENABLED = True
+            except Exception:
+                _ENABLED = False
     return _ENABLED
 
 
@@ -29,11 +39,18 @@ def _enabled() -> bool:
 def trace_range(name: str):
     """Context manager emitting a roctx range around the enclosed work."""
     if _enabled():
-        torch.cuda.nvtx.range_push(name)
+        try:
+            torch.cuda.nvtx.range_push(name)
+        except Exception:
+            yield
+            return
         try:
             yield
         finally:
-            torch.cuda.nvtx.range_pop()
+            try:
+                torch.cuda.nvtx.range_pop()
+            except Exception:
+                pass
     else:
         yield
 
@@ -41,4 +58,7 @@ def trace_range(name: str):
 def trace_mark(name: str) -> None:
     """Instantaneous roctx marker."""
     if _enabled():
-        torch.cuda.nvtx.mark(name)
+        try:
+            torch.cuda.nvtx.mark(name)
+        except Exception:
+            pass
