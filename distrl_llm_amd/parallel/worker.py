@@ -50,7 +50,9 @@ def build_worker(rank: int, world_size: int, config: Dict,
         else:
             device = torch.device("cpu")
 
-    spec = get_spec(config["model"])
+    from ..models.hf_io import (is_hf_checkpoint_dir, load_hf_checkpoint,
+                                resolve_spec)
+    spec = resolve_spec(config["model"])
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     seed = int(config.get("seed", 3407))
 
@@ -59,6 +61,11 @@ def build_worker(rank: int, world_size: int, config: Dict,
                      lora_dropout=config["lora_dropout"],
                      dtype=dtype, device=device)
     model.random_init(seed)  # identical on every rank (replicated base + LoRA)
+    if is_hf_checkpoint_dir(config["model"]):
+        # local pretrained checkpoint (the reference's from_pretrained
+        # path, reference distributed_actor.py:58-66) — loaded before any
+        # nf4 quantization
+        load_hf_checkpoint(model, config["model"])
     load_4bit = config.get("load_in_4bit")
     if load_4bit is None:
         load_4bit = is_4bit_model_name(config["model"])

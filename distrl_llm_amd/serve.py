@@ -113,8 +113,10 @@ def main():
 
     from .config import EngineConfig
     from .engine.engine import Engine
+    from .models.hf_io import (is_hf_checkpoint_dir, load_hf_checkpoint,
+                               resolve_spec)
     from .models.model import CausalLM
-    from .models.spec import get_spec, is_4bit_model_name
+    from .models.spec import is_4bit_model_name
     from .utils.tokenizer import load_tokenizer
 
     ap = argparse.ArgumentParser()
@@ -130,10 +132,12 @@ def main():
 
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
-    spec = get_spec(args.model)
+    spec = resolve_spec(args.model)
     model = CausalLM(spec, lora_r=32 if args.adapter else 0, lora_alpha=16,
                      dtype=dtype, device=device)
     model.random_init(args.seed)
+    if is_hf_checkpoint_dir(args.model):
+        load_hf_checkpoint(model, args.model)
     if is_4bit_model_name(args.model):
         model.quantize_nf4_()
     if args.adapter:

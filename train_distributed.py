@@ -139,9 +139,9 @@ def main():
     config = build_config(args)
     world_size = args.number_of_actors + args.number_of_learners
 
-    from distrl_llm_amd.models.spec import get_spec
+    from distrl_llm_amd.models.hf_io import resolve_spec
     from distrl_llm_amd.utils.tokenizer import load_tokenizer
-    tokenizer = load_tokenizer(args.model, get_spec(args.model).vocab_size)
+    tokenizer = load_tokenizer(args.model, resolve_spec(args.model).vocab_size)
     train_ds, test_ds = load_datasets(args, tokenizer)
     print(f"\nNumber of train samples: {len(train_ds)}\n")
     print(f"Number of test samples: {len(test_ds)}\n")
