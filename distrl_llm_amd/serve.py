@@ -44,6 +44,20 @@ def create_app(engine, tokenizer, model_name: str,
         echo: bool = False
         stop: Optional[Union[str, List[str]]] = None
 
+    class ChatMessage(BaseModel):
+        role: str
+        content: str
+
+    class ChatRequest(BaseModel):
+        messages: List[ChatMessage]
+        model: Optional[str] = None
+        max_tokens: int = 16
+        temperature: float = 1.0
+        top_p: float = 1.0
+        top_k: int = 0
+        n: int = 1
+        stop: Optional[Union[str, List[str]]] = None
+
     app = FastAPI(title="distrl-mi355x", version="0.1")
     batcher = DynamicBatcher(engine, max_wait_ms=batch_wait_ms)
     app.state.batcher = batcher
@@ -59,29 +73,27 @@ def create_app(engine, tokenizer, model_name: str,
                 "data": [{"id": model_name, "object": "model",
                           "created": created, "owned_by": "distrl-mi355x"}]}
 
-    @app.post("/v1/completions")
-    def completions(req: CompletionRequest):
-        prompts = [req.prompt] if isinstance(req.prompt, str) else list(req.prompt)
-        if not prompts:
-            raise HTTPException(status_code=400, detail="empty prompt")
-        stops = ([req.stop] if isinstance(req.stop, str) else req.stop) or []
+    def _sp_or_400(req):
         try:
-            sp = SamplingParams(max_tokens=req.max_tokens,
-                                temperature=req.temperature,
-                                top_p=req.top_p, top_k=req.top_k, n=req.n)
+            return SamplingParams(max_tokens=req.max_tokens,
+                                  temperature=req.temperature,
+                                  top_p=req.top_p, top_k=req.top_k, n=req.n)
         except ValueError as e:
             raise HTTPException(status_code=400, detail=str(e))
+
+    def _complete(prompts, sp, stop):
+        """Shared generation core: returns (flat [(text, finish)], usage)."""
+        stops = ([stop] if isinstance(stop, str) else stop) or []
         prompt_ids = [tokenizer.encode(p) for p in prompts]
         eos = getattr(tokenizer, "eos_token_id", None)
         outs = batcher.submit(prompt_ids, sp, eos_token_id=eos)
-
-        choices, completion_tokens = [], 0
-        for pi, (p, ids, per_prompt) in enumerate(zip(prompts, prompt_ids, outs)):
-            for ci, out_ids in enumerate(per_prompt):
+        flat, completion_tokens = [], 0
+        vocab = getattr(tokenizer, "vocab_size", None)
+        for per_prompt in outs:
+            for out_ids in per_prompt:
                 completion_tokens += len(out_ids)
                 finish = "stop" if (eos is not None and out_ids
                                     and out_ids[-1] == eos) else "length"
-                vocab = getattr(tokenizer, "vocab_size", None)
                 keep = [t for t in out_ids
                         if vocab is None or t < vocab or t > 260]
                 text = tokenizer.decode(keep, skip_special_tokens=True)
@@ -89,19 +101,50 @@ def create_app(engine, tokenizer, model_name: str,
                     cut = text.find(s)
                     if cut >= 0:
                         text, finish = text[:cut], "stop"
-                choices.append({"index": pi * sp.n + ci,
-                                "text": (p + text) if req.echo else text,
-                                "finish_reason": finish,
-                                "logprobs": None})
+                flat.append((text, finish))
         prompt_tokens = sum(len(ids) for ids in prompt_ids)
+        usage = {"prompt_tokens": prompt_tokens,
+                 "completion_tokens": completion_tokens,
+                 "total_tokens": prompt_tokens + completion_tokens}
+        return flat, usage
+
+    @app.post("/v1/completions")
+    def completions(req: CompletionRequest):
+        prompts = [req.prompt] if isinstance(req.prompt, str) else list(req.prompt)
+        if not prompts:
+            raise HTTPException(status_code=400, detail="empty prompt")
+        sp = _sp_or_400(req)
+        flat, usage = _complete(prompts, sp, req.stop)
+        choices = [{"index": i,
+                    "text": (prompts[i // sp.n] + text) if req.echo else text,
+                    "finish_reason": finish, "logprobs": None}
+                   for i, (text, finish) in enumerate(flat)]
         return {"id": f"cmpl-{created}-{int(time.time() * 1e6) & 0xFFFFFF:x}",
                 "object": "text_completion",
                 "created": int(time.time()),
                 "model": model_name,
                 "choices": choices,
-                "usage": {"prompt_tokens": prompt_tokens,
-                          "completion_tokens": completion_tokens,
-                          "total_tokens": prompt_tokens + completion_tokens}}
+                "usage": usage}
+
+    @app.post("/v1/chat/completions")
+    def chat_completions(req: ChatRequest):
+        if not req.messages:
+            raise HTTPException(status_code=400, detail="empty messages")
+        sp = _sp_or_400(req)
+        from .rl.data import apply_template
+        prompt = apply_template(tokenizer,
+                                [m.model_dump() for m in req.messages])
+        flat, usage = _complete([prompt], sp, req.stop)
+        choices = [{"index": i,
+                    "message": {"role": "assistant", "content": text},
+                    "finish_reason": finish}
+                   for i, (text, finish) in enumerate(flat)]
+        return {"id": f"chatcmpl-{created}-{int(time.time() * 1e6) & 0xFFFFFF:x}",
+                "object": "chat.completion",
+                "created": int(time.time()),
+                "model": model_name,
+                "choices": choices,
+                "usage": usage}
 
     return app
 

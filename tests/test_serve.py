@@ -58,6 +58,23 @@ def test_completion_batch_and_n(client):
     assert [c["index"] for c in body["choices"]] == list(range(6))
 
 
+def test_chat_completion(client):
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "system", "content": "Be brief."},
+                     {"role": "user", "content": "hi"}],
+        "max_tokens": 4, "temperature": 0.0})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    c = body["choices"][0]
+    assert c["message"]["role"] == "assistant"
+    assert isinstance(c["message"]["content"], str)
+    assert body["usage"]["prompt_tokens"] > 10  # chat template applied
+    r2 = client.post("/v1/chat/completions", json={
+        "messages": [], "max_tokens": 4})
+    assert r2.status_code == 400
+
+
 def test_completion_echo_and_validation(client):
     r = client.post("/v1/completions", json={
         "prompt": "hi", "max_tokens": 2, "temperature": 0.0, "echo": True})
