@@ -63,9 +63,35 @@ def create_app(engine, tokenizer, model_name: str,
     app.state.batcher = batcher
     created = int(time.time())
 
+    # Prometheus observability (per-process registry so tests and
+    # multiple apps don't collide on the global default registry)
+    from prometheus_client import (CollectorRegistry, Counter, Gauge,
+                                   Histogram, generate_latest)
+    registry = CollectorRegistry()
+    m_requests = Counter("distrl_requests_total", "completion requests",
+                         ["endpoint", "status"], registry=registry)
+    m_latency = Histogram("distrl_request_seconds", "request latency",
+                          ["endpoint"], registry=registry,
+                          buckets=(.05, .1, .25, .5, 1, 2.5, 5, 10, 30, 60))
+    m_prompt_toks = Counter("distrl_prompt_tokens_total",
+                            "prompt tokens consumed", registry=registry)
+    m_gen_toks = Counter("distrl_generated_tokens_total",
+                         "tokens generated", registry=registry)
+    m_engine_calls = Gauge("distrl_engine_calls_total",
+                           "engine.generate invocations (batched waves)",
+                           registry=registry)
+    app.state.metrics_registry = registry
+
     @app.get("/health")
     def health():
         return {"status": "ok"}
+
+    @app.get("/metrics")
+    def metrics():
+        from fastapi import Response
+        m_engine_calls.set(batcher.calls)
+        return Response(generate_latest(registry),
+                        media_type="text/plain; version=0.0.4")
 
     @app.get("/v1/models")
     def models():
@@ -81,12 +107,19 @@ def create_app(engine, tokenizer, model_name: str,
         except ValueError as e:
             raise HTTPException(status_code=400, detail=str(e))
 
-    def _complete(prompts, sp, stop):
+    def _complete(prompts, sp, stop, endpoint="completions"):
         """Shared generation core: returns (flat [(text, finish)], usage)."""
         stops = ([stop] if isinstance(stop, str) else stop) or []
         prompt_ids = [tokenizer.encode(p) for p in prompts]
         eos = getattr(tokenizer, "eos_token_id", None)
-        outs = batcher.submit(prompt_ids, sp, eos_token_id=eos)
+        t0 = time.monotonic()
+        try:
+            outs = batcher.submit(prompt_ids, sp, eos_token_id=eos)
+        except Exception:
+            m_requests.labels(endpoint=endpoint, status="error").inc()
+            raise
+        m_requests.labels(endpoint=endpoint, status="ok").inc()
+        m_latency.labels(endpoint=endpoint).observe(time.monotonic() - t0)
         flat, completion_tokens = [], 0
         vocab = getattr(tokenizer, "vocab_size", None)
         for per_prompt in outs:
@@ -103,6 +136,8 @@ def create_app(engine, tokenizer, model_name: str,
                         text, finish = text[:cut], "stop"
                 flat.append((text, finish))
         prompt_tokens = sum(len(ids) for ids in prompt_ids)
+        m_prompt_toks.inc(prompt_tokens)
+        m_gen_toks.inc(completion_tokens)
         usage = {"prompt_tokens": prompt_tokens,
                  "completion_tokens": completion_tokens,
                  "total_tokens": prompt_tokens + completion_tokens}
@@ -134,7 +169,7 @@ def create_app(engine, tokenizer, model_name: str,
         from .rl.data import apply_template
         prompt = apply_template(tokenizer,
                                 [m.model_dump() for m in req.messages])
-        flat, usage = _complete([prompt], sp, req.stop)
+        flat, usage = _complete([prompt], sp, req.stop, endpoint="chat")
         choices = [{"index": i,
                     "message": {"role": "assistant", "content": text},
                     "finish_reason": finish}

@@ -75,6 +75,16 @@ def test_chat_completion(client):
     assert r2.status_code == 400
 
 
+def test_prometheus_metrics(client):
+    client.post("/v1/completions", json={
+        "prompt": "abc", "max_tokens": 3, "temperature": 0.0})
+    body = client.get("/metrics").text
+    assert "distrl_requests_total" in body
+    assert 'endpoint="completions",status="ok"' in body
+    assert "distrl_generated_tokens_total" in body
+    assert "distrl_engine_calls_total" in body
+
+
 def test_completion_echo_and_validation(client):
     r = client.post("/v1/completions", json={
         "prompt": "hi", "max_tokens": 2, "temperature": 0.0, "echo": True})
