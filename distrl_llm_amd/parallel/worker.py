@@ -116,7 +116,8 @@ def build_worker(rank: int, world_size: int, config: Dict,
                     torch.load(opt_path, map_location="cpu",
                                weights_only=True))
 
-    fabric = Fabric(rank, world_size, num_actors, num_learners, device)
+    fabric = Fabric(rank, world_size, num_actors, num_learners, device,
+                    timeout_s=float(config.get("fabric_timeout_s", 240.0)))
 
     logger = None
     if rank == 0:
