@@ -15,6 +15,28 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 @pytest.mark.timeout(600)
+def test_bench_torchrun_two_rank_contract():
+    """The driver's exact N>1 launcher (torch.distributed.run, one rank
+    per GPU) on CPU/gloo: one JSON line from rank 0 with n_gpus=2 and
+    weak-scaled global batch."""
+    env = dict(os.environ)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(29600 + os.getpid() % 300),
+         os.path.join(REPO, "bench.py"), "--cpu", "--tiny",
+         "--gpus", "2", "--steps", "1", "--warmup", "1"],
+        capture_output=True, text=True, timeout=540, env=env, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    json_lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, out.stdout  # rank 0 only
+    rec = json.loads(json_lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["global_batch"] == 20  # 10 prompts/rank, weak
+    assert rec["value"] > 0
+
+
+@pytest.mark.timeout(600)
 def test_bench_cpu_tiny_json_contract():
     env = dict(os.environ)
     env.setdefault("MASTER_ADDR", "127.0.0.1")
