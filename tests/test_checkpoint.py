@@ -144,7 +144,7 @@ def test_mid_episode_resume_bit_identical(tmp_path):
     bit-identical (optimizer, dataset order, counters and the per-rank
     sampling-RNG streams all restored)."""
     tmpdir = str(tmp_path)
-    port = 28500 + os.getpid() % 500
+    port = 22500 + os.getpid() % 500  # base unique across test files
     mp.spawn(_ckpt_worker, nprocs=2, args=(2, tmpdir, port, "a", None),
              join=True)
 
