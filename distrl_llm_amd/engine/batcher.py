@@ -51,6 +51,7 @@ class DynamicBatcher:
         self.engine = engine
         self.max_wait_ms = max_wait_ms
         self.calls = 0  # engine.generate invocations (for tests/metrics)
+        self._closed = False
         self._q: "queue.Queue" = queue.Queue()
         self._thread = threading.Thread(target=self._loop, daemon=True,
                                         name="distrl-batcher")
@@ -61,6 +62,8 @@ class DynamicBatcher:
     def submit(self, prompts: List[List[int]], sp: SamplingParams,
                eos_token_id: Optional[int] = None) -> List[List[List[int]]]:
         """Blocking: returns the engine.generate result for ``prompts``."""
+        if self._closed:
+            raise RuntimeError("DynamicBatcher is closed")
         req = _Request(prompts, sp, eos_token_id)
         self._q.put(req)
         req.event.wait()
@@ -69,6 +72,7 @@ class DynamicBatcher:
         return req.result
 
     def close(self):
+        self._closed = True
         self._q.put(_SHUTDOWN)
         self._thread.join(timeout=10)
 
@@ -85,7 +89,7 @@ class DynamicBatcher:
         while not shutdown:
             first = self._q.get()
             if first is _SHUTDOWN:
-                return
+                break
             batch = [first]
             deadline = time.monotonic() + self.max_wait_ms / 1e3
             while True:
@@ -103,6 +107,16 @@ class DynamicBatcher:
                 groups.setdefault(self._key(req), []).append(req)
             for reqs in groups.values():
                 self._run_group(reqs)
+        # fail any submitters that raced the shutdown (never strand a
+        # blocked caller)
+        while True:
+            try:
+                it = self._q.get_nowait()
+            except queue.Empty:
+                break
+            if it is not _SHUTDOWN:
+                it.error = RuntimeError("DynamicBatcher is closed")
+                it.event.set()
 
     def _run_group(self, reqs: List[_Request]):
         merged = [p for r in reqs for p in r.prompts]

@@ -80,3 +80,7 @@ def test_close_is_idempotent_and_unblocks(engine):
     batcher = DynamicBatcher(engine, max_wait_ms=1.0)
     batcher.close()
     assert not batcher._thread.is_alive()
+    # submit after close errors instead of blocking forever
+    sp = SamplingParams(max_tokens=2, temperature=0.0, n=1)
+    with pytest.raises(RuntimeError, match="closed"):
+        batcher.submit([[1, 2]], sp, None)
