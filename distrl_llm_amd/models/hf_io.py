@@ -121,10 +121,13 @@ def save_hf_checkpoint(model, path: str) -> None:
     from safetensors.torch import save_file
 
     os.makedirs(path, exist_ok=True)
-    s = model.spec
     state = {k: v.detach().contiguous().cpu()
              for k, v in model.named_parameters() if "lora_" not in k}
     save_file(state, os.path.join(path, "model.safetensors"))
+    _write_hf_config(model.spec, path)
+
+
+def _write_hf_config(s: ModelSpec, path: str) -> None:
     cfg = {
         "architectures": ["Qwen2ForCausalLM" if s.qkv_bias
                           else "LlamaForCausalLM"],
@@ -145,6 +148,35 @@ def save_hf_checkpoint(model, path: str) -> None:
     }
     with open(os.path.join(path, "config.json"), "w") as f:
         json.dump(cfg, f, indent=2)
+
+
+def save_merged_checkpoint(model, path: str) -> int:
+    """Merge the trained LoRA adapters into the base weights and write a
+    plain HF checkpoint (the deploy artifact the reference ecosystem
+    calls ``save_pretrained_merged``): W_merged = W + scale * B@A per
+    adapted projection. The model is not mutated. Returns the number of
+    merged projection sites."""
+    from safetensors.torch import save_file
+
+    merged = {k: v.detach().clone().float()
+              for k, v in model.named_parameters() if "lora_" not in k}
+    n = 0
+    for name, mod in model.named_modules():
+        a = getattr(mod, "lora_A", None)
+        b = getattr(mod, "lora_B", None)
+        if a is None or b is None:
+            continue
+        key = f"{name}.weight"
+        merged[key] = (merged[key]
+                       + mod.scale * (b.detach().float()
+                                      @ a.detach().float()))
+        n += 1
+    dtype = next(model.parameters()).dtype
+    os.makedirs(path, exist_ok=True)
+    save_file({k: v.to(dtype).contiguous().cpu() for k, v in merged.items()},
+              os.path.join(path, "model.safetensors"))
+    _write_hf_config(model.spec, path)
+    return n
 
 
 def resolve_spec(model_name_or_path: str) -> ModelSpec:

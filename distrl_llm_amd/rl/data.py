@@ -102,6 +102,47 @@ def synthetic_math_dataset(n: int, seed: int = 0, pad_words: int = 0) -> List[Di
     return rows
 
 
+def load_local_rows(path: str) -> List[Dict[str, str]]:
+    """Load a local dataset file into [{"problem", "solution"}, ...] rows
+    — the air-gapped counterpart of the reference's HF `load_dataset`
+    (reference train_distributed.py:38-44). Accepts .json (list of
+    objects), .jsonl, or .parquet; a directory is scanned for the first
+    such file. MATH-style "answer" columns are remapped to "solution"
+    (same remap as the reference)."""
+    import json as _json
+    import os as _os
+
+    if _os.path.isdir(path):
+        for f in sorted(_os.listdir(path)):
+            if f.endswith((".json", ".jsonl", ".parquet")):
+                return load_local_rows(_os.path.join(path, f))
+        raise FileNotFoundError(f"no .json/.jsonl/.parquet file in {path}")
+
+    if path.endswith(".jsonl"):
+        with open(path) as fh:
+            raw = [_json.loads(l) for l in fh if l.strip()]
+    elif path.endswith(".json"):
+        with open(path) as fh:
+            raw = _json.load(fh)
+        if isinstance(raw, dict):  # {"data": [...]} wrappers
+            raw = next(v for v in raw.values() if isinstance(v, list))
+    elif path.endswith(".parquet"):
+        import pandas as pd
+        raw = pd.read_parquet(path).to_dict("records")
+    else:
+        raise ValueError(f"unsupported dataset file type: {path}")
+
+    rows = []
+    for r in raw:
+        sol = r.get("solution", r.get("answer"))
+        if r.get("problem") is None or sol is None:
+            raise ValueError(
+                f"{path}: rows need 'problem' and 'solution' (or 'answer') "
+                f"columns, got {sorted(r)}")
+        rows.append({"problem": str(r["problem"]), "solution": str(sol)})
+    return rows
+
+
 class ListDataset:
     """Minimal stand-in for the HF dataset surface the Trainer uses
     (reference distributed_trainer.py:245-246,386: .shuffle() and

@@ -162,6 +162,32 @@ def test_engine_generation_matches_transformers(ckpt_dir):
         assert o[0] == out[0, len(p):].tolist()
 
 
+def test_merged_checkpoint_equals_adapter_forward(tmp_path):
+    """save_merged_checkpoint: a LoRA-less model loading the merged
+    export must match the adapter-bearing model's logits (W + scale*B@A
+    folded in)."""
+    from distrl_llm_amd.models.hf_io import save_merged_checkpoint
+    spec = get_spec("tiny-qwen2")
+    m = CausalLM(spec, lora_r=4, lora_alpha=8,
+                 dtype=torch.float32).random_init(3)
+    with torch.no_grad():  # give the adapter real content
+        for name, p in m.named_parameters():
+            if "lora_" in name:
+                p.add_(torch.randn_like(p) * 0.05)
+    d = str(tmp_path / "merged")
+    n = save_merged_checkpoint(m, d)
+    assert n == 7 * spec.num_layers
+
+    flat = CausalLM(spec, lora_r=0, dtype=torch.float32).random_init(99)
+    load_hf_checkpoint(flat, d)
+    ids = torch.randint(0, 2048, (2, 10))
+    torch.testing.assert_close(flat(ids), m(ids), rtol=1e-4, atol=1e-5)
+    # and transformers agrees with the merged export too
+    from transformers import AutoModelForCausalLM
+    hf = AutoModelForCausalLM.from_pretrained(d, dtype=torch.float32)
+    torch.testing.assert_close(hf(ids).logits, m(ids), rtol=2e-4, atol=2e-4)
+
+
 def test_worker_loads_local_checkpoint_dir(ckpt_dir, tmp_path):
     """build_worker with --model <local dir> trains from the checkpoint's
     weights, not random init (single-rank gloo world)."""

@@ -100,18 +100,24 @@ def build_config(args) -> dict:
 
 
 def load_datasets(args, tokenizer):
-    from distrl_llm_amd.rl.data import (ListDataset, process_dataset,
-                                        r1_preprompt, synthetic_math_dataset)
+    from distrl_llm_amd.rl.data import (ListDataset, load_local_rows,
+                                        process_dataset, r1_preprompt,
+                                        synthetic_math_dataset)
     rows = None
     if args.synthetic_dataset <= 0:
-        try:
-            from datasets import load_dataset
-            raw = load_dataset(args.dataset)["test"]
-            raw = raw.map(lambda x: {"solution": x["answer"], "answer": x["answer"]})
-            raw = raw.remove_columns(["answer"])
-            rows = [dict(r) for r in raw]
-        except Exception as e:
-            print(f"Dataset load failed ({e}); falling back to synthetic data")
+        if os.path.exists(args.dataset):
+            # local file/dir (air-gapped real data): json/jsonl/parquet
+            # with problem+solution (or MATH-style problem+answer) columns
+            rows = load_local_rows(args.dataset)
+        else:
+            try:
+                from datasets import load_dataset
+                raw = load_dataset(args.dataset)["test"]
+                raw = raw.map(lambda x: {"solution": x["answer"], "answer": x["answer"]})
+                raw = raw.remove_columns(["answer"])
+                rows = [dict(r) for r in raw]
+            except Exception as e:
+                print(f"Dataset load failed ({e}); falling back to synthetic data")
     if rows is None:
         n = args.synthetic_dataset if args.synthetic_dataset > 0 else 500
         rows = synthetic_math_dataset(n, seed=args.seed)
