@@ -9,7 +9,7 @@ is rejected loudly rather than silently ignored.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

@@ -17,7 +17,6 @@ broadcast into these tensors.
 from __future__ import annotations
 
 import os
-import time
 from typing import List, Optional
 
 import torch

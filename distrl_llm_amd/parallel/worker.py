@@ -17,7 +17,7 @@ import torch
 from ..config import EngineConfig
 from ..engine.engine import Engine
 from ..models.model import CausalLM
-from ..models.spec import get_spec, is_4bit_model_name
+from ..models.spec import is_4bit_model_name
 from ..parallel.fabric import Fabric
 from ..rl.trainer import Trainer
 from ..train.learner import Learner
