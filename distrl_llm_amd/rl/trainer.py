@@ -304,6 +304,11 @@ class Trainer:
         return stats
 
     def _train_rank0(self):
+        if self.config.get("eval_only"):
+            # standalone evaluation (--eval_only, optionally with
+            # --load_adapter): pass@1/BoN over the test split, no training
+            self.evaluate(0)
+            return
         rs = self._resume_state
         total_batch_steps = rs["total_batch_steps"] if rs else 0
         total_samples = rs["total_samples"] if rs else 0

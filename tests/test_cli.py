@@ -35,6 +35,20 @@ def _run(tmpdir, extra):
     return out
 
 
+@pytest.mark.timeout(600)
+def test_cli_eval_only(tmp_path):
+    """--eval_only scores the test split (pass@1 / BoN@8) and exits
+    without training artifacts."""
+    tmpdir = str(tmp_path)
+    _run(tmpdir, ["--episodes", "5", "--eval_only", "--run_name", "ev"])
+    recs = [json.loads(l) for l in
+            open(os.path.join(tmpdir, "metrics_ev.jsonl"))]
+    assert any("eval/pass@1(mean8)" in r and "eval/BoN(8)" in r
+               for r in recs)
+    assert not any("loss" in r for r in recs)
+    assert not os.path.exists(os.path.join(tmpdir, "run_ev"))
+
+
 @pytest.mark.timeout(1200)
 def test_cli_train_and_resume(tmp_path):
     tmpdir = str(tmp_path)

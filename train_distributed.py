@@ -59,6 +59,10 @@ def parse_args():
                            "save cadence (run_<name>/model_<step>): restores "
                            "adapter, optimizer, trainer counters, dataset "
                            "order and per-rank RNG streams")
+    args.add_argument("--eval_only", action="store_true",
+                      help="Evaluate (pass@1 / BoN@8) on the test split and "
+                           "exit — combine with --load_adapter to score a "
+                           "trained adapter")
     args.add_argument("--seed", type=int, default=3407)
     args.add_argument("--backend_device", type=str, default="auto",
                       choices=["auto", "cuda", "cpu"])
@@ -96,6 +100,7 @@ def build_config(args) -> dict:
         "seed": args.seed,
         "load_adapter": args.load_adapter,
         "resume": args.resume,
+        "eval_only": args.eval_only,
     }
 
 
