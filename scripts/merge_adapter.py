@@ -11,6 +11,10 @@ checkpoint directory (real pretrained base).
 """
 
 import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
