@@ -11,6 +11,12 @@ gate|up GEMM per layer — halving GEMM launches and giving hipBLASLt fatter
 N at decode's skinny M. Refresh cost is ~100 ms for 7B (196 rank-32
 GEMMs + adds), amortized over a whole generation round (SURVEY.md §2.4-A:
 LoRA folded into the projection epilogue).
+
+Replaces the reference's per-generate adapter hot-swap into vLLM
+(``load_lora`` inside every fast_generate, reference
+distributed_actor.py:148-150) — there the engine applies LoRA at runtime
+per token; here the fold happens once per round, which 288 GB of HBM
+makes free to cache.
 """
 
 from __future__ import annotations
