@@ -81,8 +81,11 @@ class DynamicBatcher:
     @staticmethod
     def _key(req: _Request):
         sp = req.sp
+        # a seeded request must run alone: its reproducibility contract
+        # is over (prompts, params, seed), which co-batching would break
+        nonce = id(req) if sp.seed is not None else None
         return (sp.max_tokens, sp.temperature, sp.top_p, sp.top_k, sp.n,
-                req.eos)
+                req.eos, nonce)
 
     def _loop(self):
         shutdown = False

@@ -404,6 +404,11 @@ class Engine:
                 self.model.train()
 
     def _generate_inner(self, prompts, sp, eos_token_id, prefill_token_budget):
+        if sp.seed is not None:
+            # per-request determinism (vLLM SamplingParams.seed analogue):
+            # reseed this call's sampling stream so identical (prompts,
+            # params, seed) calls reproduce regardless of engine history
+            self.generator.manual_seed(int(sp.seed))
         bs = self.pool.block_size
         max_total = self.cfg.max_seq_length
         results: List[List[List[int]]] = [[] for _ in prompts]

@@ -42,6 +42,7 @@ def create_app(engine, tokenizer, model_name: str,
         top_k: int = 0
         n: int = 1
         echo: bool = False
+        seed: Optional[int] = None
         stop: Optional[Union[str, List[str]]] = None
 
     class ChatMessage(BaseModel):
@@ -56,6 +57,7 @@ def create_app(engine, tokenizer, model_name: str,
         top_p: float = 1.0
         top_k: int = 0
         n: int = 1
+        seed: Optional[int] = None
         stop: Optional[Union[str, List[str]]] = None
 
     app = FastAPI(title="distrl-mi355x", version="0.1")
@@ -103,7 +105,8 @@ def create_app(engine, tokenizer, model_name: str,
         try:
             return SamplingParams(max_tokens=req.max_tokens,
                                   temperature=req.temperature,
-                                  top_p=req.top_p, top_k=req.top_k, n=req.n)
+                                  top_p=req.top_p, top_k=req.top_k, n=req.n,
+                                  seed=req.seed)
         except ValueError as e:
             raise HTTPException(status_code=400, detail=str(e))
 

@@ -59,6 +59,32 @@ def test_mixed_sampling_params_grouped_separately(engine):
     assert len(rb[0]) == 2 and all(len(x) == 2 for x in rb[0])
 
 
+def test_seeded_request_reproducible_and_isolated(engine):
+    """SamplingParams.seed: identical seeded calls reproduce regardless
+    of engine history, and seeded requests never co-batch (their
+    determinism contract is over the exact call)."""
+    sp_seeded = SamplingParams(max_tokens=5, temperature=0.9, top_p=0.9,
+                               n=2, seed=1234)
+    a = engine.generate([[5, 6, 7]], sp_seeded, eos_token_id=None)
+    # perturb engine RNG history with an unseeded call
+    engine.generate([[9, 9]], SamplingParams(max_tokens=3, temperature=1.0,
+                                             n=1), eos_token_id=None)
+    b = engine.generate([[5, 6, 7]], sp_seeded, eos_token_id=None)
+    assert a == b
+
+    batcher = DynamicBatcher(engine, max_wait_ms=50.0)
+    try:
+        with ThreadPoolExecutor(max_workers=2) as ex:
+            c0 = batcher.calls
+            f1 = ex.submit(batcher.submit, [[5, 6, 7]], sp_seeded, None)
+            f2 = ex.submit(batcher.submit, [[5, 6, 7]], sp_seeded, None)
+            r1, r2 = f1.result(timeout=60), f2.result(timeout=60)
+        assert batcher.calls - c0 == 2  # ran alone, not merged
+        assert r1 == r2 == a  # and each reproduces the direct call
+    finally:
+        batcher.close()
+
+
 def test_error_propagates_to_caller(engine):
     batcher = DynamicBatcher(engine, max_wait_ms=1.0)
     try:

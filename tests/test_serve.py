@@ -75,6 +75,17 @@ def test_chat_completion(client):
     assert r2.status_code == 400
 
 
+def test_seeded_request_deterministic(client):
+    body = {"prompt": "roll", "max_tokens": 5, "temperature": 1.0,
+            "seed": 42}
+    t1 = client.post("/v1/completions", json=body).json()["choices"][0]["text"]
+    # interleave an unseeded request to disturb engine RNG state
+    client.post("/v1/completions", json={"prompt": "x", "max_tokens": 3,
+                                         "temperature": 1.0})
+    t2 = client.post("/v1/completions", json=body).json()["choices"][0]["text"]
+    assert t1 == t2
+
+
 def test_prometheus_metrics(client):
     client.post("/v1/completions", json={
         "prompt": "abc", "max_tokens": 3, "temperature": 0.0})
