@@ -27,15 +27,20 @@ _SHUTDOWN = object()
 
 
 class _Request:
-    __slots__ = ("prompts", "sp", "eos", "event", "result", "error")
+    __slots__ = ("prompts", "sp", "eos", "event", "result", "error",
+                 "stream_q")
 
-    def __init__(self, prompts, sp, eos):
+    def __init__(self, prompts, sp, eos, stream=False):
         self.prompts = prompts
         self.sp = sp
         self.eos = eos
         self.event = threading.Event()
         self.result = None
         self.error = None
+        # streaming requests get an event queue the submitter drains:
+        # ("tok", prompt_i, cand_i, [ids...]) then ("done", result) or
+        # ("err", exception)
+        self.stream_q = queue.Queue() if stream else None
 
 
 class DynamicBatcher:
@@ -71,6 +76,17 @@ class DynamicBatcher:
             raise req.error
         return req.result
 
+    def submit_stream(self, prompts: List[List[int]], sp: SamplingParams,
+                      eos_token_id: Optional[int] = None) -> "queue.Queue":
+        """Non-blocking: enqueue a streaming request and return its event
+        queue. Items: ("tok", prompt_i, cand_i, [token_ids]) deltas, then
+        a final ("done", full_result) or ("err", exception)."""
+        if self._closed:
+            raise RuntimeError("DynamicBatcher is closed")
+        req = _Request(prompts, sp, eos_token_id, stream=True)
+        self._q.put(req)
+        return req.stream_q
+
     def close(self):
         self._closed = True
         self._q.put(_SHUTDOWN)
@@ -81,9 +97,11 @@ class DynamicBatcher:
     @staticmethod
     def _key(req: _Request):
         sp = req.sp
-        # a seeded request must run alone: its reproducibility contract
-        # is over (prompts, params, seed), which co-batching would break
-        nonce = id(req) if sp.seed is not None else None
+        # seeded requests must run alone (their reproducibility contract
+        # is over the exact call); streaming requests too (the stream_cb
+        # belongs to one caller)
+        nonce = (id(req) if sp.seed is not None or req.stream_q is not None
+                 else None)
         return (sp.max_tokens, sp.temperature, sp.top_p, sp.top_k, sp.n,
                 req.eos, nonce)
 
@@ -119,21 +137,32 @@ class DynamicBatcher:
                 break
             if it is not _SHUTDOWN:
                 it.error = RuntimeError("DynamicBatcher is closed")
+                if it.stream_q is not None:
+                    it.stream_q.put(("err", it.error))
                 it.event.set()
 
     def _run_group(self, reqs: List[_Request]):
         merged = [p for r in reqs for p in r.prompts]
+        stream_q = reqs[0].stream_q  # streaming groups are singletons
+        cb = None
+        if stream_q is not None:
+            cb = lambda pi, ci, toks: stream_q.put(("tok", pi, ci, toks))
         try:
             self.calls += 1
             outs = self.engine.generate(merged, reqs[0].sp,
-                                        eos_token_id=reqs[0].eos)
+                                        eos_token_id=reqs[0].eos,
+                                        stream_cb=cb)
             off = 0
             for r in reqs:
                 r.result = outs[off:off + len(r.prompts)]
                 off += len(r.prompts)
+            if stream_q is not None:
+                stream_q.put(("done", reqs[0].result))
         except Exception as e:  # deliver the failure to every waiter
             for r in reqs:
                 r.error = e
+            if stream_q is not None:
+                stream_q.put(("err", e))
         finally:
             for r in reqs:
                 r.event.set()

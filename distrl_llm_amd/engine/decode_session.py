@@ -142,7 +142,8 @@ class DecodeSession:
 
     # -------------------------------------------------------------- run
 
-    def run(self, chunk: int = 64) -> List[List[int]]:
+    def run(self, chunk: int = 64, stream_cb=None) -> List[List[int]]:
+        reported = [0] * len(self.seqs)  # out_buf rows already streamed
         if self.max_steps > 0:
             if self.use_graph:
                 try:
@@ -162,6 +163,16 @@ class DecodeSession:
                     for _ in range(n):
                         self._step()
                 steps += n
+                if stream_cb is not None:
+                    # chunk-granular streaming: emit newly valid rows
+                    pos = self.positions.cpu().tolist()
+                    buf = self.out_buf[:steps].cpu()
+                    for i, q in enumerate(self.seqs):
+                        n_new = pos[i] - self.prompt_lens[i]
+                        if n_new > reported[i]:
+                            stream_cb(q.parent_prompt, q.cand_index,
+                                      buf[reported[i]:n_new, i].tolist())
+                            reported[i] = n_new
                 if bool(self.finished.all()):
                     break
 

@@ -347,6 +347,7 @@ class Engine:
             self._seq_counter += 1
             child = Sequence(self._seq_counter, parent.prompt_ids,
                              parent.parent_prompt)
+            child.cand_index = i
             table = []
             for b in parent.block_table[:full]:
                 self.pool.allocator.incref(b)
@@ -376,12 +377,17 @@ class Engine:
     @torch.no_grad()
     def generate(self, prompts: List[List[int]], sp: SamplingParams,
                  eos_token_id: Optional[int] = None,
-                 prefill_token_budget: int = 8192) -> List[List[List[int]]]:
+                 prefill_token_budget: int = 8192,
+                 stream_cb=None) -> List[List[List[int]]]:
         """Generate sp.n completions per prompt.
 
         prompts: token-id lists. Returns per prompt a list of n output
         token-id lists (EOS included when emitted, like vLLM's
         ``o.token_ids``).
+
+        stream_cb: optional ``f(prompt_index, cand_index, new_token_ids)``
+        called as tokens become known — per token on the eager path, per
+        decode chunk on the session path (serving SSE streaming).
         """
         was_training = self.model.training
         self.model.eval()
@@ -391,7 +397,7 @@ class Engine:
             self.fused.refresh()
         try:
             return self._generate_inner(prompts, sp, eos_token_id,
-                                        prefill_token_budget)
+                                        prefill_token_budget, stream_cb)
         except Exception:
             # a failure mid-generation strands this call's in-flight
             # sequences' KV blocks; generate calls are serialized, so no
@@ -403,7 +409,8 @@ class Engine:
             if was_training:
                 self.model.train()
 
-    def _generate_inner(self, prompts, sp, eos_token_id, prefill_token_budget):
+    def _generate_inner(self, prompts, sp, eos_token_id, prefill_token_budget,
+                        stream_cb=None):
         if sp.seed is not None:
             # per-request determinism (vLLM SamplingParams.seed analogue):
             # reseed this call's sampling stream so identical (prompts,
@@ -464,6 +471,9 @@ class Engine:
                                          generator=self.generator)
                 children = self._fork(parent, sp.n, first.tolist())
                 for c in children:
+                    if stream_cb is not None:
+                        stream_cb(c.parent_prompt, c.cand_index,
+                                  [c.output_ids[0]])
                     if ((eos_token_id is not None
                          and c.output_ids[-1] == eos_token_id)
                             or len(c.output_ids) >= sp.max_tokens
@@ -493,7 +503,7 @@ class Engine:
                         break
                 session = DecodeSession(self, running, sp, eos_token_id)
                 with trace_range(f"engine/decode_wave[{len(running)}]"):
-                    outs = session.run()
+                    outs = session.run(stream_cb=stream_cb)
                 for q, ids in zip(running, outs):
                     results[q.parent_prompt].append(ids)
                     self._finish(q)
@@ -516,6 +526,8 @@ class Engine:
             for i, q in enumerate(running):
                 t = next_list[i]
                 q.output_ids.append(t)
+                if stream_cb is not None:
+                    stream_cb(q.parent_prompt, q.cand_index, [t])
                 done = ((eos_token_id is not None and t == eos_token_id)
                         or len(q.output_ids) >= sp.max_tokens
                         or q.total_len >= max_total)

@@ -95,7 +95,8 @@ class Sequence:
     """One decoding stream (one candidate of one prompt)."""
 
     __slots__ = ("seq_id", "prompt_ids", "output_ids", "block_table",
-                 "context_len", "finished", "parent_prompt", "last_logits_idx")
+                 "context_len", "finished", "parent_prompt",
+                 "last_logits_idx", "cand_index")
 
     def __init__(self, seq_id: int, prompt_ids: List[int], parent_prompt: int):
         self.seq_id = seq_id
@@ -106,6 +107,7 @@ class Sequence:
         self.finished = False
         self.parent_prompt = parent_prompt
         self.last_logits_idx: Optional[int] = None
+        self.cand_index = 0  # which of the n candidates of its prompt
 
     @property
     def total_len(self) -> int:

@@ -43,6 +43,7 @@ def create_app(engine, tokenizer, model_name: str,
         n: int = 1
         echo: bool = False
         seed: Optional[int] = None
+        stream: bool = False
         stop: Optional[Union[str, List[str]]] = None
 
     class ChatMessage(BaseModel):
@@ -58,6 +59,7 @@ def create_app(engine, tokenizer, model_name: str,
         top_k: int = 0
         n: int = 1
         seed: Optional[int] = None
+        stream: bool = False
         stop: Optional[Union[str, List[str]]] = None
 
     app = FastAPI(title="distrl-mi355x", version="0.1")
@@ -146,12 +148,65 @@ def create_app(engine, tokenizer, model_name: str,
                  "total_tokens": prompt_tokens + completion_tokens}
         return flat, usage
 
+    def _decode_delta(toks):
+        vocab = getattr(tokenizer, "vocab_size", None)
+        keep = [t for t in toks if vocab is None or t < vocab or t > 260]
+        return tokenizer.decode(keep, skip_special_tokens=True)
+
+    def _sse_stream(prompt_ids, sp, rid, obj, delta_fn, endpoint):
+        """SSE generator over one streaming request (single prompt, n=1;
+        per-token events on CPU, per-decode-chunk on GPU). Stop-string
+        truncation is not applied mid-stream (tokens are emitted as
+        sampled)."""
+        import json as _json
+        eos = getattr(tokenizer, "eos_token_id", None)
+        q = batcher.submit_stream(prompt_ids, sp, eos_token_id=eos)
+        finish = "length"
+        while True:
+            kind, *rest = q.get()
+            if kind == "err":
+                m_requests.labels(endpoint=endpoint, status="error").inc()
+                yield "data: " + _json.dumps({"error": str(rest[0])}) + "\n\n"
+                break
+            if kind == "done":
+                result = rest[0]
+                out_ids = result[0][0]
+                if eos is not None and out_ids and out_ids[-1] == eos:
+                    finish = "stop"
+                m_requests.labels(endpoint=endpoint, status="ok").inc()
+                m_gen_toks.inc(len(out_ids))
+                yield ("data: " + _json.dumps(
+                    {"id": rid, "object": obj, "model": model_name,
+                     "choices": [delta_fn("", finish)]}) + "\n\n")
+                yield "data: [DONE]\n\n"
+                break
+            _pi, _ci, toks = rest  # kind == "tok"
+            text = _decode_delta(toks)
+            if text:
+                yield ("data: " + _json.dumps(
+                    {"id": rid, "object": obj, "model": model_name,
+                     "choices": [delta_fn(text, None)]}) + "\n\n")
+
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
         prompts = [req.prompt] if isinstance(req.prompt, str) else list(req.prompt)
         if not prompts:
             raise HTTPException(status_code=400, detail="empty prompt")
         sp = _sp_or_400(req)
+        if req.stream:
+            from fastapi.responses import StreamingResponse
+            if len(prompts) != 1 or sp.n != 1:
+                raise HTTPException(status_code=400,
+                                    detail="streaming supports a single "
+                                           "prompt with n=1")
+            rid = f"cmpl-{created}-{int(time.time() * 1e6) & 0xFFFFFF:x}"
+            gen = _sse_stream([tokenizer.encode(prompts[0])], sp, rid,
+                              "text_completion",
+                              lambda text, fin: {"index": 0, "text": text,
+                                                 "finish_reason": fin,
+                                                 "logprobs": None},
+                              "completions")
+            return StreamingResponse(gen, media_type="text/event-stream")
         flat, usage = _complete(prompts, sp, req.stop)
         choices = [{"index": i,
                     "text": (prompts[i // sp.n] + text) if req.echo else text,
@@ -172,6 +227,21 @@ def create_app(engine, tokenizer, model_name: str,
         from .rl.data import apply_template
         prompt = apply_template(tokenizer,
                                 [m.model_dump() for m in req.messages])
+        if req.stream:
+            from fastapi.responses import StreamingResponse
+            if sp.n != 1:
+                raise HTTPException(status_code=400,
+                                    detail="streaming supports n=1")
+            rid = f"chatcmpl-{created}-{int(time.time() * 1e6) & 0xFFFFFF:x}"
+            gen = _sse_stream([tokenizer.encode(prompt)], sp, rid,
+                              "chat.completion.chunk",
+                              lambda text, fin: {"index": 0,
+                                                 "delta": ({"content": text}
+                                                           if fin is None
+                                                           else {}),
+                                                 "finish_reason": fin},
+                              "chat")
+            return StreamingResponse(gen, media_type="text/event-stream")
         flat, usage = _complete([prompt], sp, req.stop, endpoint="chat")
         choices = [{"index": i,
                     "message": {"role": "assistant", "content": text},

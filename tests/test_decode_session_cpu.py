@@ -72,6 +72,24 @@ def test_session_sampling_seed_deterministic(model, force_session):
     assert a != c  # different engine seed gives a different stream
 
 
+def test_session_streaming_deltas_match_final(model, force_session):
+    """stream_cb deltas (chunk-granular on the session path) concatenate
+    exactly to the returned sequences, including the prefill-sampled
+    first token and EOS-frozen lanes."""
+    e = _engine(model)
+    prompts = [[1, 5, 9, 2, 7], [3, 3, 8]]
+    sp = SamplingParams(max_tokens=6, temperature=0.0, n=2)
+    streamed = {}  # (prompt, cand) -> [tokens]
+
+    def cb(pi, ci, toks):
+        streamed.setdefault((pi, ci), []).extend(toks)
+
+    res = e.generate(prompts, sp, eos_token_id=None, stream_cb=cb)
+    for pi, per_prompt in enumerate(res):
+        for ci, ids in enumerate(per_prompt):
+            assert streamed[(pi, ci)] == ids, (pi, ci)
+
+
 def test_session_max_seq_len_clamp(model, force_session):
     """A prompt near max_seq_length stops at the limit, not past it."""
     e = _engine(model)

@@ -86,6 +86,55 @@ def test_seeded_request_deterministic(client):
     assert t1 == t2
 
 
+def _collect_sse(resp):
+    events = []
+    for line in resp.iter_lines():
+        if line.startswith("data: "):
+            payload = line[len("data: "):]
+            if payload == "[DONE]":
+                events.append(None)
+            else:
+                import json
+                events.append(json.loads(payload))
+    return events
+
+
+def test_streaming_completion_matches_nonstreaming(client):
+    body = {"prompt": "2+2=", "max_tokens": 6, "temperature": 0.0}
+    full = client.post("/v1/completions", json=body).json()["choices"][0]["text"]
+
+    with client.stream("POST", "/v1/completions",
+                       json={**body, "stream": True}) as r:
+        assert r.status_code == 200
+        assert r.headers["content-type"].startswith("text/event-stream")
+        events = _collect_sse(r)
+    assert events[-1] is None  # [DONE]
+    chunks = [e["choices"][0]["text"] for e in events[:-1]]
+    finishes = [e["choices"][0]["finish_reason"] for e in events[:-1]]
+    assert "".join(chunks) == full
+    assert finishes[-1] in ("stop", "length")
+    assert len(chunks) >= 2  # CPU path streams incrementally, not one blob
+
+
+def test_streaming_chat_and_validation(client):
+    with client.stream("POST", "/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 4, "temperature": 0.0, "stream": True}) as r:
+        assert r.status_code == 200
+        events = _collect_sse(r)
+    assert events[-1] is None
+    assert events[0]["object"] == "chat.completion.chunk"
+    assert events[-2]["choices"][0]["finish_reason"] in ("stop", "length")
+    # multi-prompt / n>1 streaming rejected
+    r = client.post("/v1/completions", json={
+        "prompt": ["a", "b"], "max_tokens": 2, "stream": True})
+    assert r.status_code == 400
+    r = client.post("/v1/completions", json={
+        "prompt": "a", "n": 2, "max_tokens": 2, "temperature": 1.0,
+        "stream": True})
+    assert r.status_code == 400
+
+
 def test_prometheus_metrics(client):
     client.post("/v1/completions", json={
         "prompt": "abc", "max_tokens": 3, "temperature": 0.0})
