@@ -7,9 +7,14 @@ paged-KV engine can also be used as a standalone completion server:
     python -m distrl_llm_amd.serve --model qwen2.5-7b --port 8000
 
 Endpoints (OpenAI completions-compatible subset):
-    GET  /health           liveness
-    GET  /v1/models        the single served model
-    POST /v1/completions   prompt(s) -> n sampled completions
+    GET  /health              liveness
+    GET  /metrics             Prometheus counters/histograms
+    GET  /v1/models           the single served model
+    POST /v1/completions      prompt(s) -> n sampled completions
+    POST /v1/chat/completions chat-templated messages -> completions
+Both POST endpoints accept ``"stream": true`` (SSE; single prompt, n=1)
+and ``"seed"`` for per-request determinism (seeded/streaming requests
+run in their own decode wave).
 
 Handlers are sync ``def`` so FastAPI runs them in its threadpool;
 concurrent requests are merged into shared decode waves by
