@@ -49,6 +49,7 @@ def create_app(engine, tokenizer, model_name: str,
         echo: bool = False
         seed: Optional[int] = None
         stream: bool = False
+        logprobs: Optional[int] = None
         stop: Optional[Union[str, List[str]]] = None
 
     class ChatMessage(BaseModel):
@@ -118,7 +119,8 @@ def create_app(engine, tokenizer, model_name: str,
             raise HTTPException(status_code=400, detail=str(e))
 
     def _complete(prompts, sp, stop, endpoint="completions"):
-        """Shared generation core: returns (flat [(text, finish)], usage)."""
+        """Shared generation core: returns
+        (flat [(text, finish, out_ids)], usage)."""
         stops = ([stop] if isinstance(stop, str) else stop) or []
         prompt_ids = [tokenizer.encode(p) for p in prompts]
         eos = getattr(tokenizer, "eos_token_id", None)
@@ -144,7 +146,7 @@ def create_app(engine, tokenizer, model_name: str,
                     cut = text.find(s)
                     if cut >= 0:
                         text, finish = text[:cut], "stop"
-                flat.append((text, finish))
+                flat.append((text, finish, out_ids))
         prompt_tokens = sum(len(ids) for ids in prompt_ids)
         m_prompt_toks.inc(prompt_tokens)
         m_gen_toks.inc(completion_tokens)
@@ -213,16 +215,52 @@ def create_app(engine, tokenizer, model_name: str,
                               "completions")
             return StreamingResponse(gen, media_type="text/event-stream")
         flat, usage = _complete(prompts, sp, req.stop)
-        choices = [{"index": i,
-                    "text": (prompts[i // sp.n] + text) if req.echo else text,
-                    "finish_reason": finish, "logprobs": None}
-                   for i, (text, finish) in enumerate(flat)]
+        prompt_ids = [tokenizer.encode(p) for p in prompts]
+        choices = []
+        for i, (text, finish, out_ids) in enumerate(flat):
+            lp = None
+            if req.logprobs is not None:
+                lp = _rescore_logprobs(prompt_ids[i // sp.n], out_ids,
+                                       min(max(req.logprobs, 0), 5))
+            choices.append({"index": i,
+                            "text": (prompts[i // sp.n] + text)
+                                    if req.echo else text,
+                            "finish_reason": finish, "logprobs": lp})
         return {"id": f"cmpl-{created}-{int(time.time() * 1e6) & 0xFFFFFF:x}",
                 "object": "text_completion",
                 "created": int(time.time()),
                 "model": model_name,
                 "choices": choices,
                 "usage": usage}
+
+    def _rescore_logprobs(prompt_ids, out_ids, top_n):
+        """OpenAI-style logprobs by teacher-forced rescoring: one extra
+        forward over prompt+completion, model log-probs of the chosen
+        tokens (pre-temperature, as the OpenAI API defines them) plus the
+        top-N alternatives per position."""
+        import torch
+        if not out_ids:
+            return {"tokens": [], "token_logprobs": [],
+                    "top_logprobs": [], "text_offset": []}
+        ids = torch.tensor([list(prompt_ids) + list(out_ids)],
+                           device=engine.device)
+        with torch.no_grad():
+            logits = engine.model(ids)[0].float()
+        lp = torch.log_softmax(logits, -1)
+        start = len(prompt_ids) - 1
+        tokens, tlp, top = [], [], []
+        for j, t in enumerate(out_ids):
+            row = lp[start + j]
+            tokens.append(_decode_delta([t]))
+            tlp.append(float(row[t]))
+            if top_n > 0:
+                tv, ti = row.topk(top_n)
+                top.append({_decode_delta([int(i)]): float(v)
+                            for v, i in zip(tv, ti)})
+            else:
+                top.append(None)
+        return {"tokens": tokens, "token_logprobs": tlp,
+                "top_logprobs": top, "text_offset": []}
 
     @app.post("/v1/chat/completions")
     def chat_completions(req: ChatRequest):
@@ -251,7 +289,7 @@ def create_app(engine, tokenizer, model_name: str,
         choices = [{"index": i,
                     "message": {"role": "assistant", "content": text},
                     "finish_reason": finish}
-                   for i, (text, finish) in enumerate(flat)]
+                   for i, (text, finish, _ids) in enumerate(flat)]
         return {"id": f"chatcmpl-{created}-{int(time.time() * 1e6) & 0xFFFFFF:x}",
                 "object": "chat.completion",
                 "created": int(time.time()),

@@ -86,6 +86,31 @@ def test_seeded_request_deterministic(client):
     assert t1 == t2
 
 
+def test_completion_logprobs_rescoring(client):
+    """logprobs=N: greedy-chosen tokens must carry the row-max model
+    log-prob and appear first in their own top_logprobs."""
+    r = client.post("/v1/completions", json={
+        "prompt": "abc", "max_tokens": 4, "temperature": 0.0,
+        "logprobs": 2})
+    body = r.json()
+    lp = body["choices"][0]["logprobs"]
+    n = len(lp["tokens"])
+    assert n == body["usage"]["completion_tokens"] > 0
+    assert len(lp["token_logprobs"]) == n and len(lp["top_logprobs"]) == n
+    for j in range(n):
+        assert lp["token_logprobs"][j] <= 0.0
+        top = lp["top_logprobs"][j]
+        assert len(top) == 2
+        # greedy: the chosen token is the argmax -> the top entry's value
+        # equals the chosen token's logprob
+        assert max(top.values()) == pytest.approx(lp["token_logprobs"][j],
+                                                  abs=1e-5)
+    # default: no logprobs field computed
+    r2 = client.post("/v1/completions", json={
+        "prompt": "abc", "max_tokens": 2, "temperature": 0.0})
+    assert r2.json()["choices"][0]["logprobs"] is None
+
+
 def _collect_sse(resp):
     events = []
     for line in resp.iter_lines():
