@@ -73,3 +73,51 @@ def test_split_dict_lists_partition(n, sizes):
     chunks = split_dict_lists(data, sizes)
     flat = [x for c in chunks for x in c["a"]]
     assert flat == data["a"]
+
+
+@given(
+    st.lists(st.integers(min_value=1, max_value=4), min_size=1, max_size=4),
+    st.integers(min_value=2, max_value=8),
+    st.integers(min_value=1, max_value=8),
+    st.integers(min_value=1, max_value=4),
+    st.integers(min_value=0, max_value=2 ** 31 - 1),
+)
+@settings(max_examples=40, deadline=None)
+def test_pipeline_sample_conservation(group_sizes, n, topk, learners, seed):
+    """End-to-end data-path invariant: process_candidates (advantages +
+    top-k) -> merge_candidates -> even_chunk_sizes must train every kept
+    sample exactly once — min(topk, n) samples per prompt, partitioned
+    across learners with no loss or duplication."""
+    import numpy as np
+
+    from distrl_llm_amd.rl.advantage import (even_chunk_sizes,
+                                             merge_candidates,
+                                             process_candidates)
+    rng = np.random.default_rng(seed)
+    cands = []
+    uid = 0
+    total_prompts = 0
+    for g in group_sizes:
+        total_prompts += g
+        cand = {"problem": [], "answers": [], "solution": [],
+                "token_lengths": [], "rewards": []}
+        for _ in range(g):
+            cand["problem"].append([f"p{uid}"] * n)
+            cand["answers"].append([f"a{uid}_{j}" for j in range(n)])
+            cand["solution"].append([f"s{uid}"] * n)
+            cand["token_lengths"].append([int(x) for x in
+                                          rng.integers(1, 50, n)])
+            cand["rewards"].append(rng.normal(size=(n, 2)))
+            uid += 1
+        cands.append(cand)
+
+    kept, _stats = process_candidates(cands, "grpo", topk)
+    problems, answers, rewards = merge_candidates(kept)
+    expect = total_prompts * min(topk, n)
+    assert len(problems) == len(answers) == len(rewards) == expect
+    # answers are globally unique -> no duplicated training sample
+    assert len(set(answers)) == expect
+
+    sizes = even_chunk_sizes(len(problems), learners)
+    assert sum(sizes) == expect and len(sizes) == learners
+    assert max(sizes) - min(sizes) <= 1
