@@ -50,3 +50,17 @@ def test_greedy_is_argmax():
     logits = torch.randn(32)
     t = R.sample_tokens(logits.unsqueeze(0), 0.0, 1.0, 0)
     assert int(t) == int(logits.argmax())
+
+
+def test_generation_config_translation():
+    """GenerationConfig -> SamplingParams the way the reference's
+    BaseActor translates HF GenerationConfig to vLLM params (reference
+    distributed_actor.py:41-48: top_p hardcoded 0.95)."""
+    from distrl_llm_amd.config import GenerationConfig
+    gc = GenerationConfig(max_new_tokens=77, temperature=1.2,
+                          num_return_sequences=16, do_sample=True)
+    sp = gc.to_sampling_params()
+    assert (sp.max_tokens, sp.temperature, sp.n, sp.top_p) == (77, 1.2, 16, 0.95)
+    # do_sample=False means greedy regardless of temperature
+    sp2 = GenerationConfig(temperature=0.9, do_sample=False).to_sampling_params()
+    assert sp2.temperature == 0.0
