@@ -97,3 +97,40 @@ def test_eos_mid_batch_frees_blocks(model, eos_prompt):
     assert res[0][0][-1] == first  # terminated by EOS (possibly step 1)
     assert len(res[0][0]) <= 6
     assert engine.pool.allocator.num_free == 64
+
+
+@settings(max_examples=10, deadline=None)
+@given(
+    prompts=st.lists(
+        st.lists(st.integers(min_value=1, max_value=500), min_size=1,
+                 max_size=20),
+        min_size=1, max_size=4),
+    max_tokens=st.integers(min_value=1, max_value=6),
+)
+def test_llama_family_greedy_equals_naive(prompts, max_tokens):
+    """The Llama architecture branch (no qkv bias, untied lm_head,
+    different rope theta) through the same engine paths."""
+    m = _llama_model()
+    cfg = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=128,
+                       max_num_seqs=32)
+    engine = Engine(m, cfg, device=torch.device("cpu"), seed=0)
+    sp = SamplingParams(max_tokens=max_tokens, temperature=0.0, n=2)
+    results = engine.generate(prompts, sp, eos_token_id=None)
+    for p, res in zip(prompts, results):
+        expected = _naive_greedy(m, p, max_tokens)
+        for ids in res:
+            assert ids == expected
+    assert engine.pool.allocator.num_free == 128
+
+
+_LLAMA = {}
+
+
+def _llama_model():
+    if "m" not in _LLAMA:
+        m = CausalLM(get_spec("tiny-llama"), lora_r=4, lora_alpha=8,
+                     dtype=torch.float32)
+        m.random_init(seed=77)
+        assert m.lm_head is not None  # untied head actually exercised
+        _LLAMA["m"] = m
+    return _LLAMA["m"]
