@@ -141,3 +141,29 @@ def test_cli_warm_start_adapter(tmp_path):
     a, b = lora_state_dict(m1), lora_state_dict(m2)
     for k in a:
         torch.testing.assert_close(a[k], b[k])
+
+
+def test_lora_dropout_train_eval_semantics():
+    """--lora_dropout > 0: stochastic adapter path in train mode,
+    deterministic (no-drop) in eval mode (PEFT semantics; reference
+    helper.py:25-46 passes lora_dropout through)."""
+    import torch
+
+    from distrl_llm_amd.models import CausalLM, get_spec
+    m = CausalLM(get_spec("tiny-qwen2"), lora_r=4, lora_alpha=8,
+                 lora_dropout=0.5, dtype=torch.float32).random_init(2)
+    with torch.no_grad():  # nonzero B so the adapter contributes
+        for name, p in m.named_parameters():
+            if "lora_B" in name:
+                p.add_(torch.randn_like(p))
+    ids = torch.randint(0, 2048, (1, 6))
+    m.train()
+    torch.manual_seed(0)
+    a = m(ids)
+    torch.manual_seed(1)
+    b = m(ids)
+    assert not torch.equal(a, b)  # dropout is live in train mode
+    m.eval()
+    with torch.no_grad():
+        c, d = m(ids), m(ids)
+    assert torch.equal(c, d)  # eval: deterministic, no dropping
