@@ -120,3 +120,25 @@ def test_sampling_params_validation():
         SamplingParams(temperature=-1.0)
     with _pytest.raises(ValueError):
         SamplingParams(n=0)
+
+
+def test_nf4_quantized_model_generates():
+    """A 4-bit (nf4) quantized model runs the same engine paths: the
+    decode math operates on the quantized weight image (identical to the
+    merged nf4(W)+BA semantics the GPU path uses)."""
+    from distrl_llm_amd.models import CausalLM, get_spec
+    model = CausalLM(get_spec("tiny-qwen2"), lora_r=4, lora_alpha=8,
+                     dtype=torch.float32)
+    model.random_init(seed=13)
+    model.quantize_nf4_()
+    cfg = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=128,
+                       max_num_seqs=16)
+    engine = Engine(model, cfg, device=torch.device("cpu"), seed=0)
+    prompts = [[1, 2, 3, 4], [9, 8]]
+    sp = SamplingParams(max_tokens=5, temperature=0.0, n=2)
+    res = engine.generate(prompts, sp, eos_token_id=None)
+    for p, r in zip(prompts, res):
+        expected = _naive_greedy(model, p, 5)
+        for ids in r:
+            assert ids == expected
+    assert engine.pool.allocator.num_free == 128
