@@ -45,10 +45,16 @@ _REGISTRY = {
                               1e6, 1e-6, True, True),
     "qwen2.5-7b": ModelSpec("qwen2.5-7b", 152064, 3584, 18944, 28, 28, 4, 128,
                             1e6, 1e-6, False, True),
+    "qwen2.5-14b": ModelSpec("qwen2.5-14b", 152064, 5120, 13824, 48, 40, 8, 128,
+                             1e6, 1e-5, False, True),
     "qwen2.5-32b": ModelSpec("qwen2.5-32b", 152064, 5120, 27648, 64, 40, 8, 128,
+                             1e6, 1e-5, False, True),
+    "qwen2.5-72b": ModelSpec("qwen2.5-72b", 152064, 8192, 29568, 80, 64, 8, 128,
                              1e6, 1e-5, False, True),
     "llama-3-8b": ModelSpec("llama-3-8b", 128256, 4096, 14336, 32, 32, 8, 128,
                             5e5, 1e-5, False, False),
+    "llama-3-70b": ModelSpec("llama-3-70b", 128256, 8192, 28672, 80, 64, 8, 128,
+                             5e5, 1e-5, False, False),
     # tiny models for CPU tests / the gloo plumbing config
     "tiny-qwen2": ModelSpec("tiny-qwen2", 2048, 64, 128, 2, 4, 2, 16,
                             1e4, 1e-6, True, True, max_position=512),
@@ -73,10 +79,16 @@ def get_spec(model_name: str) -> ModelSpec:
         return _REGISTRY["qwen2.5-1.5b"]
     if "qwen2.5-7b" in low:
         return _REGISTRY["qwen2.5-7b"]
+    if "qwen2.5-14b" in low:
+        return _REGISTRY["qwen2.5-14b"]
     if "qwen2.5-32b" in low:
         return _REGISTRY["qwen2.5-32b"]
+    if "qwen2.5-72b" in low:
+        return _REGISTRY["qwen2.5-72b"]
     if "llama-3" in low and "8b" in low:
         return _REGISTRY["llama-3-8b"]
+    if "llama-3" in low and "70b" in low:
+        return _REGISTRY["llama-3-70b"]
     raise ValueError(f"Unknown model architecture for name: {model_name!r}; "
                      f"known: {sorted(_REGISTRY)}")
 

@@ -167,3 +167,28 @@ def test_lora_dropout_train_eval_semantics():
     with torch.no_grad():
         c, d = m(ids), m(ids)
     assert torch.equal(c, d)  # eval: deterministic, no dropping
+
+
+def test_spec_registry_resolution_and_shapes():
+    """Registry name resolution for every supported family/size, and
+    internal dimension consistency (published architecture shapes)."""
+    from distrl_llm_amd.models.spec import get_spec
+    cases = {
+        "unsloth/Qwen2.5-7B-Instruct-bnb-4bit": (3584, 28, 28, 4),
+        "Qwen/Qwen2.5-0.5B-Instruct": (896, 24, 14, 2),
+        "qwen2.5-1.5b": (1536, 28, 12, 2),
+        "Qwen/Qwen2.5-14B-Instruct": (5120, 48, 40, 8),
+        "unsloth/Qwen2.5-32B-Instruct-bnb-4bit": (5120, 64, 40, 8),
+        "Qwen/Qwen2.5-72B-Instruct": (8192, 80, 64, 8),
+        "meta-llama/Meta-Llama-3-8B": (4096, 32, 32, 8),
+        "meta-llama/Meta-Llama-3-70B-Instruct": (8192, 80, 64, 8),
+    }
+    for name, (h, L, q, kv) in cases.items():
+        s = get_spec(name)
+        assert (s.hidden_size, s.num_layers, s.num_heads,
+                s.num_kv_heads) == (h, L, q, kv), name
+        assert s.q_size == s.num_heads * s.head_dim
+        assert s.num_heads % s.num_kv_heads == 0
+    import pytest as _p
+    with _p.raises(ValueError):
+        get_spec("mystery-model-99b")
