@@ -140,6 +140,17 @@ def save_adapter(model: nn.Module, path: str, base_model_name: str,
     save_file(state, os.path.join(path, "adapter_model.safetensors"))
 
 
+def adapter_hyperparams(path: str):
+    """(r, alpha, dropout) from a PEFT adapter directory's
+    adapter_config.json — lets loaders build a matching model instead of
+    guessing the rank."""
+    import json
+    with open(os.path.join(path, "adapter_config.json")) as f:
+        cfg = json.load(f)
+    return (int(cfg.get("r", 32)), float(cfg.get("lora_alpha", 16)),
+            float(cfg.get("lora_dropout", 0.0)))
+
+
 def load_adapter(model: nn.Module, path: str) -> int:
     """Load a PEFT adapter directory saved by save_adapter (or by PEFT)."""
     from safetensors.torch import load_file

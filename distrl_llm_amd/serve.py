@@ -327,7 +327,11 @@ def main():
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     spec = resolve_spec(args.model)
-    model = CausalLM(spec, lora_r=32 if args.adapter else 0, lora_alpha=16,
+    lora_r, lora_alpha = 0, 16.0
+    if args.adapter:
+        from .models.lora import adapter_hyperparams
+        lora_r, lora_alpha, _ = adapter_hyperparams(args.adapter)
+    model = CausalLM(spec, lora_r=lora_r, lora_alpha=lora_alpha,
                      dtype=dtype, device=device)
     model.random_init(args.seed)
     if is_hf_checkpoint_dir(args.model):

@@ -30,14 +30,22 @@ def main():
     ap.add_argument("--model", required=True)
     ap.add_argument("--adapter", required=True)
     ap.add_argument("--out", required=True)
-    ap.add_argument("--lora_r", type=int, default=32)
-    ap.add_argument("--lora_alpha", type=int, default=16)
+    ap.add_argument("--lora_r", type=int, default=0,
+                    help="0 = read from the adapter's adapter_config.json")
+    ap.add_argument("--lora_alpha", type=float, default=0.0,
+                    help="0 = read from the adapter's adapter_config.json")
     ap.add_argument("--seed", type=int, default=3407,
                     help="random-init seed when --model is a name")
     args = ap.parse_args()
 
     spec = resolve_spec(args.model)
-    model = CausalLM(spec, lora_r=args.lora_r, lora_alpha=args.lora_alpha,
+    r, alpha = args.lora_r, args.lora_alpha
+    if r <= 0 or alpha <= 0:
+        from distrl_llm_amd.models.lora import adapter_hyperparams
+        ar, aa, _ = adapter_hyperparams(args.adapter)
+        r = r if r > 0 else ar
+        alpha = alpha if alpha > 0 else aa
+    model = CausalLM(spec, lora_r=r, lora_alpha=alpha,
                      dtype=torch.float32)
     model.random_init(args.seed)
     if is_hf_checkpoint_dir(args.model):

@@ -39,10 +39,11 @@ def test_train_merge_serve_journey(tmp_path):
 
     # 2. merge via the CLI script
     merged_dir = str(tmp_path / "merged")
+    # rank/alpha auto-read from the adapter's adapter_config.json
     out = subprocess.run(
         [sys.executable, os.path.join(REPO, "scripts/merge_adapter.py"),
          "--model", base_dir, "--adapter", adapter_dir,
-         "--out", merged_dir, "--lora_r", "4", "--lora_alpha", "8"],
+         "--out", merged_dir],
         capture_output=True, text=True, timeout=300, cwd=REPO)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "merged 14 projection sites" in out.stdout  # 7 sites x 2 layers
