@@ -76,8 +76,17 @@ def build_worker(rank: int, world_size: int, config: Dict,
         # warm-start from a PEFT adapter directory (every rank loads the
         # same file, so replicas stay identical); --resume implies loading
         # the checkpoint's adapter
-        from ..models.lora import load_adapter
-        load_adapter(model, config.get("resume") or config["load_adapter"])
+        from ..models.lora import adapter_hyperparams, load_adapter
+        adir = config.get("resume") or config["load_adapter"]
+        ar, aa, _ad = adapter_hyperparams(adir)
+        if ar != config["max_lora_rank"] or aa != config["lora_alpha"]:
+            raise ValueError(
+                f"adapter at {adir} was trained with r={ar}, alpha={aa} "
+                f"but the run is configured with "
+                f"--max_lora_rank {config['max_lora_rank']} "
+                f"--lora_alpha {config['lora_alpha']} — pass the matching "
+                f"values")
+        load_adapter(model, adir)
 
     tokenizer = load_tokenizer(config["model"], spec.vocab_size)
 

@@ -177,3 +177,36 @@ def test_mid_episode_resume_bit_identical(tmp_path):
     # guard against the degenerate all-zero-advantage case: training must
     # actually have updated the adapter for the comparison to mean anything
     assert moved
+
+
+def test_resume_rank_mismatch_clear_error(tmp_path):
+    """Resuming with a different --max_lora_rank than the checkpoint's
+    adapter fails with an explicit message, not a shape error."""
+    import torch
+
+    from distrl_llm_amd.models import CausalLM, get_spec
+    from distrl_llm_amd.models.lora import save_adapter
+    from distrl_llm_amd.parallel.worker import build_worker
+
+    m = CausalLM(get_spec("tiny-qwen2"), lora_r=4, lora_alpha=8,
+                 dtype=torch.float32).random_init(0)
+    adir = str(tmp_path / "a")
+    save_adapter(m, adir, "tiny-qwen2", r=4, alpha=8)
+
+    config = {
+        "run_name": "x", "lora_save_path": str(tmp_path / "s"),
+        "lr": 1e-3, "max_prompt_tokens": 16, "max_new_tokens": 8,
+        "episodes": 1, "num_candidates": 2, "batch_size": 2,
+        "train_batch_size": 2, "temperature": 1.0, "save_every": 0,
+        "eval_every": 0, "model": "tiny-qwen2", "dataset": "synthetic",
+        "number_of_actors": 0, "number_of_learners": 1, "learner": "grpo",
+        "max_lora_rank": 8,  # adapter has r=4
+        "topk": 2, "learner_chunk_size": 2, "actor_gpu_usage": 0.9,
+        "learner_gpu_usage": 0.35, "lora_alpha": 8, "lora_dropout": 0.0,
+        "seed": 1, "load_adapter": adir,
+    }
+    with pytest.raises(ValueError, match="trained with r=4"):
+        build_worker(0, 1, config, device=torch.device("cpu"),
+                     engine_overrides={"num_kv_blocks": 64,
+                                       "kv_block_size": 8,
+                                       "max_seq_length": 64})
