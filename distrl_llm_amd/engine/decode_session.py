@@ -184,3 +184,178 @@ class DecodeSession:
             n_new = final_pos[i] - self.prompt_lens[i]
             results.append(q.output_ids[:1] + out[:n_new, i].tolist())
         return results
+
+
+# --------------------------------------------------------- session cache
+
+class CachedDecodeSession(DecodeSession):
+    """A DecodeSession whose device buffers (and captured hipGraph) are
+    owned by a SessionCache and REUSED across generation waves
+    (docs/ROADMAP.md #4: capture cost ~0.2-0.4 s/round otherwise).
+
+    Shapes are fixed at worst case for the cache key — N padded up, block
+    tables at blocks_for(max_seq_length), out_buf at sp.max_tokens — so a
+    graph captured once stays valid; ``reset`` refills the same storages
+    for each new wave. Lanes beyond the real batch are born finished and
+    point at a per-wave scratch KV block (frozen lanes re-forward
+    idempotently, so their writes land in the scratch block and their
+    outputs are never extracted).
+
+    Gated off by default (DISTRL_GRAPH_CACHE=1): the padding/reset state
+    machine is CPU-tested via DISTRL_FORCE_SESSION; flipping the default
+    needs a GPU validation pass (round-2).
+    """
+
+    def __init__(self, engine, n_pad: int, sp: SamplingParams,
+                 eos_token_id: Optional[int]):
+        # deliberately NOT calling DecodeSession.__init__: buffers are
+        # allocated once at worst-case shapes, then refilled by reset()
+        self.engine = engine
+        self.sp = sp
+        self.eos = eos_token_id
+        dev = engine.device
+        bs = engine.pool.block_size
+        self.n_pad = n_pad
+        max_total = engine.cfg.max_seq_length
+        max_nb = KVCachePool.blocks_for(max_total, bs)
+        self.block_tables = torch.zeros(n_pad, max_nb, dtype=torch.int32,
+                                        device=dev)
+        self.tokens = torch.zeros(n_pad, dtype=torch.long, device=dev)
+        self.positions = torch.zeros(n_pad, dtype=torch.long, device=dev)
+        self.ctx_lens = torch.ones(n_pad, dtype=torch.int32, device=dev)
+        self.limit_pos = torch.zeros(n_pad, dtype=torch.long, device=dev)
+        self.finished = torch.zeros(n_pad, dtype=torch.bool, device=dev)
+        self.step_idx = torch.zeros(1, dtype=torch.long, device=dev)
+        self.seeds = torch.zeros(n_pad, dtype=torch.int64, device=dev)
+        self.out_buf = torch.zeros(max(sp.max_tokens, 1), n_pad,
+                                   dtype=torch.long, device=dev)
+        self.graph = None
+        self.use_graph = dev.type == "cuda"
+        self.seqs: List[Sequence] = []
+        self.prompt_lens: List[int] = []
+        self.max_steps = 0
+        self._scratch_block: Optional[int] = None
+
+    def reset(self, seqs: List[Sequence], scratch_block: Optional[int]):
+        """Refill the cached buffers for a new wave (same storages, so a
+        previously captured graph remains valid)."""
+        engine = self.engine
+        bs = engine.pool.block_size
+        sp = self.sp
+        N = len(seqs)
+        assert N <= self.n_pad
+        max_total = engine.cfg.max_seq_length
+        prompt_lens = [len(q.prompt_ids) for q in seqs]
+        limits = [min(pl + sp.max_tokens, max_total) - 1 for pl in prompt_lens]
+        for q, lim in zip(seqs, limits):
+            need = KVCachePool.blocks_for(lim + 1, bs)
+            while len(q.block_table) < need:
+                q.block_table.append(engine.pool.allocator.alloc())
+
+        self.seqs = seqs
+        self.prompt_lens = list(prompt_lens)
+        self.max_steps = max(lim - pl for lim, pl in zip(limits, prompt_lens))
+        self._scratch_block = scratch_block
+
+        n_pad, dev = self.n_pad, engine.device
+        bt = torch.zeros(n_pad, self.block_tables.shape[1], dtype=torch.int32)
+        for i, q in enumerate(seqs):
+            bt[i, :len(q.block_table)] = torch.tensor(q.block_table,
+                                                      dtype=torch.int32)
+        if N < n_pad:
+            # dummy lanes: frozen from step 0, KV writes land in scratch
+            sb = scratch_block if scratch_block is not None else 0
+            bt[N:, 0] = sb
+        self.block_tables.copy_(bt.to(dev, non_blocking=True))
+
+        def fill(t, vals, pad):
+            host = torch.full((n_pad,), pad, dtype=t.dtype)
+            host[:N] = torch.tensor(vals, dtype=t.dtype)
+            t.copy_(host.to(dev, non_blocking=True))
+
+        fill(self.tokens, [q.output_ids[-1] for q in seqs], 0)
+        fill(self.positions, prompt_lens, 0)
+        fill(self.ctx_lens, [p + 1 for p in prompt_lens], 1)
+        fill(self.limit_pos, limits, 0)  # dummy limit 0 -> finished
+        host_fin = torch.zeros(n_pad, dtype=torch.bool)
+        host_fin[N:] = True
+        self.finished.copy_(host_fin.to(dev, non_blocking=True))
+        self.step_idx.zero_()
+        self.seeds.copy_(torch.randint(0, 2**31 - 1, (n_pad,),
+                                       dtype=torch.int64,
+                                       generator=engine.generator
+                                       ).to(dev, non_blocking=True))
+        self.out_buf.zero_()
+        return self
+
+    def run(self, chunk: int = 64, stream_cb=None):
+        # a graph captured on a previous wave stays valid (same storages)
+        if self.use_graph and self.graph is not None and self.max_steps > 0:
+            steps = 0
+            reported = [0] * len(self.seqs)
+            while steps < self.max_steps:
+                n = min(chunk, self.max_steps - steps)
+                for _ in range(n):
+                    self.graph.replay()
+                steps += n
+                if stream_cb is not None:
+                    pos = self.positions.cpu().tolist()
+                    buf = self.out_buf[:steps].cpu()
+                    for i, q in enumerate(self.seqs):
+                        n_new = pos[i] - self.prompt_lens[i]
+                        if n_new > reported[i]:
+                            stream_cb(q.parent_prompt, q.cand_index,
+                                      buf[reported[i]:n_new, i].tolist())
+                            reported[i] = n_new
+                if bool(self.finished.all()):
+                    break
+            final_pos = self.positions.cpu().tolist()
+            out = self.out_buf.cpu()
+            return [q.output_ids[:1] + out[:final_pos[i] - self.prompt_lens[i],
+                                           i].tolist()
+                    for i, q in enumerate(self.seqs)]
+        return super().run(chunk=chunk, stream_cb=stream_cb)
+
+
+class SessionCache:
+    """Per-engine cache of CachedDecodeSession state keyed on
+    (padded batch size, sampling params, eos). One scratch KV block is
+    borrowed from the pool per wave for the padding lanes and returned
+    after."""
+
+    def __init__(self, engine):
+        self.engine = engine
+        self._cache = {}
+
+    @staticmethod
+    def _pad(n: int) -> int:
+        p = 8
+        while p < n:
+            p *= 2
+        return p
+
+    def acquire(self, seqs: List[Sequence], sp: SamplingParams,
+                eos_token_id: Optional[int]) -> CachedDecodeSession:
+        n_pad = self._pad(len(seqs))
+        key = (n_pad, sp.max_tokens, sp.temperature, sp.top_p, sp.top_k,
+               eos_token_id)
+        sess = self._cache.get(key)
+        if sess is None:
+            sess = CachedDecodeSession(self.engine, n_pad, sp, eos_token_id)
+            self._cache[key] = sess
+        scratch = None
+        if n_pad > len(seqs):
+            scratch = self.engine.pool.allocator.alloc()
+        try:
+            return sess.reset(seqs, scratch)
+        except MemoryError:
+            # per-seq worst-case prealloc failed after the scratch was
+            # taken — return it before falling back to an exact session
+            if scratch is not None:
+                self.engine.pool.allocator.free(scratch)
+            raise
+
+    def release(self, sess: CachedDecodeSession):
+        if sess._scratch_block is not None:
+            self.engine.pool.allocator.free(sess._scratch_block)
+            sess._scratch_block = None

@@ -40,6 +40,7 @@ class Engine:
         self.dtype = model.dtype_
         self.scale = self.spec.head_dim ** -0.5
         self._seq_counter = 0
+        self._session_cache = None  # DISTRL_GRAPH_CACHE=1 lazily creates
         self.generator = torch.Generator(device=self.device)
         self.generator.manual_seed(seed)
 
@@ -219,6 +220,20 @@ class Engine:
         context_lens = torch.tensor(ctx_lens, dtype=torch.int32, device=device)
         return self._decode_forward(input_ids, positions_t, slot_mapping,
                                     block_tables, context_lens)
+
+    def _make_session(self, running, sp, eos_token_id):
+        """DecodeSession per wave; with DISTRL_GRAPH_CACHE=1, reuse
+        cached state buffers + captured graph keyed on (padded N, params)
+        (docs/ROADMAP.md #4; default off pending GPU validation)."""
+        from .decode_session import DecodeSession, SessionCache
+        if os.environ.get("DISTRL_GRAPH_CACHE") == "1":
+            if self._session_cache is None:
+                self._session_cache = SessionCache(self)
+            try:
+                return self._session_cache.acquire(running, sp, eos_token_id)
+            except MemoryError:
+                pass  # pool too tight for the padding scratch block
+        return DecodeSession(self, running, sp, eos_token_id)
 
     def _decode_forward(self, input_ids: torch.Tensor, positions: torch.Tensor,
                         slot_mapping: torch.Tensor, block_tables: torch.Tensor,
@@ -501,9 +516,15 @@ class Engine:
                             raise MemoryError(
                                 "KV pool too small to admit any waiting prompt")
                         break
-                session = DecodeSession(self, running, sp, eos_token_id)
-                with trace_range(f"engine/decode_wave[{len(running)}]"):
-                    outs = session.run(stream_cb=stream_cb)
+                session = self._make_session(running, sp, eos_token_id)
+                try:
+                    with trace_range(f"engine/decode_wave[{len(running)}]"):
+                        outs = session.run(stream_cb=stream_cb)
+                finally:
+                    if self._session_cache is not None:
+                        from .decode_session import CachedDecodeSession
+                        if isinstance(session, CachedDecodeSession):
+                            self._session_cache.release(session)
                 for q, ids in zip(running, outs):
                     results[q.parent_prompt].append(ids)
                     self._finish(q)
