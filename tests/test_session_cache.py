@@ -28,6 +28,16 @@ def cached_session(monkeypatch):
     monkeypatch.setenv("DISTRL_GRAPH_CACHE", "1")
 
 
+def _naive(model, prompt, steps):
+    ids = list(prompt)
+    out = []
+    for _ in range(steps):
+        t = int(model(torch.tensor([ids]))[0, -1].argmax())
+        out.append(t)
+        ids.append(t)
+    return out
+
+
 def _engine(model, seed=0, blocks=256):
     cfg = EngineConfig(max_seq_length=64, kv_block_size=8,
                        num_kv_blocks=blocks, max_num_seqs=32)
@@ -60,9 +70,8 @@ def test_cache_reuses_state_across_waves(model, cached_session):
     assert len(cache._cache) == 2
     assert e.pool.allocator.num_free == 256
     # results still correct on the reused buffers
-    from tests.test_engine_fuzz import _naive_greedy
-    assert r1[0][0] == _naive_greedy(model, [1, 2, 3], 4)
-    assert r2[0][0] == _naive_greedy(model, [4, 5, 6, 7], 4)
+    assert r1[0][0] == _naive(model, [1, 2, 3], 4)
+    assert r2[0][0] == _naive(model, [4, 5, 6, 7], 4)
 
 
 def test_cached_streaming_and_eos(model, cached_session):
@@ -99,6 +108,5 @@ def test_cached_tight_pool_falls_back(model, cached_session, monkeypatch):
     e = _engine(model)
     sp = SamplingParams(max_tokens=4, temperature=0.0, n=1)
     res = e.generate([[1, 2, 3]], sp, eos_token_id=None)
-    from tests.test_engine_fuzz import _naive_greedy
-    assert res[0][0] == _naive_greedy(model, [1, 2, 3], 4)
+    assert res[0][0] == _naive(model, [1, 2, 3], 4)
     assert e.pool.allocator.num_free == 256
