@@ -426,6 +426,13 @@ class Engine:
 
     def _generate_inner(self, prompts, sp, eos_token_id, prefill_token_budget,
                         stream_cb=None):
+        if sp.n > self.cfg.max_num_seqs:
+            # the n-candidate fan-out of one prompt is admitted atomically,
+            # so it can never fit — fail with the real reason instead of
+            # the admission loop's pool-exhaustion error
+            raise ValueError(
+                f"SamplingParams.n={sp.n} exceeds max_num_seqs="
+                f"{self.cfg.max_num_seqs}; raise EngineConfig.max_num_seqs")
         if sp.seed is not None:
             # per-request determinism (vLLM SamplingParams.seed analogue):
             # reseed this call's sampling stream so identical (prompts,

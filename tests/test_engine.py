@@ -142,3 +142,30 @@ def test_nf4_quantized_model_generates():
         for ids in r:
             assert ids == expected
     assert engine.pool.allocator.num_free == 128
+
+
+def test_n_exceeding_max_num_seqs_clear_error(setup):
+    """n-fan-out is admitted atomically per prompt; an impossible n must
+    fail with the real reason, not a pool-exhaustion error."""
+    model, _ = setup
+    cfg = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=256,
+                       max_num_seqs=8)
+    engine = Engine(model, cfg, device=torch.device("cpu"), seed=0)
+    with pytest.raises(ValueError, match="max_num_seqs"):
+        engine.generate([[1, 2, 3]],
+                        SamplingParams(max_tokens=2, temperature=0.0, n=20),
+                        eos_token_id=None)
+    # error path leaves the pool fully drained
+    assert engine.pool.allocator.num_free == 256
+
+
+def test_duplicate_prompts_independent(setup):
+    """Identical prompts in one batch are independent sequences (each
+    prefills its own blocks) and produce identical greedy outputs."""
+    model, engine = setup
+    p = [5, 5, 6, 7]
+    res = engine.generate([p, p, p],
+                          SamplingParams(max_tokens=4, temperature=0.0, n=1),
+                          eos_token_id=None)
+    exp = _naive_greedy(model, p, 4)
+    assert [r[0] for r in res] == [exp, exp, exp]
