@@ -64,3 +64,13 @@ def test_generation_config_translation():
     # do_sample=False means greedy regardless of temperature
     sp2 = GenerationConfig(temperature=0.9, do_sample=False).to_sampling_params()
     assert sp2.temperature == 0.0
+
+
+def test_byte_tokenizer_specials_rendering():
+    """skip_special_tokens=False renders specials visibly; True drops
+    them (the learner/serve paths rely on this split)."""
+    from distrl_llm_amd.utils.tokenizer import ByteTokenizer
+    tok = ByteTokenizer(vocab_size=152064)
+    ids = tok.encode("hi") + [tok.eos_token_id]
+    assert tok.decode(ids, skip_special_tokens=True) == "hi"
+    assert "<|257|>" in tok.decode(ids, skip_special_tokens=False)
