@@ -44,3 +44,28 @@ def test_reward_function_shape():
     assert r[0, 1] == 1.0 and r[1, 1] == 0.0
     assert r[0, 0] > 0.2  # 0.1 soft + xml tag credit (minus trailing penalty)
     assert isinstance(r, np.ndarray)
+
+
+def test_reward_adversarial_strings():
+    """Adversarial completions must never crash the reward stack
+    (reference reward_functions.py handles these the same way: split on
+    literal tags, regex with DOTALL)."""
+    from distrl_llm_amd.rl.rewards import extract_xml_answer, reward_function
+    cases = [
+        "",                                       # empty completion
+        "<answer>",                               # unterminated tag
+        "</answer><answer>",                      # reversed tags
+        "<answer><answer>42</answer></answer>",   # nested tags
+        "<think>é中文</think><answer>∞</answer>",  # unicode
+        "x" * 10000,                              # long no-tag blob
+        "<answer>42</answer><answer>7</answer>",  # two answers
+    ]
+    r = reward_function(cases, ["42"] * len(cases))
+    assert r.shape == (len(cases), 2)
+    import numpy as np
+    assert np.isfinite(r).all()
+    # reference split semantics (reward_functions.py:4-7:
+    # text.split("<answer>")[-1].split("</answer>")[0]): the LAST
+    # <answer> tag wins
+    assert extract_xml_answer("<answer><answer>42</answer></answer>") == "42"
+    assert extract_xml_answer("<answer>42</answer><answer>7</answer>") == "7"
