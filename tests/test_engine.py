@@ -169,3 +169,22 @@ def test_duplicate_prompts_independent(setup):
                           eos_token_id=None)
     exp = _naive_greedy(model, p, 4)
     assert [r[0] for r in res] == [exp, exp, exp]
+
+
+def test_derive_num_blocks_cpu_sizing(setup):
+    """CPU pool sizing: 64 MB budget, floor of 16 blocks, capped at
+    max_num_seqs * blocks_for(max_seq_length)."""
+    model, _ = setup
+    cfg = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=0,
+                       max_num_seqs=4)
+    e = Engine(model, cfg, device=torch.device("cpu"), seed=0)
+    # cap: 4 seqs x 8 blocks each
+    assert e.pool.num_blocks == 4 * 8
+    cfg2 = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=0,
+                        max_num_seqs=100000)
+    e2 = Engine(model, cfg2, device=torch.device("cpu"), seed=0)
+    # budget-bound: 64 MB / bytes-per-block, at least the floor
+    spec = model.spec
+    per_block = (2 * spec.num_layers * 8 * spec.num_kv_heads * spec.head_dim
+                 * 4)  # fp32 on CPU
+    assert e2.pool.num_blocks == max(64 * 1024 * 1024 // per_block, 16)
