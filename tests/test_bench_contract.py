@@ -37,6 +37,25 @@ def test_bench_torchrun_two_rank_contract():
 
 
 @pytest.mark.timeout(600)
+def test_bench_split_pool_roles():
+    """--actors selects split actor/learner pools (bench role
+    arithmetic): 2 ranks as 1 actor + 1 learner."""
+    env = dict(os.environ)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(29100 + os.getpid() % 300),
+         os.path.join(REPO, "bench.py"), "--cpu", "--tiny",
+         "--gpus", "2", "--steps", "1", "--warmup", "0", "--actors", "1"],
+        capture_output=True, text=True, timeout=540, env=env, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    rec = json.loads([l for l in out.stdout.splitlines()
+                      if l.startswith("{")][0])
+    assert "1 actors + 1 learners" in rec["config"]["parallelism"]
+    assert rec["value"] > 0
+
+
+@pytest.mark.timeout(600)
 def test_bench_cpu_tiny_json_contract():
     env = dict(os.environ)
     env.setdefault("MASTER_ADDR", "127.0.0.1")
