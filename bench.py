@@ -191,7 +191,7 @@ def main():
         "scaling": "weak",
         "vs_baseline": (round(samples_per_sec / REFERENCE_SAMPLES_PER_SEC, 3)
                         if not args.tiny and use_cuda else None),
-        "dtype": "bf16",
+        "dtype": "bf16" if use_cuda else "fp32",
         "data": "synthetic",
         "config": {
             "model": model_name,
