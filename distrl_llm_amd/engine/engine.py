@@ -476,7 +476,7 @@ class Engine:
                 reserved += need
                 if batch and batch_tokens + L > prefill_token_budget:
                     break
-                if len(running) + len(batch) * sp.n > self.cfg.max_num_seqs:
+                if len(running) + (len(batch) + 1) * sp.n > self.cfg.max_num_seqs:
                     break
                 waiting.pop(0)
                 self._seq_counter += 1

@@ -34,7 +34,11 @@ def group_advantages(group_rewards: np.ndarray) -> np.ndarray:
 
 def topk_indices(scores: np.ndarray, topk: int) -> np.ndarray:
     """Indices of the top-k scores, in the reference's argsort order
-    (ascending, last k — ties resolved identically)."""
+    (ascending, last k). Ties are resolved by a stable sort here (lowest
+    index first within a tie); the reference's np.argsort default is an
+    unstable quicksort, so tied-reward candidates may differ from it —
+    equivalent in expectation, deterministic here (an improvement, not
+    exact tie parity). Reference: distributed_trainer.py:287."""
     return np.argsort(scores, kind="stable")[-topk:]
 
 

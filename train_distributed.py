@@ -127,7 +127,11 @@ def load_datasets(args, tokenizer):
         n = args.synthetic_dataset if args.synthetic_dataset > 0 else 500
         rows = synthetic_math_dataset(n, seed=args.seed)
     rows = process_dataset(tokenizer, rows, r1_preprompt, postprompt="")
-    # 90/10 split (reference train_distributed.py:44)
+    # Seeded shuffle before the 90/10 split (reference train_distributed.py:44
+    # uses train_test_split which shuffles; an unshuffled tail split would make
+    # eval non-random for datasets ordered by subject/difficulty).
+    import random as _random
+    _random.Random(args.seed).shuffle(rows)
     n_test = max(1, len(rows) // 10)
     return ListDataset(rows[:-n_test], seed=args.seed), ListDataset(rows[-n_test:], seed=args.seed)
 
