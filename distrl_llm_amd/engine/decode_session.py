@@ -281,10 +281,11 @@ class CachedDecodeSession(DecodeSession):
         host_fin[N:] = True
         self.finished.copy_(host_fin.to(dev, non_blocking=True))
         self.step_idx.zero_()
+        # generate on the engine's device: its generator is device-bound
+        # (a CUDA generator cannot seed a CPU-side randint)
         self.seeds.copy_(torch.randint(0, 2**31 - 1, (n_pad,),
-                                       dtype=torch.int64,
-                                       generator=engine.generator
-                                       ).to(dev, non_blocking=True))
+                                       dtype=torch.int64, device=dev,
+                                       generator=engine.generator))
         self.out_buf.zero_()
         return self
 

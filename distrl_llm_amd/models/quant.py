@@ -1,10 +1,12 @@
 """nf4 packing utilities + MFMA fragment prepacking for the gfx950 kernels.
 
-The fused nf4 GEMM consumes weights in B-fragment order (one dword per
-(n-tile, k-step, lane) = the lane's 8 nibbles — see ops/csrc/nf4_gemm.hip).
-Because the base quantizer packs two nibbles per byte K-contiguously, the
-fragment dword is exactly 4 consecutive packed bytes, so prepacking is a
-pure int32 gather (no bit twiddling).
+The fused nf4 GEMM consumes weights in B-fragment order, layout v2: the
+lane's 8 nibble-dwords for one 64-deep K chunk of a wave's 4 n-tiles are
+contiguous ([N/64][K/64][64][8] — see ops/csrc/nf4_gemm.hip), so one chunk
+streams as two dwordx4 loads per lane. Because the base quantizer packs
+two nibbles per byte K-contiguously, each fragment dword is exactly 4
+consecutive packed bytes, so prepacking is a pure int32 gather (no bit
+twiddling).
 """
 
 from __future__ import annotations
@@ -26,16 +28,27 @@ def _frag_maps(ntiles: int, ksteps: int, device):
 def prepack_nf4_fragments(packed: torch.Tensor, absmax: torch.Tensor,
                           N: int, K: int) -> Tuple[torch.Tensor, torch.Tensor]:
     """packed: uint8 (N*K/2,), absmax: fp32 (N*K/64,) — the row-major (N, K)
-    output of ops.reference.quantize_nf4. Returns (w4f int32 flat
-    (ntiles*ksteps*64,), amaxf fp32 (ntiles*(K/64)*16,))."""
-    assert N % 16 == 0 and K % 64 == 0
+    output of ops.reference.quantize_nf4.
+
+    Layout v2 (vectorized chunk loads): the kernel's unit of streaming is
+    one 64-deep K chunk of one wave's 4 n-tiles. All of a lane's data for
+    a chunk is contiguous so it loads as 2x dwordx4 (weights) + 1x dwordx4
+    (absmax):
+
+      w4f  int32 [N/64][K/64][64 lanes][8]   (8 = ks*4 + nt)
+      amaxf fp32 [N/64][K/64][16 lrow][4 nt]
+    """
+    assert N % 64 == 0 and K % 64 == 0
     device = packed.device
     ntiles, ksteps = N // 16, K // 32
+    ngr, nkb = N // 64, K // 64
     pd = packed.contiguous().view(N, K // 2).view(torch.int32)  # (N, K/8)
     n_map, k_map = _frag_maps(ntiles, ksteps, device)
-    w4f = pd[n_map, k_map].contiguous()
-    am = absmax.view(N, K // 64).view(ntiles, 16, K // 64)
-    amaxf = am.permute(0, 2, 1).contiguous().float()
+    w4f1 = pd[n_map, k_map]                       # (ntiles, ksteps, 64)
+    v1 = w4f1.view(ngr, 4, nkb, 2, 64)            # (g4, nt, kb, ks, lane)
+    w4f = v1.permute(0, 2, 4, 3, 1).contiguous()  # (g4, kb, lane, ks, nt)
+    am = absmax.view(N, nkb).view(ngr, 4, 16, nkb)  # (g4, nt, lrow, kb)
+    amaxf = am.permute(0, 3, 2, 1).contiguous().float()  # (g4, kb, lrow, nt)
     return w4f.view(-1), amaxf.view(-1)
 
 

@@ -6,10 +6,12 @@
 //   y[M,N] = x[M,K] @ dequant(W4)[N,K]^T (+ bias) (+ u[M,r] @ Bs[N,r]^T)
 //
 // Weights are PREPACKED into MFMA B-fragment order at load time (we own
-// the format): one dword per (n-tile, k-step, lane) holding the lane's 8
-// nf4 nibbles, so the weight stream is perfectly coalesced 256 B per wave
-// instruction and dequant is 8 LUT-mul-cvt ops straight into the MFMA
-// B fragment. absmax is fragment-ordered fp32. The LoRA correction rides
+// the format): the lane's 8 nibble-dwords for one 64-deep K chunk of a
+// wave's 4 n-tiles are CONTIGUOUS (layout v2: [ngroup4][kchunk][lane][8]),
+// so a whole chunk is two dwordx4 loads per lane (v1 was 8 scalar dword
+// loads) and dequant is 8 LUT-mul-cvt ops straight into the MFMA
+// B fragment. absmax is fragment-ordered fp32 ([ngroup4][kchunk][16][4],
+// one dwordx4 per lane per chunk). The LoRA correction rides
 // the same accumulators: u = x@A^T (computed by the split-K lora_u kernel
 // below) enters as ONE extra MFMA k-step per rank-32 block against the
 // bf16-prepacked, scale-folded B matrix — the adapter stays exact bf16
@@ -31,6 +33,7 @@
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16v8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
 namespace {
 
@@ -49,11 +52,12 @@ DEV_INLINE bf16v8 lds_read_frag(const char* base, int byte_off) {
   return *reinterpret_cast<const bf16v8*>(base + byte_off);
 }
 
-template <int MT, int DBG = 0>  // DBG: 1 = skip LUT dequant, 2 = skip x LDS
+template <int MT, int DBG = 0, bool NT = false>
+// DBG: 1 = skip LUT dequant, 2 = skip x LDS; NT: non-temporal W stream
 __global__ __launch_bounds__(256)
 void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
-                     const uint32_t* __restrict__ w4f,       // frag-packed
-                     const float* __restrict__ amaxf,        // frag-ordered
+                     const uint32_t* __restrict__ w4f,       // frag v2
+                     const float* __restrict__ amaxf,        // frag v2
                      const __hip_bfloat16* __restrict__ bias,  // (N) | null
                      const float* __restrict__ u,            // (M, r) | null
                      const uint32_t* __restrict__ bfrag,     // B frag | null
@@ -71,19 +75,15 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   // share the same weight panel, so the XCD L2 absorbs re-reads when the
   // m dimension is tiled (T1 locality without an explicit remap)
   const int mbase = blockIdx.x * BM;
-  const int ntile0 = blockIdx.y * 16 + wave * 4;  // this wave's 4 n-tiles
+  const int g4 = blockIdx.y * 4 + wave;           // n-group of 4 tiles
+  const int ntile0 = g4 * 4;                      // this wave's 4 n-tiles
   const int zid = blockIdx.z;                     // split-K slice
-  const int ksteps = K / 32;
+  const int nkb = K / 64;                         // total 64-deep chunks
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // 256-entry paired LUT: entry b = (code[b&15], code[b>>4]) — one
-  // ds_read_b64 + one v_pk_mul dequantizes a whole packed BYTE
   constexpr int SK_ = (MT <= 2) ? 512 : 256;  // x super-panel columns
-  float2* lut2 = reinterpret_cast<float2*>(smem);     // 256 * 8 B
-  char* x_lds = smem + 2048;                  // BM * SK * 2 bytes
+  char* x_lds = smem;                         // BM * SK * 2 bytes
   char* u_lds = x_lds + BM * SK_ * 2;         // BM * r * 2 bytes
-  if (tid < 256)
-    lut2[tid] = make_float2(NF4_LUT[tid & 15], NF4_LUT[tid >> 4]);
   // per-lane register copy of the codebook for the shuffle-LUT dequant
   // (a pure-VALU cndmask-tree variant measured 1.5x SLOWER: ~9 VALU/value
   // beats ds_bpermute's latency only on paper — the compiler pipelines
@@ -114,36 +114,46 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
     }
   }
 
-  // ---- main K loop: 64-deep chunks with one-chunk-ahead weight
-  // prefetch. Without it the compiler issues each weight dword right
-  // before its use with an immediate vmcnt wait — every fragment pays
-  // full HBM latency serially (measured 8-20x slowdown).
-  const int nkb = K / 64;
-  uint32_t wb_cur[8] = {}, wb_nxt[8] = {}, wb_nx2[8] = {};
-  float am_cur[8] = {}, am_nxt[8] = {}, am_nx2[8] = {};
+  // ---- main K loop: 64-deep chunks streamed through a static 4-slot
+  // register ring (3 chunks prefetched ahead). One chunk = 2 dwordx4 W
+  // loads + 1 dwordx4 absmax load per lane (layout v2) — hipcc's counted
+  // vmcnt bookkeeping then keeps up to 9 loads in flight. Without
+  // prefetch the compiler issues each weight dword right before its use
+  // behind vmcnt(0) — every fragment pays full HBM latency serially
+  // (measured 8-20x slowdown).
+  uint32_t wb[4][8];
+  f32x4 am[4];
 
-  #define LOAD_WCHUNK(KB, WB, AM)                                         \
-    _Pragma("unroll")                                                     \
-    for (int ks = 0; ks < 2; ++ks) {                                      \
-      _Pragma("unroll")                                                   \
-      for (int nt = 0; nt < 4; ++nt) {                                    \
-        const int kstep = (KB) * 2 + ks;                                  \
-        const int ntg = ntile0 + nt;                                      \
-        WB[ks * 4 + nt] = w4f[((int64_t)ntg * ksteps + kstep) * 64 + l];  \
-        AM[ks * 4 + nt] = amaxf[((int64_t)ntg * (K / 64)                  \
-                                 + (kstep >> 1)) * 16 + lrow];            \
+  #define LOAD_WCHUNK(KB, RS)                                             \
+    {                                                                     \
+      const int64_t cb_ = (int64_t)g4 * nkb + (KB);                       \
+      const u32x4* wp_ = reinterpret_cast<const u32x4*>(w4f)              \
+                         + cb_ * 128 + l * 2;                             \
+      if constexpr (NT) {                                                 \
+        *reinterpret_cast<u32x4*>(&wb[RS][0]) =                           \
+            __builtin_nontemporal_load(wp_);                              \
+        *reinterpret_cast<u32x4*>(&wb[RS][4]) =                           \
+            __builtin_nontemporal_load(wp_ + 1);                          \
+        am[RS] = __builtin_nontemporal_load(                              \
+            reinterpret_cast<const f32x4*>(amaxf + cb_ * 64) + lrow);     \
+      } else {                                                            \
+        *reinterpret_cast<u32x4*>(&wb[RS][0]) = wp_[0];                   \
+        *reinterpret_cast<u32x4*>(&wb[RS][4]) = wp_[1];                   \
+        am[RS] = reinterpret_cast<const f32x4*>(amaxf + cb_ * 64)[lrow];  \
       }                                                                   \
     }
 
   // this slice's super-panel range (split-K over whole panels)
   constexpr int SKC = ((MT <= 2) ? 512 : 256) / 64;  // chunks per panel
+  static_assert(SKC % 4 == 0, "ring phase must stay aligned per panel");
   const int npanels = K / ((MT <= 2) ? 512 : 256);
   const int sp0 = zid * nsp_per;
   const int sp1 = min(npanels, sp0 + nsp_per);
   const int kb0 = sp0 * SKC;
   const int kb_end = sp1 * SKC;
-  if (kb0 < kb_end) LOAD_WCHUNK(kb0, wb_cur, am_cur);
-  if (kb0 + 1 < kb_end) LOAD_WCHUNK(kb0 + 1, wb_nxt, am_nxt);
+  if (kb0 < kb_end) LOAD_WCHUNK(kb0, 0);
+  if (kb0 + 1 < kb_end) LOAD_WCHUNK(kb0 + 1, 1);
+  if (kb0 + 2 < kb_end) LOAD_WCHUNK(kb0 + 2, 2);
 
   // x is staged in SUPER-panels of SK columns: one barrier pair per
   // SK/64 weight chunks, so the chunk loop in between runs barrier-free
@@ -184,11 +194,14 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
     }
     __syncthreads();
 
-    for (int kc = 0; kc < SK / 64; ++kc) {
-      const int kb = (sk + kc * 64) / 64;
-      if (kb + 2 < kb_end) {
-        LOAD_WCHUNK(kb + 2, wb_nx2, am_nx2);
-      }
+    #pragma unroll
+    for (int kc = 0; kc < SKC; ++kc) {
+      const int kb = sp * SKC + kc;
+      // ring slot indices are compile-time: SKC % 4 == 0 keeps the
+      // panel-relative phase aligned, so slot = kc & 3 in the unrolled
+      // body (no register-shuffle rotation between chunks)
+      if (kb + 3 < kb_end) LOAD_WCHUNK(kb + 3, (kc + 3) & 3);
+      const int cur = kc & 3;
       #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         // A fragments for every m-tile
@@ -209,13 +222,13 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
         // 4 n-tiles: dequant B fragment + MFMA
         #pragma unroll
         for (int nt = 0; nt < 4; ++nt) {
-          const uint32_t wbits = wb_cur[ks * 4 + nt];
-          const float am = am_cur[ks * 4 + nt];
+          const uint32_t wbits = wb[cur][ks * 4 + nt];
+          const float amv = am[cur][nt];
           bf16v8 bfr;
           if constexpr (DBG == 1) {
             #pragma unroll
             for (int j = 0; j < 8; ++j)
-              bfr[j] = (__bf16)(am + (float)(wbits & 1));
+              bfr[j] = (__bf16)(amv + (float)(wbits & 1));
           } else {
             // register LUT via cross-lane shuffle (ds_bpermute):
             // lanes 0-15 hold the 16 nf4 codes; no LDS traffic, no
@@ -223,7 +236,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
             #pragma unroll
             for (int j = 0; j < 8; ++j) {
               const int nib = (wbits >> (4 * j)) & 0xF;
-              bfr[j] = (__bf16)(__shfl(lut_reg, nib, WAVE) * am);
+              bfr[j] = (__bf16)(__shfl(lut_reg, nib, WAVE) * amv);
             }
           }
           #pragma unroll
@@ -231,13 +244,6 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
             acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 afrag[mt], bfr, acc[mt][nt], 0, 0, 0);
         }
-      }
-      #pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        wb_cur[i] = wb_nxt[i];
-        am_cur[i] = am_nxt[i];
-        wb_nxt[i] = wb_nx2[i];
-        am_nxt[i] = am_nx2[i];
       }
     }
   }
@@ -436,7 +442,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   const int nsp_per = (npanels + ksplit - 1) / ksplit;
   ksplit = (npanels + nsp_per - 1) / nsp_per;
   dim3 grid((M + BM - 1) / BM, N / 256, ksplit), block(256);
-  size_t smem = 2048 + (size_t)BM * SK * 2
+  size_t smem = (size_t)BM * SK * 2
                 + (has_lora ? (size_t)BM * r * 2 : 0);
   if (getenv("DISTRL_NF4_LDSPAD")) smem += 8192;  // debug: OOB guard
   torch::Tensor ws;
@@ -457,28 +463,26 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
 
   int dbg = 0;
   if (const char* e = getenv("DISTRL_NF4_DBG")) dbg = atoi(e);
-  if (dbg == 1 && mt == 1) {
-    hipLaunchKernelGGL((nf4_gemm_kernel<1, 1>), grid, block, smem, stream,
-        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-        reinterpret_cast<const uint32_t*>(w4f.data_ptr()),
-        amaxf.data_ptr<float>(), bias_p, u_p, bf_p,
-        reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), ws_p,
-        M, (int)N, (int)K, (int)r, u_stride, nsp_per);
-  } else if (dbg == 2 && mt == 1) {
-    hipLaunchKernelGGL((nf4_gemm_kernel<1, 2>), grid, block, smem, stream,
-        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-        reinterpret_cast<const uint32_t*>(w4f.data_ptr()),
-        amaxf.data_ptr<float>(), bias_p, u_p, bf_p,
-        reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), ws_p,
-        M, (int)N, (int)K, (int)r, u_stride, nsp_per);
-  } else
-  #define LAUNCH(MTV) \
-    hipLaunchKernelGGL((nf4_gemm_kernel<MTV>), grid, block, smem, stream, \
+  // non-temporal W stream (weights are read once per CU at decode
+  // shapes); default on, DISTRL_NF4_NT=0 reverts to cached loads
+  bool nt = true;
+  if (const char* e = getenv("DISTRL_NF4_NT")) nt = atoi(e) != 0;
+  #define LAUNCH_1(MTV, DBGV, NTV) \
+    hipLaunchKernelGGL((nf4_gemm_kernel<MTV, DBGV, NTV>), grid, block, smem, \
+        stream, \
         reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()), \
         reinterpret_cast<const uint32_t*>(w4f.data_ptr()), \
         amaxf.data_ptr<float>(), bias_p, u_p, bf_p, \
         reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), ws_p, \
         M, (int)N, (int)K, (int)r, u_stride, nsp_per)
+  #define LAUNCH(MTV) \
+    do { if (nt) LAUNCH_1(MTV, 0, true); else LAUNCH_1(MTV, 0, false); } \
+    while (0)
+  if (dbg == 1 && mt == 1) {
+    LAUNCH_1(1, 1, false);
+  } else if (dbg == 2 && mt == 1) {
+    LAUNCH_1(1, 2, false);
+  } else
   switch (mt) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -487,6 +491,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
     default: LAUNCH(5); break;
   }
   #undef LAUNCH
+  #undef LAUNCH_1
   if (ksplit > 1 && getenv("DISTRL_NF4_RETWS")) return ws;  // debug
   if (ksplit > 1) {
     const int64_t mn = (int64_t)M * N;
