@@ -30,25 +30,25 @@ def prepack_nf4_fragments(packed: torch.Tensor, absmax: torch.Tensor,
     """packed: uint8 (N*K/2,), absmax: fp32 (N*K/64,) — the row-major (N, K)
     output of ops.reference.quantize_nf4.
 
-    Layout v2 (vectorized chunk loads): the kernel's unit of streaming is
-    one 64-deep K chunk of one wave's 4 n-tiles. All of a lane's data for
-    a chunk is contiguous so it loads as 2x dwordx4 (weights) + 1x dwordx4
+    Layout v3 (vectorized chunk loads): the kernel's unit of streaming is
+    one 64-deep K chunk of one wave's 2 n-tiles. All of a lane's data for
+    a chunk is contiguous so it loads as 1x dwordx4 (weights) + 1x dwordx2
     (absmax):
 
-      w4f  int32 [N/64][K/64][64 lanes][8]   (8 = ks*4 + nt)
-      amaxf fp32 [N/64][K/64][16 lrow][4 nt]
+      w4f  int32 [N/32][K/64][64 lanes][4]   (4 = ks*2 + nt)
+      amaxf fp32 [N/32][K/64][16 lrow][2 nt]
     """
-    assert N % 64 == 0 and K % 64 == 0
+    assert N % 32 == 0 and K % 64 == 0
     device = packed.device
     ntiles, ksteps = N // 16, K // 32
-    ngr, nkb = N // 64, K // 64
+    ngr, nkb = N // 32, K // 64
     pd = packed.contiguous().view(N, K // 2).view(torch.int32)  # (N, K/8)
     n_map, k_map = _frag_maps(ntiles, ksteps, device)
     w4f1 = pd[n_map, k_map]                       # (ntiles, ksteps, 64)
-    v1 = w4f1.view(ngr, 4, nkb, 2, 64)            # (g4, nt, kb, ks, lane)
-    w4f = v1.permute(0, 2, 4, 3, 1).contiguous()  # (g4, kb, lane, ks, nt)
-    am = absmax.view(N, nkb).view(ngr, 4, 16, nkb)  # (g4, nt, lrow, kb)
-    amaxf = am.permute(0, 3, 2, 1).contiguous().float()  # (g4, kb, lrow, nt)
+    v1 = w4f1.view(ngr, 2, nkb, 2, 64)            # (g2, nt, kb, ks, lane)
+    w4f = v1.permute(0, 2, 4, 3, 1).contiguous()  # (g2, kb, lane, ks, nt)
+    am = absmax.view(N, nkb).view(ngr, 2, 16, nkb)  # (g2, nt, lrow, kb)
+    amaxf = am.permute(0, 3, 2, 1).contiguous().float()  # (g2, kb, lrow, nt)
     return w4f.view(-1), amaxf.view(-1)
 
 

@@ -6,12 +6,12 @@
 //   y[M,N] = x[M,K] @ dequant(W4)[N,K]^T (+ bias) (+ u[M,r] @ Bs[N,r]^T)
 //
 // Weights are PREPACKED into MFMA B-fragment order at load time (we own
-// the format): the lane's 8 nibble-dwords for one 64-deep K chunk of a
-// wave's 4 n-tiles are CONTIGUOUS (layout v2: [ngroup4][kchunk][lane][8]),
-// so a whole chunk is two dwordx4 loads per lane (v1 was 8 scalar dword
+// the format): the lane's 4 nibble-dwords for one 64-deep K chunk of a
+// wave's 2 n-tiles are CONTIGUOUS (layout v3: [ngroup2][kchunk][lane][4]),
+// so a whole chunk is ONE dwordx4 load per lane (v1 was 8 scalar dword
 // loads) and dequant is 8 LUT-mul-cvt ops straight into the MFMA
-// B fragment. absmax is fragment-ordered fp32 ([ngroup4][kchunk][16][4],
-// one dwordx4 per lane per chunk). The LoRA correction rides
+// B fragment. absmax is fragment-ordered fp32 ([ngroup2][kchunk][16][2],
+// one dwordx2 per lane per chunk). The LoRA correction rides
 // the same accumulators: u = x@A^T (computed by the split-K lora_u kernel
 // below) enters as ONE extra MFMA k-step per rank-32 block against the
 // bf16-prepacked, scale-folded B matrix — the adapter stays exact bf16
@@ -23,16 +23,21 @@
 //   B: lane l holds B[col = l&15][k = (l>>4)*8 + j]   (B consumed as N x K)
 //   D: lane l, reg r -> row = (l>>4)*4 + r, col = l&15
 //
-// Geometry: 256 threads = 4 waves; block tile = BN 256 cols (4 waves x 4
-// n-tiles) x BM = 16*MT rows; K loop in 64-deep chunks with the x tile
-// staged in LDS behind an XOR swizzle (byte ^= (row&15)<<4) so the
-// ds_read_b128 A-fragment reads are conflict-free (guide G4/T2).
+// Geometry: NW waves of 2 n-tiles each (NW=8 -> BN 256, 512 threads,
+// 1 block/CU; NW=4 -> BN 128, 256 threads, 2 blocks/CU) x BM = 16*MT
+// rows; K loop in 64-deep chunks with the x tile staged in LDS behind an
+// XOR swizzle (byte ^= (row&15)<<4) so the ds_read_b128 A-fragment reads
+// are conflict-free (guide G4/T2). The 2-tile wave keeps total VGPR+AGPR
+// under 256 -> 2 waves/SIMD, which is what hides the ds_bpermute dequant
+// latency (the 4-tile variant allocated 338 regs = 1 wave/SIMD and ran
+// ~2x slower end-to-end).
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16v8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(2))) float f32x2;
 typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
 namespace {
@@ -52,12 +57,13 @@ DEV_INLINE bf16v8 lds_read_frag(const char* base, int byte_off) {
   return *reinterpret_cast<const bf16v8*>(base + byte_off);
 }
 
-template <int MT, int DBG = 0, bool NT = false>
-// DBG: 1 = skip LUT dequant, 2 = skip x LDS; NT: non-temporal W stream
-__global__ __launch_bounds__(256)
+template <int MT, int NW, int DBG = 0>
+// NW = waves per block (each wave owns TWO n-tiles; BN = NW*32);
+// DBG: 1 = skip LUT dequant, 2 = skip x LDS
+__global__ __launch_bounds__(NW * 64)
 void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
-                     const uint32_t* __restrict__ w4f,       // frag v2
-                     const float* __restrict__ amaxf,        // frag v2
+                     const uint32_t* __restrict__ w4f,       // frag v3
+                     const float* __restrict__ amaxf,        // frag v3
                      const __hip_bfloat16* __restrict__ bias,  // (N) | null
                      const float* __restrict__ u,            // (M, r) | null
                      const uint32_t* __restrict__ bfrag,     // B frag | null
@@ -66,6 +72,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
                      int M, int N, int K, int r, int u_stride,
                      int nsp_per) {
   constexpr int BM = 16 * MT;
+  constexpr int THREADS = NW * 64;
   const int tid = threadIdx.x;
   const int l = tid & 63;
   const int wave = tid >> 6;
@@ -75,8 +82,8 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   // share the same weight panel, so the XCD L2 absorbs re-reads when the
   // m dimension is tiled (T1 locality without an explicit remap)
   const int mbase = blockIdx.x * BM;
-  const int g4 = blockIdx.y * 4 + wave;           // n-group of 4 tiles
-  const int ntile0 = g4 * 4;                      // this wave's 4 n-tiles
+  const int g2 = blockIdx.y * NW + wave;          // n-group of 2 tiles
+  const int ntile0 = g2 * 2;                      // this wave's 2 n-tiles
   const int zid = blockIdx.z;                     // split-K slice
   const int nkb = K / 64;                         // total 64-deep chunks
 
@@ -90,15 +97,15 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   // the shuffles ~2-deep and the VALU tree just adds issue pressure)
   const float lut_reg = NF4_LUT[tid & 15];
 
-  f32x4 acc[MT][4];
+  f32x4 acc[MT][2];
   #pragma unroll
   for (int mt = 0; mt < MT; ++mt)
     #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) acc[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int nt = 0; nt < 2; ++nt) acc[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   // stage u (fp32 -> bf16) once; it is tiny (BM x r)
   if (u != nullptr) {
-    for (int i = tid; i < BM * (r / 8); i += 256) {
+    for (int i = tid; i < BM * (r / 8); i += THREADS) {
       const int row = i / (r / 8);
       const int unit = i % (r / 8);
       bf16x8 s;
@@ -115,32 +122,23 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   }
 
   // ---- main K loop: 64-deep chunks streamed through a static 4-slot
-  // register ring (3 chunks prefetched ahead). One chunk = 2 dwordx4 W
-  // loads + 1 dwordx4 absmax load per lane (layout v2) — hipcc's counted
-  // vmcnt bookkeeping then keeps up to 9 loads in flight. Without
-  // prefetch the compiler issues each weight dword right before its use
-  // behind vmcnt(0) — every fragment pays full HBM latency serially
-  // (measured 8-20x slowdown).
-  uint32_t wb[4][8];
-  f32x4 am[4];
+  // register ring (3 chunks prefetched ahead). One chunk = 1 dwordx4 W
+  // load + 1 dwordx2 absmax load per lane (layout v3) — hipcc's counted
+  // vmcnt bookkeeping keeps up to 8 loads in flight, and the 2-tile wave
+  // keeps total registers low enough for 2 waves/SIMD (the co-resident
+  // wave is what hides the dequant-shuffle latency). Without prefetch the
+  // compiler issues each weight dword right before its use behind
+  // vmcnt(0) — every fragment pays full HBM latency serially (measured
+  // 8-20x slowdown).
+  uint32_t wb[4][4];
+  f32x2 am[4];
 
   #define LOAD_WCHUNK(KB, RS)                                             \
     {                                                                     \
-      const int64_t cb_ = (int64_t)g4 * nkb + (KB);                       \
-      const u32x4* wp_ = reinterpret_cast<const u32x4*>(w4f)              \
-                         + cb_ * 128 + l * 2;                             \
-      if constexpr (NT) {                                                 \
-        *reinterpret_cast<u32x4*>(&wb[RS][0]) =                           \
-            __builtin_nontemporal_load(wp_);                              \
-        *reinterpret_cast<u32x4*>(&wb[RS][4]) =                           \
-            __builtin_nontemporal_load(wp_ + 1);                          \
-        am[RS] = __builtin_nontemporal_load(                              \
-            reinterpret_cast<const f32x4*>(amaxf + cb_ * 64) + lrow);     \
-      } else {                                                            \
-        *reinterpret_cast<u32x4*>(&wb[RS][0]) = wp_[0];                   \
-        *reinterpret_cast<u32x4*>(&wb[RS][4]) = wp_[1];                   \
-        am[RS] = reinterpret_cast<const f32x4*>(amaxf + cb_ * 64)[lrow];  \
-      }                                                                   \
+      const int64_t cb_ = (int64_t)g2 * nkb + (KB);                       \
+      *reinterpret_cast<u32x4*>(&wb[RS][0]) =                             \
+          reinterpret_cast<const u32x4*>(w4f)[cb_ * 64 + l];              \
+      am[RS] = reinterpret_cast<const f32x2*>(amaxf + cb_ * 32)[lrow];    \
     }
 
   // this slice's super-panel range (split-K over whole panels)
@@ -161,7 +159,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   // a vmcnt(0) on any in-loop ds_write/barrier path — guide §5 traps).
   constexpr int SK = SK_;                     // host asserts K % SK == 0
   constexpr int SKU = SK / 8;                 // bf16x8 units per row
-  constexpr int XIT = (BM * SKU + 255) / 256; // staging iters (VGPR-resident)
+  constexpr int XIT = (BM * SKU + THREADS - 1) / THREADS;  // staging iters
 
   for (int sp = sp0; sp < sp1; ++sp) {
     const int sk = sp * SK;
@@ -169,7 +167,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
     bf16x8 xs[XIT];
     #pragma unroll
     for (int it = 0; it < XIT; ++it) {
-      const int i = tid + it * 256;
+      const int i = tid + it * THREADS;
       if (i < BM * SKU) {
         const int row = i / SKU;
         if (mbase + row < M) {
@@ -184,7 +182,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
     __syncthreads();  // previous panel fully consumed
     #pragma unroll
     for (int it = 0; it < XIT; ++it) {
-      const int i = tid + it * 256;
+      const int i = tid + it * THREADS;
       if (i < BM * SKU) {
         const int row = i / SKU;
         const int off = (row * (SK * 2) + (i % SKU) * 16)
@@ -219,10 +217,10 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
             afrag[mt] = lds_read_frag(x_lds, off);
           }
         }
-        // 4 n-tiles: dequant B fragment + MFMA
+        // 2 n-tiles: dequant B fragment + MFMA
         #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) {
-          const uint32_t wbits = wb[cur][ks * 4 + nt];
+        for (int nt = 0; nt < 2; ++nt) {
+          const uint32_t wbits = wb[cur][ks * 2 + nt];
           const float amv = am[cur][nt];
           bf16v8 bfr;
           if constexpr (DBG == 1) {
@@ -262,7 +260,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
             u_lds + (int64_t)row * r * 2 + (rs * 32 + lk * 8) * 2);
       }
       #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < 2; ++nt) {
         const int ntg = ntile0 + nt;
         const uint4 bw = *reinterpret_cast<const uint4*>(
             bfrag + ((int64_t)ntg * (r / 32) + rs) * 64 * 4 + l * 4);
@@ -277,7 +275,7 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
 
   // ---- epilogue: bias + store (fp32 slab under split-K) ----
   #pragma unroll
-  for (int nt = 0; nt < 4; ++nt) {
+  for (int nt = 0; nt < 2; ++nt) {
     const int n = (ntile0 + nt) * 16 + lrow;
     const float bv = (bias != nullptr && zid == 0) ? bf2f(bias[n]) : 0.f;
     #pragma unroll
@@ -407,7 +405,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   TORCH_CHECK(w4f.scalar_type() == at::kInt || w4f.scalar_type() == at::kUInt32);
   const int M = x.size(0);
   TORCH_CHECK(x.size(1) == K);
-  TORCH_CHECK(N % 256 == 0, "nf4_gemm: N must be a multiple of 256, got ", N);
+  TORCH_CHECK(N % 128 == 0, "nf4_gemm: N must be a multiple of 128, got ", N);
   TORCH_CHECK(K % 64 == 0);
   auto y = torch::empty({(int64_t)M, N}, x.options());
   if (M == 0) return y;
@@ -420,18 +418,26 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
     u_stride = u->stride(0);
   }
 
-  // m-tile size: MT5 costs 232 regs (2 waves/SIMD); MT3/4 fit 3 waves/SIMD
+  // m-tile size: MT5 + 2-tile waves fits 2 waves/SIMD
   int mtcap = 5;
   if (const char* e = getenv("DISTRL_NF4_MTCAP")) mtcap = atoi(e);
   int mt = std::min<int>((M + 15) / 16, std::max(1, mtcap));
   const int BM = 16 * mt;
   const int SK = (mt <= 2) ? 512 : 256;  // must mirror the kernel constexpr
   TORCH_CHECK(K % SK == 0, "nf4_gemm: K (", K, ") % ", SK, " != 0");
+  // waves per block: wide (8-wave, BN=256) blocks when N alone nearly
+  // fills the chip at 1 block/CU; narrow (4-wave, BN=128, 2 blocks/CU)
+  // blocks for small N so the grid still fills without deep split-K
+  int nw = (N % 256 == 0 && (N / 256) * ((M + BM - 1) / BM) >= 192) ? 8 : 4;
+  if (const char* e = getenv("DISTRL_NF4_NW")) nw = atoi(e);
+  TORCH_CHECK(nw == 4 || nw == 8);
+  TORCH_CHECK(N % (nw * 32) == 0);
+  const int nblk = (int)N / (nw * 32);
   // split-K over super-panels until the grid fills the chip (the skinny
-  // decode shapes otherwise run at ~1 wave/SIMD, fully latency-exposed)
+  // decode shapes otherwise run at ~1 block-wave, fully latency-exposed)
   const int npanels = (int)K / SK;
-  const int base_blocks = ((M + BM - 1) / BM) * ((int)N / 256);
-  int blk_target = 1024;
+  const int base_blocks = ((M + BM - 1) / BM) * nblk;
+  int blk_target = (nw == 8) ? 384 : 768;
   if (const char* e = getenv("DISTRL_NF4_BLKTGT")) blk_target = atoi(e);
   int ksplit = 1;
   while (ksplit * 2 <= npanels && base_blocks * ksplit < blk_target
@@ -441,7 +447,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   ksplit = std::max(1, std::min(ksplit, npanels));
   const int nsp_per = (npanels + ksplit - 1) / ksplit;
   ksplit = (npanels + nsp_per - 1) / nsp_per;
-  dim3 grid((M + BM - 1) / BM, N / 256, ksplit), block(256);
+  dim3 grid((M + BM - 1) / BM, nblk, ksplit), block(nw * 64);
   size_t smem = (size_t)BM * SK * 2
                 + (has_lora ? (size_t)BM * r * 2 : 0);
   if (getenv("DISTRL_NF4_LDSPAD")) smem += 8192;  // debug: OOB guard
@@ -463,12 +469,8 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
 
   int dbg = 0;
   if (const char* e = getenv("DISTRL_NF4_DBG")) dbg = atoi(e);
-  // non-temporal W stream (weights are read once per CU at decode
-  // shapes); default on, DISTRL_NF4_NT=0 reverts to cached loads
-  bool nt = true;
-  if (const char* e = getenv("DISTRL_NF4_NT")) nt = atoi(e) != 0;
-  #define LAUNCH_1(MTV, DBGV, NTV) \
-    hipLaunchKernelGGL((nf4_gemm_kernel<MTV, DBGV, NTV>), grid, block, smem, \
+  #define LAUNCH_1(MTV, NWV, DBGV) \
+    hipLaunchKernelGGL((nf4_gemm_kernel<MTV, NWV, DBGV>), grid, block, smem, \
         stream, \
         reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()), \
         reinterpret_cast<const uint32_t*>(w4f.data_ptr()), \
@@ -476,12 +478,12 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
         reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), ws_p, \
         M, (int)N, (int)K, (int)r, u_stride, nsp_per)
   #define LAUNCH(MTV) \
-    do { if (nt) LAUNCH_1(MTV, 0, true); else LAUNCH_1(MTV, 0, false); } \
+    do { if (nw == 8) LAUNCH_1(MTV, 8, 0); else LAUNCH_1(MTV, 4, 0); } \
     while (0)
-  if (dbg == 1 && mt == 1) {
-    LAUNCH_1(1, 1, false);
-  } else if (dbg == 2 && mt == 1) {
-    LAUNCH_1(1, 2, false);
+  if (dbg == 1 && mt == 1 && nw == 4) {
+    LAUNCH_1(1, 4, 1);
+  } else if (dbg == 2 && mt == 1 && nw == 4) {
+    LAUNCH_1(1, 4, 2);
   } else
   switch (mt) {
     case 1: LAUNCH(1); break;
