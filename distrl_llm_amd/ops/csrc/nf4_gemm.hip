@@ -484,6 +484,10 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
     LAUNCH_1(1, 4, 1);
   } else if (dbg == 2 && mt == 1 && nw == 4) {
     LAUNCH_1(1, 4, 2);
+  } else if (dbg == 1 && mt == 5) {        // ablation: no LUT dequant
+    if (nw == 8) LAUNCH_1(5, 8, 1); else LAUNCH_1(5, 4, 1);
+  } else if (dbg == 2 && mt == 5) {        // ablation: no x LDS reads
+    if (nw == 8) LAUNCH_1(5, 8, 2); else LAUNCH_1(5, 4, 2);
   } else
   switch (mt) {
     case 1: LAUNCH(1); break;
