@@ -88,14 +88,20 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   const int nkb = K / 64;                         // total 64-deep chunks
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  constexpr int SK_ = (MT <= 2) ? 512 : 256;  // x super-panel columns
+  // x super-panel columns: 512 when the LDS budget allows (8-wave blocks
+  // run 1 block/CU so the whole 160 KB is ours) — halves the barrier
+  // frequency of the staging loop
+  constexpr int SK_ = (MT <= 2 || NW == 8) ? 512 : 256;
   char* x_lds = smem;                         // BM * SK * 2 bytes
   char* u_lds = x_lds + BM * SK_ * 2;         // BM * r * 2 bytes
   // per-lane register copy of the codebook for the shuffle-LUT dequant
   // (a pure-VALU cndmask-tree variant measured 1.5x SLOWER: ~9 VALU/value
   // beats ds_bpermute's latency only on paper — the compiler pipelines
-  // the shuffles ~2-deep and the VALU tree just adds issue pressure)
+  // the shuffles ~2-deep and the VALU tree just adds issue pressure).
+  // The table is replicated every 16 lanes, which makes the raw-address
+  // bpermute trick below legal (garbage bits [7:6] only pick the replica).
   const float lut_reg = NF4_LUT[tid & 15];
+  const int lut_bits = __builtin_bit_cast(int, lut_reg);
 
   f32x4 acc[MT][2];
   #pragma unroll
@@ -217,31 +223,41 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
             afrag[mt] = lds_read_frag(x_lds, off);
           }
         }
-        // 2 n-tiles: dequant B fragment + MFMA
+        // 2 n-tiles: dequant both B fragments first (one 16-shuffle
+        // batch — decouples the bpermute latency chains from the MFMAs),
+        // then issue all MFMAs
+        bf16v8 bfr[2];
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt) {
           const uint32_t wbits = wb[cur][ks * 2 + nt];
           const float amv = am[cur][nt];
-          bf16v8 bfr;
           if constexpr (DBG == 1) {
             #pragma unroll
             for (int j = 0; j < 8; ++j)
-              bfr[j] = (__bf16)(amv + (float)(wbits & 1));
+              bfr[nt][j] = (__bf16)(amv + (float)(wbits & 1));
           } else {
-            // register LUT via cross-lane shuffle (ds_bpermute):
-            // lanes 0-15 hold the 16 nf4 codes; no LDS traffic, no
-            // divergent-LDS hazards
+            // register LUT via raw-address ds_bpermute: the byte address
+            // only uses bits [7:2], so shifting the nibble to bits [5:2]
+            // needs ONE shift per element (no mask — bits [7:6] carry the
+            // next nibble's low bits and merely select one of the four
+            // 16-lane table replicas). No LDS traffic, no divergent-LDS
+            // hazards.
             #pragma unroll
             for (int j = 0; j < 8; ++j) {
-              const int nib = (wbits >> (4 * j)) & 0xF;
-              bfr[j] = (__bf16)(__shfl(lut_reg, nib, WAVE) * amv);
+              const int addr = (j == 0) ? (int)(wbits << 2)
+                                        : (int)(wbits >> (4 * j - 2));
+              const float c = __builtin_bit_cast(
+                  float, __builtin_amdgcn_ds_bpermute(addr, lut_bits));
+              bfr[nt][j] = (__bf16)(c * amv);
             }
           }
+        }
+        #pragma unroll
+        for (int nt = 0; nt < 2; ++nt)
           #pragma unroll
           for (int mt = 0; mt < MT; ++mt)
             acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag[mt], bfr, acc[mt][nt], 0, 0, 0);
-        }
+                afrag[mt], bfr[nt], acc[mt][nt], 0, 0, 0);
       }
     }
   }
@@ -423,15 +439,16 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   if (const char* e = getenv("DISTRL_NF4_MTCAP")) mtcap = atoi(e);
   int mt = std::min<int>((M + 15) / 16, std::max(1, mtcap));
   const int BM = 16 * mt;
-  const int SK = (mt <= 2) ? 512 : 256;  // must mirror the kernel constexpr
-  TORCH_CHECK(K % SK == 0, "nf4_gemm: K (", K, ") % ", SK, " != 0");
   // waves per block: wide (8-wave, BN=256) blocks when N alone nearly
   // fills the chip at 1 block/CU; narrow (4-wave, BN=128, 2 blocks/CU)
   // blocks for small N so the grid still fills without deep split-K
   int nw = (N % 256 == 0 && (N / 256) * ((M + BM - 1) / BM) >= 192) ? 8 : 4;
+  if (nw == 8 && mt > 2 && K % 512 != 0) nw = 4;  // 8-wave blocks need K%512
   if (const char* e = getenv("DISTRL_NF4_NW")) nw = atoi(e);
   TORCH_CHECK(nw == 4 || nw == 8);
   TORCH_CHECK(N % (nw * 32) == 0);
+  const int SK = (mt <= 2 || nw == 8) ? 512 : 256;  // mirror kernel SK_
+  TORCH_CHECK(K % SK == 0, "nf4_gemm: K (", K, ") % ", SK, " != 0");
   const int nblk = (int)N / (nw * 32);
   // split-K over super-panels until the grid fills the chip (the skinny
   // decode shapes otherwise run at ~1 block-wave, fully latency-exposed)
