@@ -148,9 +148,9 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
     }
 
   // this slice's super-panel range (split-K over whole panels)
-  constexpr int SKC = ((MT <= 2) ? 512 : 256) / 64;  // chunks per panel
+  constexpr int SKC = SK_ / 64;                      // chunks per panel
   static_assert(SKC % 4 == 0, "ring phase must stay aligned per panel");
-  const int npanels = K / ((MT <= 2) ? 512 : 256);
+  const int npanels = K / SK_;
   const int sp0 = zid * nsp_per;
   const int sp1 = min(npanels, sp0 + nsp_per);
   const int kb0 = sp0 * SKC;
