@@ -167,21 +167,37 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   constexpr int SKU = SK / 8;                 // bf16x8 units per row
   constexpr int XIT = (BM * SKU + THREADS - 1) / THREADS;  // staging iters
 
+  // full-tile blocks (every row valid) issue the staging loads
+  // unpredicated — a per-iteration row guard makes hipcc branch around
+  // EACH load (one basic block per load, guide §5 trap c); only the
+  // ragged last m-block pays the guarded path
+  const bool full_rows = (mbase + BM <= M);
+
   for (int sp = sp0; sp < sp1; ++sp) {
     const int sk = sp * SK;
     // issue-early x loads for the whole super-panel
     bf16x8 xs[XIT];
-    #pragma unroll
-    for (int it = 0; it < XIT; ++it) {
-      const int i = tid + it * THREADS;
-      if (i < BM * SKU) {
-        const int row = i / SKU;
-        if (mbase + row < M) {
+    if (full_rows) {
+      #pragma unroll
+      for (int it = 0; it < XIT; ++it) {
+        const int i = tid + it * THREADS;
+        if (i < BM * SKU)
           xs[it] = *reinterpret_cast<const bf16x8*>(
-              x + (int64_t)(mbase + row) * K + sk + (i % SKU) * 8);
-        } else {
-          #pragma unroll
-          for (int j = 0; j < 8; ++j) xs[it].v[j] = f2bf(0.f);
+              x + (int64_t)(mbase + i / SKU) * K + sk + (i % SKU) * 8);
+      }
+    } else {
+      #pragma unroll
+      for (int it = 0; it < XIT; ++it) {
+        const int i = tid + it * THREADS;
+        if (i < BM * SKU) {
+          const int row = i / SKU;
+          if (mbase + row < M) {
+            xs[it] = *reinterpret_cast<const bf16x8*>(
+                x + (int64_t)(mbase + row) * K + sk + (i % SKU) * 8);
+          } else {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) xs[it].v[j] = f2bf(0.f);
+          }
         }
       }
     }
@@ -223,33 +239,46 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
             afrag[mt] = lds_read_frag(x_lds, off);
           }
         }
-        // 2 n-tiles: dequant both B fragments first (one 16-shuffle
-        // batch — decouples the bpermute latency chains from the MFMAs),
-        // then issue all MFMAs
+        // 2 n-tiles: dequant both B fragments first, THEN scale, THEN
+        // issue all MFMAs. The shuffle results land in 16 DISTINCT
+        // registers (cval) so the 16 ds_bpermutes issue as one batch and
+        // pipeline in the DS unit — when the destination doubles as the
+        // address register (what minimal-register codegen produces), the
+        // chain waits ~50-60 cycles per PAIR instead.
         bf16v8 bfr[2];
-        #pragma unroll
-        for (int nt = 0; nt < 2; ++nt) {
-          const uint32_t wbits = wb[cur][ks * 2 + nt];
-          const float amv = am[cur][nt];
-          if constexpr (DBG == 1) {
+        if constexpr (DBG == 1) {
+          #pragma unroll
+          for (int nt = 0; nt < 2; ++nt) {
+            const uint32_t wbits = wb[cur][ks * 2 + nt];
             #pragma unroll
             for (int j = 0; j < 8; ++j)
-              bfr[nt][j] = (__bf16)(amv + (float)(wbits & 1));
-          } else {
-            // register LUT via raw-address ds_bpermute: the byte address
-            // only uses bits [7:2], so shifting the nibble to bits [5:2]
-            // needs ONE shift per element (no mask — bits [7:6] carry the
-            // next nibble's low bits and merely select one of the four
-            // 16-lane table replicas). No LDS traffic, no divergent-LDS
-            // hazards.
+              bfr[nt][j] = (__bf16)(am[cur][nt] + (float)(wbits & 1));
+          }
+        } else {
+          // register LUT via raw-address ds_bpermute: the byte address
+          // only uses bits [7:2], so shifting the nibble to bits [5:2]
+          // needs ONE shift per element (no mask — bits [7:6] carry the
+          // next nibble's low bits and merely select one of the four
+          // 16-lane table replicas). No LDS traffic, no divergent-LDS
+          // hazards.
+          float cval[2][8];
+          #pragma unroll
+          for (int nt = 0; nt < 2; ++nt) {
+            const uint32_t wbits = wb[cur][ks * 2 + nt];
             #pragma unroll
             for (int j = 0; j < 8; ++j) {
               const int addr = (j == 0) ? (int)(wbits << 2)
                                         : (int)(wbits >> (4 * j - 2));
-              const float c = __builtin_bit_cast(
+              cval[nt][j] = __builtin_bit_cast(
                   float, __builtin_amdgcn_ds_bpermute(addr, lut_bits));
-              bfr[nt][j] = (__bf16)(c * amv);
             }
+          }
+          #pragma unroll
+          for (int nt = 0; nt < 2; ++nt) {
+            const float amv = am[cur][nt];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+              bfr[nt][j] = (__bf16)(cval[nt][j] * amv);
           }
         }
         #pragma unroll
