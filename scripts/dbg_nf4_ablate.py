@@ -43,6 +43,23 @@ def main():
     x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
     os.environ["DISTRL_NF4_KSPLIT"] = ks
 
+    if os.environ.get("SWEEP2") == "1":
+        # geometry sweep: waves-per-block x m-tile x ksplit
+        for nw in (4, 8):
+            for mtc in (3, 4, 5):
+                for ksv in (1, 2, 4, 8):
+                    os.environ["DISTRL_NF4_NW"] = str(nw)
+                    os.environ["DISTRL_NF4_MTCAP"] = str(mtc)
+                    os.environ["DISTRL_NF4_KSPLIT"] = str(ksv)
+                    t = bench(lambda: ext.nf4_gemm(x, w4f, amaxf, None, None,
+                                                   None, N, K, 0), iters=60)
+                    eff = (N * K / 2) / (t * 1e-6) / 1e12
+                    print(f"nw={nw} mt={mtc} ks={ksv}: {t:7.1f}us "
+                          f"({eff:.2f} TB/s)", flush=True)
+        for v in ("DISTRL_NF4_NW", "DISTRL_NF4_MTCAP", "DISTRL_NF4_KSPLIT"):
+            os.environ.pop(v, None)
+        return
+
     if os.environ.get("PROF") == "1":
         for _ in range(100):
             ext.nf4_gemm(x, w4f, amaxf, None, None, None, N, K, 0)
