@@ -468,11 +468,11 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   if (const char* e = getenv("DISTRL_NF4_MTCAP")) mtcap = atoi(e);
   int mt = std::min<int>((M + 15) / 16, std::max(1, mtcap));
   const int BM = 16 * mt;
-  // waves per block: wide (8-wave, BN=256) blocks when N alone nearly
-  // fills the chip at 1 block/CU; narrow (4-wave, BN=128, 2 blocks/CU)
-  // blocks for small N so the grid still fills without deep split-K
-  int nw = (N % 256 == 0 && (N / 256) * ((M + BM - 1) / BM) >= 192) ? 8 : 4;
-  if (nw == 8 && mt > 2 && K % 512 != 0) nw = 4;  // 8-wave blocks need K%512
+  // waves per block: 4-wave (BN=128, 2 blocks/CU) measured >= the 8-wave
+  // variant on every decode shape (co-resident independent blocks
+  // decorrelate the dequant-latency stalls); the 8-wave template is kept
+  // behind DISTRL_NF4_NW=8 for experiments
+  int nw = 4;
   if (const char* e = getenv("DISTRL_NF4_NW")) nw = atoi(e);
   TORCH_CHECK(nw == 4 || nw == 8);
   TORCH_CHECK(N % (nw * 32) == 0);
@@ -483,7 +483,7 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   // decode shapes otherwise run at ~1 block-wave, fully latency-exposed)
   const int npanels = (int)K / SK;
   const int base_blocks = ((M + BM - 1) / BM) * nblk;
-  int blk_target = (nw == 8) ? 384 : 768;
+  int blk_target = (nw == 8) ? 384 : 640;  // sweep-tuned (profiles/)
   if (const char* e = getenv("DISTRL_NF4_BLKTGT")) blk_target = atoi(e);
   int ksplit = 1;
   while (ksplit * 2 <= npanels && base_blocks * ksplit < blk_target

@@ -142,3 +142,30 @@ def test_graph_vs_eager_decode_identical(setup):
     out_e = Engine(model, cfg_e, device=torch.device("cuda:0"),
                    seed=3).generate(prompts, sp, eos_token_id=None)
     assert out_g == out_e, (out_g, out_e)
+
+
+def test_session_cache_vs_fresh_identical(setup):
+    """CachedDecodeSession (DISTRL_GRAPH_CACHE=1: padded lanes, reused
+    buffers, hipGraph reused across waves) must produce exactly the same
+    greedy tokens as per-wave fresh sessions (docs/ROADMAP.md #4)."""
+    import os
+    model, _ = setup
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    cfg = dict(max_seq_length=256, kv_block_size=16, num_kv_blocks=512,
+               max_num_seqs=64)
+    prompts_a = [list(range(3, 40)), [7, 11, 13, 17, 19], [2, 4]]
+    prompts_b = [[5, 6, 7, 300, 9], list(range(100, 120))]
+    sp = SamplingParams(max_tokens=12, temperature=0.0, n=2)
+    outs = {}
+    for cache in ("0", "1"):
+        os.environ["DISTRL_GRAPH_CACHE"] = cache
+        try:
+            eng = Engine(model, EngineConfig(**cfg),
+                         device=torch.device("cuda:0"), seed=3)
+            # two waves: the second exercises buffer/graph REUSE
+            outs[cache] = (eng.generate(prompts_a, sp, eos_token_id=None),
+                           eng.generate(prompts_b, sp, eos_token_id=None))
+        finally:
+            os.environ.pop("DISTRL_GRAPH_CACHE", None)
+    assert outs["0"] == outs["1"], (outs["0"], outs["1"])
