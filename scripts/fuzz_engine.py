@@ -25,6 +25,7 @@ ap = argparse.ArgumentParser()
 ap.add_argument("--trials", type=int, default=200)
 ap.add_argument("--model", type=str, default="tiny-qwen2")
 ap.add_argument("--seed", type=int, default=20260913)
+ap.add_argument("--start", type=int, default=0, help="skip trials < start (RNG still advances)")
 args = ap.parse_args()
 
 dev = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
@@ -69,6 +70,8 @@ for trial in range(args.trials):
     streamed = {}
     cb = (lambda pi, ci, toks: streamed.setdefault((pi, ci), []).extend(toks)) \
         if rng.random() < 0.5 else None
+    if trial < args.start:
+        continue
     try:
         res = eng.generate(prompts, sp, eos_token_id=eos, stream_cb=cb)
     except MemoryError:
