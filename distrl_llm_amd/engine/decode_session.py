@@ -70,7 +70,11 @@ class DecodeSession:
         self.seeds = torch.randint(0, 2**31 - 1, (N,), dtype=torch.int64,
                                    device=dev, generator=engine.generator)
         self.max_steps = max(lim - pl for lim, pl in zip(limits, prompt_lens))
-        self.out_buf = torch.zeros(max(self.max_steps, 1), N,
+        # >= 2 rows: _capture() warms up with TWO _step() calls before the
+        # state restore, so step_idx reaches 1 — a 1-row buffer would make
+        # the warmup's index_copy_ write out of bounds (an async device
+        # fault that surfaces as an HSA exception a trial later)
+        self.out_buf = torch.zeros(max(self.max_steps, 2), N,
                                    dtype=torch.long, device=dev)
         self.graph = None
         self.use_graph = use_graph and dev.type == "cuda"
@@ -227,7 +231,8 @@ class CachedDecodeSession(DecodeSession):
         self.finished = torch.zeros(n_pad, dtype=torch.bool, device=dev)
         self.step_idx = torch.zeros(1, dtype=torch.long, device=dev)
         self.seeds = torch.zeros(n_pad, dtype=torch.int64, device=dev)
-        self.out_buf = torch.zeros(max(sp.max_tokens, 1), n_pad,
+        # >= 2 rows for the same capture-warmup reason as the base class
+        self.out_buf = torch.zeros(max(sp.max_tokens, 2), n_pad,
                                    dtype=torch.long, device=dev)
         self.graph = None
         self.use_graph = dev.type == "cuda"
