@@ -56,16 +56,26 @@ class Attention(nn.Module):
         k = self.k_proj(x).view(B, T, s.num_kv_heads, s.head_dim)
         v = self.v_proj(x).view(B, T, s.num_kv_heads, s.head_dim)
         q, k = R.apply_rope(q, k, cos, sin)
-        group = s.num_heads // s.num_kv_heads
-        k = k.repeat_interleave(group, dim=2)
-        v = v.repeat_interleave(group, dim=2)
-        q, k, v = (t.transpose(1, 2) for t in (q, k, v))  # (B, H, T, D)
-        if attn_bias is None:
-            o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
-                                               scale=self.scale)
+        if (attn_bias is None and q.is_cuda and q.dtype == torch.bfloat16
+                and s.head_dim in (64, 128)):
+            # first-party CDNA4 flash attention (ops/csrc/attention.hip):
+            # native GQA (no repeat_interleave materialization), causal,
+            # fused online softmax — replaces torch SDPA's aotriton
+            # (Triton-derived) backend on the learner hot path
+            from ..ops import functional as OF
+            q, k, v = (t.transpose(1, 2).contiguous() for t in (q, k, v))
+            o = OF.flash_attention(q, k, v, self.scale)
         else:
-            o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias,
-                                               scale=self.scale)
+            group = s.num_heads // s.num_kv_heads
+            k = k.repeat_interleave(group, dim=2)
+            v = v.repeat_interleave(group, dim=2)
+            q, k, v = (t.transpose(1, 2) for t in (q, k, v))  # (B, H, T, D)
+            if attn_bias is None:
+                o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                                   scale=self.scale)
+            else:
+                o = F.scaled_dot_product_attention(
+                    q, k, v, attn_mask=attn_bias, scale=self.scale)
         o = o.transpose(1, 2).reshape(B, T, s.q_size)
         return self.o_proj(o)
 

@@ -47,6 +47,12 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
 void lora_u(torch::Tensor x, torch::Tensor afrag, torch::Tensor u,
             int64_t r, int64_t ksplit);
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
+std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
+                                          torch::Tensor v, double scale);
+std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                          torch::Tensor k, torch::Tensor v,
+                                          torch::Tensor o, torch::Tensor lse,
+                                          double scale);
 void adam8bit_step(torch::Tensor p, torch::Tensor g, torch::Tensor m_q,
                    torch::Tensor v_q, torch::Tensor m_absmax,
                    torch::Tensor v_absmax, double lr, double b1, double b2,
@@ -81,4 +87,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused nf4-dequant MFMA GEMM with bias + LoRA-B epilogue");
   m.def("lora_u", &lora_u, "split-K LoRA A projection (u = x @ A^T)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA mapping probe");
+  m.def("flash_attn_fwd", &flash_attn_fwd,
+        "causal GQA flash attention forward (returns o, lse)");
+  m.def("flash_attn_bwd", &flash_attn_bwd,
+        "causal GQA flash attention backward (returns dq, dk, dv)");
 }

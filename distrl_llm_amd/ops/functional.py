@@ -177,6 +177,38 @@ def sample_tokens(logits: torch.Tensor, temperature: float, top_p: float,
     return R.sample_tokens(logits, temperature, top_p, top_k, generator=generator)
 
 
+# --------------------------------------------------- flash attention
+
+class _FlashAttnFn(torch.autograd.Function):
+    """First-party CDNA4 causal GQA flash attention (ops/csrc/attention.hip)
+    — the learner's training attention without aotriton/Triton (SURVEY.md
+    §2.4-B row "causal attention fwd/bwd"; reference hot path
+    distributed_actor.py:241-243 + loss.backward())."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        ext = _require_ext("flash_attention")
+        o, lse = ext.flash_attn_fwd(q, k, v, float(scale))
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = float(scale)
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, o, lse = ctx.saved_tensors
+        ext = _require_ext("flash_attention")
+        dq, dk, dv = ext.flash_attn_bwd(dout.contiguous(), q, k, v, o, lse,
+                                        ctx.scale)
+        return dq, dk, dv, None
+
+
+def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                    scale: float) -> torch.Tensor:
+    """Causal GQA attention, (B, H, T, D) layout, bf16. k/v carry the KV
+    head count natively (no repeat_interleave materialization)."""
+    return _FlashAttnFn.apply(q, k, v, scale)
+
+
 # ------------------------------------------------------------ fused loss
 
 class _LogprobLossFn(torch.autograd.Function):

@@ -374,3 +374,48 @@ def test_prefill_attention(ext, D, H, KV, lens):
     ref = R.varlen_prefill_attention(q.float(), k.float(), v.float(), lens,
                                      scale)
     torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("B,Hq,Hkv,T,D", [
+    (2, 8, 2, 100, 64),
+    (1, 4, 4, 333, 128),
+    (2, 28, 4, 160, 128),   # Qwen2.5-7B head geometry, short T
+])
+def test_flash_attention_fwd_bwd(ext, B, Hq, Hkv, T, D):
+    """First-party flash attention vs fp32 torch reference: forward and all
+    three input grads (SURVEY.md §2.4-B training attention row)."""
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    scale = D ** -0.5
+    q = (torch.randn(B, Hq, T, D, device=dev) * 0.5).to(torch.bfloat16)
+    k = (torch.randn(B, Hkv, T, D, device=dev) * 0.5).to(torch.bfloat16)
+    v = (torch.randn(B, Hkv, T, D, device=dev) * 0.5).to(torch.bfloat16)
+    dout = (torch.randn(B, Hq, T, D, device=dev) * 0.5).to(torch.bfloat16)
+
+    from distrl_llm_amd.ops import functional as OF
+    qg, kg, vg = (t.clone().requires_grad_(True) for t in (q, k, v))
+    o = OF.flash_attention(qg, kg, vg, scale)
+    o.backward(dout)
+
+    # fp32 reference with GQA expansion
+    group = Hq // Hkv
+    qr = q.float().requires_grad_(True)
+    kr = k.float().requires_grad_(True)
+    vr = v.float().requires_grad_(True)
+    ke = kr.repeat_interleave(group, 1)
+    ve = vr.repeat_interleave(group, 1)
+    s = torch.einsum("bhqd,bhkd->bhqk", qr, ke) * scale
+    mask = torch.ones(T, T, dtype=torch.bool, device=dev).tril()
+    s = s.masked_fill(~mask, float("-inf"))
+    p = s.softmax(-1)
+    orf = torch.einsum("bhqk,bhkd->bhqd", p, ve)
+    orf.backward(dout.float())
+
+    assert torch.allclose(o.float(), orf, atol=3e-2, rtol=3e-2), \
+        (o.float() - orf).abs().max()
+    assert torch.allclose(qg.grad.float(), qr.grad, atol=8e-2, rtol=8e-2), \
+        (qg.grad.float() - qr.grad).abs().max()
+    assert torch.allclose(kg.grad.float(), kr.grad, atol=8e-2, rtol=8e-2), \
+        (kg.grad.float() - kr.grad).abs().max()
+    assert torch.allclose(vg.grad.float(), vr.grad, atol=8e-2, rtol=8e-2), \
+        (vg.grad.float() - vr.grad).abs().max()
