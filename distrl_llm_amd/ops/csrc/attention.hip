@@ -85,10 +85,17 @@ DEV_INLINE bf16v8 relayout_frag(const f32x4& t0, const f32x4& t1) {
   return *reinterpret_cast<const bf16v8*>(w);
 }
 
-// LDS tile staging: [ROWS][COLS] bf16 row-major with the XOR swizzle
-// byte ^= (row & 15) << 4 (COLS*2 must be a multiple of 256 so the
-// swizzle stays inside the row). Readers use lds_frag_off with the same
-// XOR. Guards rows >= limit with zero fill.
+// LDS tile staging: [ROWS][COLS] bf16 row-major with an XOR swizzle on
+// the 16-B column slot, MASKED to the row width (COLS/8 slots per row)
+// so it never escapes the row — (row & 15) << 4 for 256-B rows,
+// (row & 3) << 4 for 64-B rows. Readers use lds_frag with the same XOR.
+// Guards rows >= limit with zero fill.
+template <int COLS>
+DEV_INLINE int tile_off(int row, int cu) {
+  constexpr int SWZ_MASK = (COLS / 8) - 1;
+  return (row * (COLS * 2) + cu * 16) ^ ((row & SWZ_MASK) << 4);
+}
+
 template <int ROWS, int COLS>
 DEV_INLINE void stage_tile(char* lds, const __hip_bfloat16* src,
                            int64_t src_row_stride, int rows_valid, int tid,
@@ -105,15 +112,13 @@ DEV_INLINE void stage_tile(char* lds, const __hip_bfloat16* src,
       #pragma unroll
       for (int j = 0; j < 8; ++j) v.v[j] = f2bf(0.f);
     }
-    const int off = (row * (COLS * 2) + cu * 16) ^ ((row & 15) << 4);
-    *reinterpret_cast<bf16x8*>(lds + off) = v;
+    *reinterpret_cast<bf16x8*>(lds + tile_off<COLS>(row, cu)) = v;
   }
 }
 
 template <int COLS>
 DEV_INLINE bf16v8 lds_frag(const char* lds, int row, int col8) {
-  const int off = (row * (COLS * 2) + col8 * 16) ^ ((row & 15) << 4);
-  return *reinterpret_cast<const bf16v8*>(lds + off);
+  return *reinterpret_cast<const bf16v8*>(lds + tile_off<COLS>(row, col8));
 }
 
 // ---------------------------------------------------------------- forward
