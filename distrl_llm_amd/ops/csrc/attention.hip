@@ -233,11 +233,17 @@ void flash_fwd_kernel(const __hip_bfloat16* __restrict__ q,
       psum = wave_xor_sum(psum);
       l_run = l_run * alpha + psum;
       m_run = m_new;
-      // rescale O
+      // rescale O. The softmax stats live per LANE COLUMN (q = lc), but
+      // the accumulator's rows are q = 4*lg + r — fetch each row's alpha
+      // from the lane that owns that q column.
+      float alpha_row[4];
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        alpha_row[r] = __shfl(alpha, 4 * lg + r, 64);
       #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
         #pragma unroll
-        for (int r = 0; r < 4; ++r) oacc[dt][r] *= alpha;
+        for (int r = 0; r < 4; ++r) oacc[dt][r] *= alpha_row[r];
       }
       // PV: A = PT^T fragments (relayout), B = v_t rows
       const bf16v8 pa = relayout_frag(st[0], st[1]);
@@ -251,15 +257,19 @@ void flash_fwd_kernel(const __hip_bfloat16* __restrict__ q,
   }
 
   // epilogue: normalize, stage O tile through LDS (D-layout columns ->
-  // row-major global), write lse
+  // row-major global), write lse. As with alpha, 1/l for accumulator row
+  // q = 4*lg+r lives in lane lc = 4*lg+r.
   const float inv = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  float inv_row[4];
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) inv_row[r] = __shfl(inv, 4 * lg + r, 64);
   float* ow = o_lds + (size_t)wave * QB * D;
   #pragma unroll
   for (int dt = 0; dt < D / 16; ++dt) {
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qq = 4 * lg + r;            // row within the wave's tile
-      ow[qq * D + dt * 16 + lc] = oacc[dt][r] * inv;
+      ow[qq * D + dt * 16 + lc] = oacc[dt][r] * inv_row[r];
     }
   }
   if (q0 < T && lg == 0) {

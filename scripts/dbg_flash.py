@@ -57,8 +57,9 @@ if err2.max() > 1e-2:
 torch.manual_seed(1)
 q3 = (torch.randn(B, H, T, D, device=dev) * 0.3).to(torch.bfloat16)
 k3 = (torch.randn(B, H, T, D, device=dev) * 0.3).to(torch.bfloat16)
-v3 = torch.eye(T, device=dev)[:, :D].view(1, 1, T, D).to(torch.bfloat16) \
-    .expand(B, H, T, D).contiguous()
+v3 = torch.zeros(T, D, device=dev)
+v3[:, :T] = torch.eye(T, device=dev)
+v3 = v3.view(1, 1, T, D).to(torch.bfloat16).expand(B, H, T, D).contiguous()
 o3, lse3 = run(q3, k3, v3, scale=D ** -0.5)
 r3 = ref(q3, k3, v3, scale=D ** -0.5)
 err3 = (o3.float() - r3).abs()
