@@ -456,8 +456,8 @@ void flash_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ q,
                           const __hip_bfloat16* __restrict__ dot_t,
                           const float* __restrict__ lse,
                           const float* __restrict__ delta,
-                          float* __restrict__ dkt,  // (B,Hkv,D,T) f32 accum
-                          float* __restrict__ dvt,
+                          __hip_bfloat16* __restrict__ dkt,  // (B,Hq,D,T)
+                          __hip_bfloat16* __restrict__ dvt,   // per Q-HEAD
                           int B, int Hq, int Hkv, int T, int Tp, float scale) {
   constexpr int KB = 16;     // kv cols per wave
   const int kvblk = blockIdx.x;          // 64 kv rows per block
@@ -574,8 +574,9 @@ void flash_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ q,
     }
   }
 
-  // accumulate into f32 (B,Hkv,D,T): GQA heads of one group all add into
-  // the same kv head -> atomic (group > 1) or direct store (group == 1)
+  // per-Q-HEAD bf16 outputs (B,Hq,D,T): each (kvblk, h) block owns its
+  // region exclusively — no atomics; the host sums the GQA group in fp32
+  // (288 GB makes the extra head-resolved image free)
   if (kv0 < T) {
     #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
@@ -584,18 +585,14 @@ void flash_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ q,
         const int d = dt * 16 + 4 * lg + r;   // D-layout row = d
         const int kv = kv0 + lc;              // col = kv
         if (kv < T) {
-          const int64_t off = (bkv * D + d) * T + kv;
-          if (group == 1) {
-            dkt[off] = dkacc[dt][r];
-            dvt[off] = dvacc[dt][r];
-          } else {
-            atomicAdd(&dkt[off], dkacc[dt][r]);
-            atomicAdd(&dvt[off], dvacc[dt][r]);
-          }
+          const int64_t off = (bh * D + d) * T + kv;
+          dkt[off] = f2bf(dkacc[dt][r]);
+          dvt[off] = f2bf(dvacc[dt][r]);
         }
       }
     }
   }
+  (void)group;
 }
 
 }  // namespace
@@ -651,8 +648,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                 .contiguous();
   auto delta = (doc.to(at::kFloat) * o.to(at::kFloat)).sum(-1);  // (B,Hq,T)
   auto dq = torch::empty_like(qc);
-  auto dkt = torch::zeros({B, Hkv, D, T}, q.options().dtype(at::kFloat));
-  auto dvt = torch::zeros({B, Hkv, D, T}, q.options().dtype(at::kFloat));
+  auto dkt = torch::zeros({B, Hq, D, T}, q.options());
+  auto dvt = torch::zeros({B, Hq, D, T}, q.options());
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 block(256);
   dim3 grid_q((T + QB * NWAVE - 1) / (QB * NWAVE), Hq, B);
@@ -682,12 +679,16 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
         reinterpret_cast<const __hip_bfloat16*>(doc.data_ptr()), \
         reinterpret_cast<const __hip_bfloat16*>(dot_t.data_ptr()), \
         lse.data_ptr<float>(), delta.data_ptr<float>(), \
-        dkt.data_ptr<float>(), dvt.data_ptr<float>(), \
+        reinterpret_cast<__hip_bfloat16*>(dkt.data_ptr()), \
+        reinterpret_cast<__hip_bfloat16*>(dvt.data_ptr()), \
         B, Hq, Hkv, T, Tp, (float)scale)
   if (D == 128) LAUNCH_DKV(128); else LAUNCH_DKV(64);
   #undef LAUNCH_DKV
   HIP_CHECK_LAST();
-  auto dk = dkt.transpose(2, 3).contiguous().to(at::kBFloat16);
-  auto dv = dvt.transpose(2, 3).contiguous().to(at::kBFloat16);
+  const int group = Hq / Hkv;
+  auto dk = dkt.view({B, Hkv, group, D, T}).to(at::kFloat).sum(2)
+                .transpose(2, 3).contiguous().to(at::kBFloat16);
+  auto dv = dvt.view({B, Hkv, group, D, T}).to(at::kFloat).sum(2)
+                .transpose(2, 3).contiguous().to(at::kBFloat16);
   return {dq, dk, dv};
 }
