@@ -335,10 +335,15 @@ class SessionCache:
 
     @staticmethod
     def _pad(n: int) -> int:
-        p = 8
-        while p < n:
-            p *= 2
-        return p
+        """Pad the wave to a multiple of 16 (min 8). Power-of-two padding
+        measured a 12% END-TO-END regression on the headline bench: batch
+        160 padded to 256 runs every decode GEMM/attention 60% fatter
+        (gen 13.5 -> 17.0 s/round). Multiple-of-16 keeps the graph-count
+        bounded (<= max_num_seqs/16 cache entries) at near-zero padding
+        waste."""
+        if n <= 8:
+            return 8
+        return (n + 15) // 16 * 16
 
     def acquire(self, seqs: List[Sequence], sp: SamplingParams,
                 eos_token_id: Optional[int]) -> CachedDecodeSession:
