@@ -164,7 +164,12 @@ class Engine:
                 torch.tensor(tkv0, dtype=torch.int32, device=dev),
                 max(lens), self.scale)
         if q.is_cuda:
-            group = self.spec.num_heads // self.spec.num_kv_heads
+            # prompts beyond the varlen kernel's score-tile cap: batch-pad
+            # each prompt into its own row and run the first-party flash
+            # kernel (ops/csrc/attention.hip) — native GQA, causal, no
+            # length cap, no aotriton/SDPA (docs/ROADMAP.md #9)
+            from ..ops.build import get_extension
+            ext = get_extension()
             B, Lmax = len(lens), max(lens)
             H, KV, D = self.spec.num_heads, self.spec.num_kv_heads, self.spec.head_dim
             qp = q.new_zeros(B, Lmax, H, D)
@@ -176,11 +181,10 @@ class Engine:
                 kp[i, :L] = k[start:start + L]
                 vp[i, :L] = v[start:start + L]
                 start += L
-            kp = kp.repeat_interleave(group, 2)
-            vp = vp.repeat_interleave(group, 2)
-            o = F.scaled_dot_product_attention(
-                qp.transpose(1, 2), kp.transpose(1, 2), vp.transpose(1, 2),
-                is_causal=True, scale=self.scale).transpose(1, 2)
+            o, _ = ext.flash_attn_fwd(
+                qp.transpose(1, 2).contiguous(), kp.transpose(1, 2).contiguous(),
+                vp.transpose(1, 2).contiguous(), self.scale)
+            o = o.transpose(1, 2)
             # trailing pad rows never influence real rows under causal
             return torch.cat([o[i, :L] for i, L in enumerate(lens)], 0)
         return R.varlen_prefill_attention(q, k, v, lens, self.scale)

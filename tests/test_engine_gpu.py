@@ -169,3 +169,21 @@ def test_session_cache_vs_fresh_identical(setup):
         finally:
             os.environ.pop("DISTRL_GRAPH_CACHE", None)
     assert outs["0"] == outs["1"], (outs["0"], outs["1"])
+
+
+def test_long_prompt_prefill_no_cap(setup):
+    """Prompts beyond the varlen prefill kernel's ~2200-token score-tile
+    cap route through the first-party flash kernel (no SDPA/aotriton, no
+    length cap — docs/ROADMAP.md #9)."""
+    model, _ = setup
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    cfg = EngineConfig(max_seq_length=3072, kv_block_size=16,
+                       num_kv_blocks=1024, max_num_seqs=8)
+    eng = Engine(model, cfg, device=torch.device("cuda:0"), seed=0)
+    prompt = [(i * 7 + 3) % 900 + 1 for i in range(2500)]
+    sp = SamplingParams(max_tokens=4, temperature=0.0, n=1)
+    res = eng.generate([prompt], sp, eos_token_id=None)
+    expected = _naive_greedy(model, prompt, 4)
+    agree = sum(a == b for a, b in zip(res[0][0], expected))
+    assert agree >= 3, (res[0][0], expected)
