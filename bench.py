@@ -44,6 +44,11 @@ def parse_args():
     p.add_argument("--actors", type=int, default=-1,
                    help="-1: all ranks are dual-role learners (0 actors)")
     p.add_argument("--cpu", action="store_true", help="CPU/gloo debug mode")
+    p.add_argument("--eos-mean", type=float, default=0.0,
+                   help="EOS-realistic mode: per-candidate exponential "
+                        "output-length caps with this mean (e.g. 450 — the "
+                        "reference eval's observed mean; 0 = fixed-length "
+                        "no-EOS rounds). Measures in-wave retirement.")
     p.add_argument("--tiny", action="store_true",
                    help="tiny model + short generations (debug only; NOT the "
                         "headline config)")
@@ -157,7 +162,9 @@ def main():
 
     # ---- rank 0 drives the bench ----
     batches = list(train.iter(batch_size=batch_size))
-    sp_dict = trainer.sampling_params.__dict__
+    sp_dict = dict(trainer.sampling_params.__dict__)
+    if args.eos_mean > 0:
+        sp_dict["geom_len_mean"] = args.eos_mean
 
     for i in range(args.warmup):
         trainer.rl_round(batches[i], sp_dict)

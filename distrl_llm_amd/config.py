@@ -28,6 +28,10 @@ class SamplingParams:
     top_p: float = 1.0
     top_k: int = 0  # 0 = disabled
     seed: Optional[int] = None
+    # optional realistic-length mode: each candidate draws an exponential
+    # output cap with this mean (clamped to [1, max_tokens]) — used by the
+    # EOS-realistic bench so in-wave retirement is measured, not assumed
+    geom_len_mean: Optional[float] = None
 
     def __post_init__(self) -> None:
         if self.max_tokens <= 0:

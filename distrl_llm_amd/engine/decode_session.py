@@ -44,8 +44,12 @@ class DecodeSession:
         N = len(seqs)
         max_total = engine.cfg.max_seq_length
 
-        prompt_lens = [len(q.prompt_ids) for q in seqs]
-        limits = [min(pl + sp.max_tokens, max_total) - 1 for pl in prompt_lens]
+        # sequences may be RESUMED mid-generation (in-wave retirement re-
+        # waves survivors): the session starts each lane at its CURRENT
+        # last-token position, not at the prompt boundary
+        pos0 = [q.total_len - 1 for q in seqs]
+        limits = [min(len(q.prompt_ids) + (q.max_tokens or sp.max_tokens),
+                      max_total) - 1 for q in seqs]
 
         # pre-allocate every block each sequence can ever need
         for q, lim in zip(seqs, limits):
@@ -61,15 +65,15 @@ class DecodeSession:
         self.block_tables = bt.to(dev, non_blocking=True)
         self.tokens = torch.tensor([q.output_ids[-1] for q in seqs],
                                    dtype=torch.long, device=dev)
-        self.positions = torch.tensor(prompt_lens, dtype=torch.long, device=dev)
-        self.prompt_lens = list(prompt_lens)
+        self.positions = torch.tensor(pos0, dtype=torch.long, device=dev)
+        self.pos0 = pos0
         self.ctx_lens = (self.positions + 1).to(torch.int32)
         self.limit_pos = torch.tensor(limits, dtype=torch.long, device=dev)
         self.finished = torch.zeros(N, dtype=torch.bool, device=dev)
         self.step_idx = torch.zeros(1, dtype=torch.long, device=dev)
         self.seeds = torch.randint(0, 2**31 - 1, (N,), dtype=torch.int64,
                                    device=dev, generator=engine.generator)
-        self.max_steps = max(lim - pl for lim, pl in zip(limits, prompt_lens))
+        self.max_steps = max(lim - p0 for lim, p0 in zip(limits, pos0))
         # >= 2 rows: _capture() warms up with TWO _step() calls before the
         # state restore, so step_idx reaches 1 — a 1-row buffer would make
         # the warmup's index_copy_ write out of bounds (an async device
@@ -146,10 +150,18 @@ class DecodeSession:
 
     # -------------------------------------------------------------- run
 
-    def run(self, chunk: int = 64, stream_cb=None) -> List[List[int]]:
+    def run(self, chunk: int = 16, stream_cb=None, retire_at=None):
+        """Decode until every lane finishes, or — when ``retire_at`` is
+        set — until at least that many lanes have finished (in-wave
+        retirement: the engine then retires them, admits waiting prompts
+        and re-waves the survivors; reference parity: vLLM continuous
+        batching inside fast_generate, distributed_actor.py:147-172).
+
+        Returns (new_tokens_per_seq, finished_per_seq): only the tokens
+        generated by THIS session (the caller extends q.output_ids)."""
         reported = [0] * len(self.seqs)  # out_buf rows already streamed
         if self.max_steps > 0:
-            if self.use_graph:
+            if self.use_graph and self.graph is None:
                 try:
                     self._capture()
                 except Exception as err:  # capture unsupported -> eager
@@ -172,22 +184,26 @@ class DecodeSession:
                     pos = self.positions.cpu().tolist()
                     buf = self.out_buf[:steps].cpu()
                     for i, q in enumerate(self.seqs):
-                        n_new = pos[i] - self.prompt_lens[i]
+                        n_new = pos[i] - self.pos0[i]
                         if n_new > reported[i]:
                             stream_cb(q.parent_prompt, q.cand_index,
                                       buf[reported[i]:n_new, i].tolist())
                             reported[i] = n_new
-                if bool(self.finished.all()):
+                n_fin = int(self.finished.sum())
+                if n_fin >= self.finished.numel():  # incl. padded dummies
+                    break
+                if retire_at is not None and n_fin >= retire_at:
                     break
 
-        # ---- extraction ----
+        # ---- extraction: this session's new tokens + finished flags ----
         final_pos = self.positions.cpu().tolist()
+        fin = self.finished.cpu().tolist()
         out = self.out_buf.cpu()
-        results = []
+        new_tokens = []
         for i, q in enumerate(self.seqs):
-            n_new = final_pos[i] - self.prompt_lens[i]
-            results.append(q.output_ids[:1] + out[:n_new, i].tolist())
-        return results
+            n_new = final_pos[i] - self.pos0[i]
+            new_tokens.append(out[:n_new, i].tolist())
+        return new_tokens, fin[:len(self.seqs)]
 
 
 # --------------------------------------------------------- session cache
@@ -237,7 +253,7 @@ class CachedDecodeSession(DecodeSession):
         self.graph = None
         self.use_graph = dev.type == "cuda"
         self.seqs: List[Sequence] = []
-        self.prompt_lens: List[int] = []
+        self.pos0: List[int] = []
         self.max_steps = 0
         self._scratch_block: Optional[int] = None
 
@@ -250,16 +266,17 @@ class CachedDecodeSession(DecodeSession):
         N = len(seqs)
         assert N <= self.n_pad
         max_total = engine.cfg.max_seq_length
-        prompt_lens = [len(q.prompt_ids) for q in seqs]
-        limits = [min(pl + sp.max_tokens, max_total) - 1 for pl in prompt_lens]
+        pos0 = [q.total_len - 1 for q in seqs]
+        limits = [min(len(q.prompt_ids) + (q.max_tokens or sp.max_tokens),
+                      max_total) - 1 for q in seqs]
         for q, lim in zip(seqs, limits):
             need = KVCachePool.blocks_for(lim + 1, bs)
             while len(q.block_table) < need:
                 q.block_table.append(engine.pool.allocator.alloc())
 
         self.seqs = seqs
-        self.prompt_lens = list(prompt_lens)
-        self.max_steps = max(lim - pl for lim, pl in zip(limits, prompt_lens))
+        self.pos0 = pos0
+        self.max_steps = max(lim - p0 for lim, p0 in zip(limits, pos0))
         self._scratch_block = scratch_block
 
         n_pad, dev = self.n_pad, engine.device
@@ -279,8 +296,8 @@ class CachedDecodeSession(DecodeSession):
             t.copy_(host.to(dev, non_blocking=True))
 
         fill(self.tokens, [q.output_ids[-1] for q in seqs], 0)
-        fill(self.positions, prompt_lens, 0)
-        fill(self.ctx_lens, [p + 1 for p in prompt_lens], 1)
+        fill(self.positions, pos0, 0)
+        fill(self.ctx_lens, [p + 1 for p in pos0], 1)
         fill(self.limit_pos, limits, 0)  # dummy limit 0 -> finished
         host_fin = torch.zeros(n_pad, dtype=torch.bool)
         host_fin[N:] = True
@@ -294,33 +311,18 @@ class CachedDecodeSession(DecodeSession):
         self.out_buf.zero_()
         return self
 
-    def run(self, chunk: int = 64, stream_cb=None):
-        # a graph captured on a previous wave stays valid (same storages)
-        if self.use_graph and self.graph is not None and self.max_steps > 0:
-            steps = 0
-            reported = [0] * len(self.seqs)
-            while steps < self.max_steps:
-                n = min(chunk, self.max_steps - steps)
-                for _ in range(n):
-                    self.graph.replay()
-                steps += n
-                if stream_cb is not None:
-                    pos = self.positions.cpu().tolist()
-                    buf = self.out_buf[:steps].cpu()
-                    for i, q in enumerate(self.seqs):
-                        n_new = pos[i] - self.prompt_lens[i]
-                        if n_new > reported[i]:
-                            stream_cb(q.parent_prompt, q.cand_index,
-                                      buf[reported[i]:n_new, i].tolist())
-                            reported[i] = n_new
-                if bool(self.finished.all()):
-                    break
-            final_pos = self.positions.cpu().tolist()
-            out = self.out_buf.cpu()
-            return [q.output_ids[:1] + out[:final_pos[i] - self.prompt_lens[i],
-                                           i].tolist()
-                    for i, q in enumerate(self.seqs)]
-        return super().run(chunk=chunk, stream_cb=stream_cb)
+    # run() is inherited: the base loop only captures when no graph
+    # exists yet, so a graph captured on a previous wave (same storages)
+    # replays directly, and the padded dummy lanes are born finished so
+    # the retirement/extraction logic never sees them (finished count is
+    # compared against n_pad via retire_at offsetting in SessionCache).
+
+    def run(self, chunk: int = 16, stream_cb=None, retire_at=None):
+        # dummy lanes count as finished: shift the retirement threshold
+        if retire_at is not None:
+            retire_at = retire_at + (self.n_pad - len(self.seqs))
+        return super().run(chunk=chunk, stream_cb=stream_cb,
+                           retire_at=retire_at)
 
 
 class SessionCache:

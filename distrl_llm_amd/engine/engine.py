@@ -401,7 +401,9 @@ class Engine:
     def generate(self, prompts: List[List[int]], sp: SamplingParams,
                  eos_token_id: Optional[int] = None,
                  prefill_token_budget: int = 8192,
-                 stream_cb=None) -> List[List[List[int]]]:
+                 stream_cb=None,
+                 token_limits: Optional[List[List[int]]] = None
+                 ) -> List[List[List[int]]]:
         """Generate sp.n completions per prompt.
 
         prompts: token-id lists. Returns per prompt a list of n output
@@ -420,7 +422,8 @@ class Engine:
             self.fused.refresh()
         try:
             return self._generate_inner(prompts, sp, eos_token_id,
-                                        prefill_token_budget, stream_cb)
+                                        prefill_token_budget, stream_cb,
+                                        token_limits)
         except Exception:
             # a failure mid-generation strands this call's in-flight
             # sequences' KV blocks; generate calls are serialized, so no
@@ -433,7 +436,11 @@ class Engine:
                 self.model.train()
 
     def _generate_inner(self, prompts, sp, eos_token_id, prefill_token_budget,
-                        stream_cb=None):
+                        stream_cb=None, token_limits=None):
+        # token_limits: optional per-prompt lists of per-candidate output
+        # caps (each <= sp.max_tokens) — the EOS-realistic bench mode and
+        # per-request limits use these; admission still reserves the
+        # worst case from sp.max_tokens.
         if sp.n > self.cfg.max_num_seqs:
             # the n-candidate fan-out of one prompt is admitted atomically,
             # so it can never fit — fail with the real reason instead of
@@ -448,7 +455,10 @@ class Engine:
             self.generator.manual_seed(int(sp.seed))
         bs = self.pool.block_size
         max_total = self.cfg.max_seq_length
-        results: List[List[List[int]]] = [[] for _ in prompts]
+        # results indexed by (prompt, candidate): with EOS / per-seq
+        # limits and in-wave retirement, candidates finish out of order —
+        # the contract (like vLLM's) is CANDIDATE order
+        results: List[List[List[int]]] = [[None] * sp.n for _ in prompts]
 
         waiting = list(range(len(prompts)))
         running: List[Sequence] = []
@@ -500,16 +510,29 @@ class Engine:
                 first = OF.sample_tokens(lg, sp.temperature, sp.top_p, sp.top_k,
                                          generator=self.generator)
                 children = self._fork(parent, sp.n, first.tolist())
+                if token_limits is not None:
+                    for ci, c in enumerate(children):
+                        c.max_tokens = token_limits[parent.parent_prompt][ci]
+                elif sp.geom_len_mean:
+                    # EOS-realistic mode: per-candidate exponential caps
+                    uu = torch.rand(len(children), generator=self.generator,
+                                    device=self.device)
+                    lims = (1.0 - float(sp.geom_len_mean)
+                            * uu.clamp_min(1e-9).log()).long() \
+                        .clamp(1, sp.max_tokens)
+                    for c, lim in zip(children, lims.tolist()):
+                        c.max_tokens = int(lim)
                 for c in children:
                     if stream_cb is not None:
                         stream_cb(c.parent_prompt, c.cand_index,
                                   [c.output_ids[0]])
                     if ((eos_token_id is not None
                          and c.output_ids[-1] == eos_token_id)
-                            or len(c.output_ids) >= sp.max_tokens
+                            or len(c.output_ids) >= (c.max_tokens
+                                                     or sp.max_tokens)
                             or c.total_len >= max_total):
                         # still need its KV? No: sequence is done.
-                        results[c.parent_prompt].append(c.output_ids)
+                        results[c.parent_prompt][c.cand_index] = c.output_ids
                         self._finish(c)
                     else:
                         running.append(c)
@@ -521,8 +544,15 @@ class Engine:
                 or force_session):
             # wave-based decode sessions (device state + hipGraph replay;
             # DISTRL_FORCE_SESSION=1 runs the same state machine on CPU
-            # so CI covers the graph step body without a GPU)
-            from .decode_session import DecodeSession
+            # so CI covers the graph step body without a GPU).
+            # In-wave continuous batching: a wave exits early once
+            # `retire_unit` lanes finish; finished lanes are retired,
+            # waiting prompts admitted, and the survivors re-waved (the
+            # session cache keeps the per-padded-size graphs, so a
+            # re-wave costs a reset, not a re-capture). Reference parity:
+            # vLLM's continuous batching inside fast_generate
+            # (distributed_actor.py:147-172).
+            retire_unit = int(os.environ.get("DISTRL_RETIRE_UNIT", "16"))
             while running or waiting:
                 if not running:
                     try_admit()
@@ -532,18 +562,27 @@ class Engine:
                                 "KV pool too small to admit any waiting prompt")
                         break
                 session = self._make_session(running, sp, eos_token_id)
+                retire_at = retire_unit if retire_unit > 0 else None
                 try:
                     with trace_range(f"engine/decode_wave[{len(running)}]"):
-                        outs = session.run(stream_cb=stream_cb)
+                        new_toks, fin = session.run(stream_cb=stream_cb,
+                                                    retire_at=retire_at)
                 finally:
                     if self._session_cache is not None:
                         from .decode_session import CachedDecodeSession
                         if isinstance(session, CachedDecodeSession):
                             self._session_cache.release(session)
-                for q, ids in zip(running, outs):
-                    results[q.parent_prompt].append(ids)
-                    self._finish(q)
-                running = []
+                still = []
+                for q, toks, f in zip(running, new_toks, fin):
+                    q.output_ids.extend(toks)
+                    if f:
+                        results[q.parent_prompt][q.cand_index] = q.output_ids
+                        self._finish(q)
+                    else:
+                        still.append(q)
+                running = still
+                if waiting:
+                    try_admit()  # mid-flight admission into the next wave
             return results
 
         while running or waiting:
@@ -565,10 +604,10 @@ class Engine:
                 if stream_cb is not None:
                     stream_cb(q.parent_prompt, q.cand_index, [t])
                 done = ((eos_token_id is not None and t == eos_token_id)
-                        or len(q.output_ids) >= sp.max_tokens
+                        or len(q.output_ids) >= (q.max_tokens or sp.max_tokens)
                         or q.total_len >= max_total)
                 if done:
-                    results[q.parent_prompt].append(q.output_ids)
+                    results[q.parent_prompt][q.cand_index] = q.output_ids
                     self._finish(q)
                 else:
                     still.append(q)

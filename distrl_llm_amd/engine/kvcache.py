@@ -96,7 +96,7 @@ class Sequence:
 
     __slots__ = ("seq_id", "prompt_ids", "output_ids", "block_table",
                  "context_len", "finished", "parent_prompt",
-                 "last_logits_idx", "cand_index")
+                 "last_logits_idx", "cand_index", "max_tokens")
 
     def __init__(self, seq_id: int, prompt_ids: List[int], parent_prompt: int):
         self.seq_id = seq_id
@@ -108,6 +108,7 @@ class Sequence:
         self.parent_prompt = parent_prompt
         self.last_logits_idx: Optional[int] = None
         self.cand_index = 0  # which of the n candidates of its prompt
+        self.max_tokens: Optional[int] = None  # per-seq cap (None -> sp)
 
     @property
     def total_len(self) -> int:

@@ -98,3 +98,45 @@ def test_session_max_seq_len_clamp(model, force_session):
     res = e.generate([prompt], sp, eos_token_id=None)
     assert len(res[0][0]) == 4  # 64 - 60
     assert e.pool.allocator.num_free == 256
+
+
+def test_in_wave_retirement_and_per_seq_limits(tiny_engine_factory=None):
+    """Session path with in-wave retirement (retire_at) + per-candidate
+    token limits: outputs must equal the eager path's (reference parity:
+    vLLM continuous batching, distributed_actor.py:147-172)."""
+    import os
+    import torch
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    spec = get_spec("tiny-qwen2")
+    model = CausalLM(spec, lora_r=2, lora_alpha=4, dtype=torch.float32,
+                     device=torch.device("cpu"))
+    model.random_init(seed=11)
+    prompts = [[3, 5, 7], [11, 13], [2, 4, 6, 8], [9]]
+    limits = [[3, 9], [1, 5], [7, 2], [4, 4]]
+    sp = SamplingParams(max_tokens=9, temperature=0.0, n=2)
+    outs = {}
+    for mode, env in (("eager", {}),
+                      ("session", {"DISTRL_FORCE_SESSION": "1",
+                                   "DISTRL_RETIRE_UNIT": "2"}),
+                      ("session_cache", {"DISTRL_FORCE_SESSION": "1",
+                                         "DISTRL_GRAPH_CACHE": "1",
+                                         "DISTRL_RETIRE_UNIT": "2"})):
+        for k, v in env.items():
+            os.environ[k] = v
+        try:
+            cfg = EngineConfig(max_seq_length=32, kv_block_size=4,
+                               num_kv_blocks=256, max_num_seqs=8)
+            eng = Engine(model, cfg, device=torch.device("cpu"), seed=5)
+            outs[mode] = eng.generate(prompts, sp, eos_token_id=None,
+                                      token_limits=limits)
+            assert eng.pool.allocator.num_free == 256
+        finally:
+            for k in env:
+                os.environ.pop(k, None)
+    assert outs["eager"] == outs["session"], (outs["eager"], outs["session"])
+    assert outs["eager"] == outs["session_cache"]
+    for pi, per_prompt in enumerate(outs["eager"]):
+        for ci, ids in enumerate(per_prompt):
+            assert len(ids) == limits[pi][ci]  # greedy, no EOS -> exact cap
