@@ -49,15 +49,16 @@ def build_tokenizer_dir(d: str) -> str:
     tok = Tokenizer(BPE(unk_token=None))
     tok.pre_tokenizer = ByteLevel(add_prefix_space=False)
     tok.decoder = ByteLevelDecoder()
-    # the four count_xml tag patterns are single special tokens: the
-    # tokenizer DESIGN choice that makes the format reward discoverable
-    # by exploration (each tag is one sampling event, not a 5-token
-    # coincidence)
-    trainer = BpeTrainer(vocab_size=576, special_tokens=[
-        "<|endoftext|>", "<|im_start|>", "<|im_end|>",
-        "<think>\n", "\n</think>\n", "\n<answer>\n", "\n</answer>",
-    ])
+    trainer = BpeTrainer(vocab_size=572, special_tokens=[
+        "<|endoftext|>", "<|im_start|>", "<|im_end|>"])
     tok.train_from_iterator(corpus, trainer)
+    # the four count_xml tag patterns are single REGULAR tokens (added,
+    # not special — decode(skip_special_tokens=True) must keep them or
+    # the reward never sees them): the tokenizer DESIGN choice that makes
+    # the format reward discoverable by exploration (each tag is one
+    # sampling event, not a 5-token coincidence)
+    tok.add_tokens(["<think>\n", "\n</think>\n", "\n<answer>\n",
+                    "\n</answer>"])
     tok.save(os.path.join(d, "tokenizer.json"))
     template = (
         "{% for message in messages %}"
@@ -126,7 +127,7 @@ def main():
         "lr": args.lr, "max_new_tokens": 24, "max_prompt_tokens": 96,
         "num_candidates": args.cands, "episodes": 1,
         "batch_size": args.batch, "train_batch_size": 16,
-        "temperature": 1.0, "save_every": 10**9, "eval_every": 0,
+        "temperature": 1.5, "save_every": 10**9, "eval_every": 0,
         "model": tok_dir, "dataset": "synthetic",
         "number_of_actors": 0, "number_of_learners": 1,
         "learner": "grpo", "use_vllm": True, "max_lora_rank": 32,
