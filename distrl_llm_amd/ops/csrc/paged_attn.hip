@@ -54,18 +54,17 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   float* denom = q_lds + GROUP * D;                           // padded to 4
   float* gmax = denom + ((GROUP + 3) & ~3);                   // slice maxes
   float* scores = gmax + ((GROUP + 3) & ~3);                  // GROUP*Lpad
-  int* bt = reinterpret_cast<int*>(scores + (size_t)GROUP * Lpad);
+  __hip_bfloat16* v_lds = reinterpret_cast<__hip_bfloat16*>(
+      scores + (size_t)GROUP * Lpad);                         // VTILE*D
 
-  const int* bt_g = block_tables + (int64_t)seq * max_nb;
+  const int* bt = block_tables + (int64_t)seq * max_nb;
   const int64_t kv_row = (int64_t)KV * D;
 
-  // ---- stage Q (group heads of this kv head) + the seq's block table
-  // into LDS (the PV loop walks the table per token; LDS beats L1) ----
+  // ---- stage Q (group heads of this kv head) into LDS as fp32 ----
   for (int i = tid; i < GROUP * D; i += blockDim.x) {
     const int h = kv * GROUP + i / D;
     q_lds[i] = bf2f(q[(int64_t)seq * q_row_stride + h * D + i % D]) * scale;
   }
-  for (int i = tid; i < max_nb; i += blockDim.x) bt[i] = bt_g[i];
   __syncthreads();
 
   // ---- phase 1: one token per lane, per-head partials in VGPRs ----
@@ -119,32 +118,36 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   }
   __syncthreads();
 
-  // ---- phase 3: PV accumulation with DIRECT global V reads ----
-  // The old 32-token LDS V tiles cost 2 barriers per tile (~48 barriers
-  // per 776-token slice: SQ_WAIT_ANY was 73% of wave cycles). Reading V
-  // rows directly instead: the 16 d-chunk threads of one head read one
-  // contiguous 256 B row (coalesced), the GROUP heads re-read it from
-  // L2, and the loop has no barriers at all.
+  // ---- phase 3: PV accumulation through LDS V tiles ----
   constexpr int DV = D / 8;             // bf16x8 units per row
   constexpr int UNITS = GROUP * DV;     // <= 16*16 = 256
   float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   const int u = tid;                    // one unit per thread (tid < UNITS)
   const int uh = u / DV, ud = u % DV;
 
-  if (u < UNITS) {
-    const float* ps = scores + (size_t)uh * Lpad;
-    const __hip_bfloat16* vbase = vcache + (int64_t)kv * D + ud * 8;
-    #pragma unroll 4
-    for (int j = 0; j < Ls; ++j) {
-      const int tt = c0 + j;
+  for (int base = 0; base < Ls; base += VTILE) {
+    const int tile = min(VTILE, Ls - base);
+    // stage V rows [base, base+tile) for this kv head
+    for (int i = tid; i < tile * DV; i += blockDim.x) {
+      const int tt = c0 + base + i / DV;
       const int64_t row = (int64_t)bt[tt / block_size] * block_size
                           + tt % block_size;
-      const float p = ps[j];
-      const bf16x8 vv = *reinterpret_cast<const bf16x8*>(vbase
-                                                         + row * kv_row);
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) acc[e] += p * bf2f(vv.v[e]);
+      reinterpret_cast<bf16x8*>(v_lds)[i] =
+          *reinterpret_cast<const bf16x8*>(
+              vcache + row * kv_row + (int64_t)kv * D + (i % DV) * 8);
     }
+    __syncthreads();
+    if (u < UNITS) {
+      const float* ps = scores + (size_t)uh * Lpad + base;
+      #pragma unroll 4
+      for (int j = 0; j < tile; ++j) {
+        const float p = ps[j];
+        const bf16x8 vv = reinterpret_cast<const bf16x8*>(v_lds)[j * DV + ud];
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) acc[e] += p * bf2f(vv.v[e]);
+      }
+    }
+    __syncthreads();
   }
 
   if (u < UNITS) {
@@ -272,7 +275,7 @@ torch::Tensor paged_attention_decode_strided(
   const int slice_len = (max_ctx + cs - 1) / cs;
   const int Lpad = slice_len + 4;  // +pad to stagger LDS banks across heads
   size_t smem = (size_t)group * D * 4 + 2 * ((group + 3) & ~3) * 4
-                + (size_t)group * Lpad * 4 + (size_t)max_nb * 4 + 16;
+                + (size_t)group * Lpad * 4 + (size_t)VTILE * D * 2;
   TORCH_CHECK(smem <= 160 * 1024,
               "context too long for single-pass decode kernel: ", max_ctx);
 
