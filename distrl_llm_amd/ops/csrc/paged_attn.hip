@@ -118,70 +118,36 @@ void paged_decode_kernel(const __hip_bfloat16* __restrict__ q,      // (N,H,D)
   }
   __syncthreads();
 
-  // ---- phase 3: PV accumulation through DOUBLE-BUFFERED LDS V tiles.
-  // The single-buffer version pays 2 barriers per 32-token tile (~48 per
-  // 776-token slice; SQ_WAIT_ANY 73% of wave cycles); a direct-global-V
-  // variant measured 1.8x SLOWER (per-token load chains). Here each
-  // thread issues its next-tile V loads into registers BEFORE computing
-  // the current tile (guide T14 async-stage split), writes them to the
-  // spare buffer after, and crosses ONE barrier per tile with the HBM
-  // latency hidden under the PV math.
+  // ---- phase 3: PV accumulation through LDS V tiles ----
   constexpr int DV = D / 8;             // bf16x8 units per row
   constexpr int UNITS = GROUP * DV;     // <= 16*16 = 256
-  constexpr int SU = (VTILE * DV + 255) / 256;  // staged units per thread
   float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   const int u = tid;                    // one unit per thread (tid < UNITS)
   const int uh = u / DV, ud = u % DV;
-  __hip_bfloat16* vbuf[2] = {v_lds, v_lds + (size_t)VTILE * D};
 
-  if (Ls > 0) {
-    bf16x8 st[SU];
-    #define PA_ISSUE(BASE)                                                  \
-      _Pragma("unroll")                                                     \
-      for (int it = 0; it < SU; ++it) {                                     \
-        const int i = tid + it * 256;                                       \
-        if (i < DV * min(VTILE, Ls - (BASE))) {                             \
-          const int tt = c0 + (BASE) + i / DV;                              \
-          const int64_t row = (int64_t)bt[tt / block_size] * block_size     \
-                              + tt % block_size;                            \
-          st[it] = *reinterpret_cast<const bf16x8*>(                        \
-              vcache + row * kv_row + (int64_t)kv * D + (i % DV) * 8);      \
-        }                                                                   \
-      }
-    #define PA_WRITE(BASE, BUF)                                             \
-      _Pragma("unroll")                                                     \
-      for (int it = 0; it < SU; ++it) {                                     \
-        const int i = tid + it * 256;                                       \
-        if (i < DV * min(VTILE, Ls - (BASE)))                               \
-          reinterpret_cast<bf16x8*>(vbuf[BUF])[i] = st[it];                 \
-      }
-    PA_ISSUE(0)
-    PA_WRITE(0, 0)
-    __syncthreads();
-    int cur = 0;
-    for (int base = 0; base < Ls; base += VTILE) {
-      const int tile = min(VTILE, Ls - base);
-      const bool more = base + VTILE < Ls;
-      if (more) PA_ISSUE(base + VTILE)
-      if (u < UNITS) {
-        const float* ps = scores + (size_t)uh * Lpad + base;
-        #pragma unroll 4
-        for (int j = 0; j < tile; ++j) {
-          const float p = ps[j];
-          const bf16x8 vv =
-              reinterpret_cast<const bf16x8*>(vbuf[cur])[j * DV + ud];
-          #pragma unroll
-          for (int e = 0; e < 8; ++e) acc[e] += p * bf2f(vv.v[e]);
-        }
-      }
-      if (more) {
-        PA_WRITE(base + VTILE, cur ^ 1)
-        cur ^= 1;
-      }
-      __syncthreads();
+  for (int base = 0; base < Ls; base += VTILE) {
+    const int tile = min(VTILE, Ls - base);
+    // stage V rows [base, base+tile) for this kv head
+    for (int i = tid; i < tile * DV; i += blockDim.x) {
+      const int tt = c0 + base + i / DV;
+      const int64_t row = (int64_t)bt[tt / block_size] * block_size
+                          + tt % block_size;
+      reinterpret_cast<bf16x8*>(v_lds)[i] =
+          *reinterpret_cast<const bf16x8*>(
+              vcache + row * kv_row + (int64_t)kv * D + (i % DV) * 8);
     }
-    #undef PA_ISSUE
-    #undef PA_WRITE
+    __syncthreads();
+    if (u < UNITS) {
+      const float* ps = scores + (size_t)uh * Lpad + base;
+      #pragma unroll 4
+      for (int j = 0; j < tile; ++j) {
+        const float p = ps[j];
+        const bf16x8 vv = reinterpret_cast<const bf16x8*>(v_lds)[j * DV + ud];
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) acc[e] += p * bf2f(vv.v[e]);
+      }
+    }
+    __syncthreads();
   }
 
   if (u < UNITS) {
@@ -309,7 +275,7 @@ torch::Tensor paged_attention_decode_strided(
   const int slice_len = (max_ctx + cs - 1) / cs;
   const int Lpad = slice_len + 4;  // +pad to stagger LDS banks across heads
   size_t smem = (size_t)group * D * 4 + 2 * ((group + 3) & ~3) * 4
-                + (size_t)group * Lpad * 4 + (size_t)VTILE * D * 2 * 2;
+                + (size_t)group * Lpad * 4 + (size_t)VTILE * D * 2;
   TORCH_CHECK(smem <= 160 * 1024,
               "context too long for single-pass decode kernel: ", max_ctx);
 
@@ -488,7 +454,7 @@ torch::Tensor prefill_attention(torch::Tensor q, torch::Tensor k,
   if (n_tiles == 0) return out;
   const int Lpad = (int)max_len + 4;
   size_t smem = 16 * D * 4 + 16 * 4 + (size_t)16 * Lpad * 4
-                + (size_t)VTILE * D * 2 * 2;
+                + (size_t)VTILE * D * 2;
   TORCH_CHECK(smem <= 160 * 1024, "prompt too long for prefill kernel: ",
               max_len);
   dim3 grid(n_tiles, H), block(256);
