@@ -272,6 +272,10 @@ torch::Tensor paged_attention_decode_strided(
   while (cs * 2 <= 8 && N * KV * cs < 1024
          && (max_ctx + cs * 2 - 1) / (cs * 2) >= 128)
     cs *= 2;
+  if (const char* e = getenv("DISTRL_PAGED_CS")) {
+    cs = std::max(1, std::min(atoi(e), 16));
+    while (cs > 1 && (max_ctx + cs - 1) / cs < 32) cs /= 2;
+  }
   const int slice_len = (max_ctx + cs - 1) / cs;
   const int Lpad = slice_len + 4;  // +pad to stagger LDS banks across heads
   size_t smem = (size_t)group * D * 4 + 2 * ((group + 3) & ~3) * 4
