@@ -268,8 +268,11 @@ torch::Tensor paged_attention_decode_strided(
   const int max_ctx = max_nb * block_size;
   // flash-decoding context split: fill the chip (N*KV blocks alone leave
   // most CUs idle at decode batch sizes) and shorten the per-block chain
+  // sweep-tuned: ~2048 blocks beats ~1024 at decode batch 160 (113 vs
+  // 137 us at ctx 800 — smaller LDS score tiles double the blocks/CU);
+  // cs=8+ loses to combine overhead (profiles/)
   int cs = 1;
-  while (cs * 2 <= 8 && N * KV * cs < 1024
+  while (cs * 2 <= 8 && N * KV * cs < 2048
          && (max_ctx + cs * 2 - 1) / (cs * 2) >= 128)
     cs *= 2;
   if (const char* e = getenv("DISTRL_PAGED_CS")) {
