@@ -2,6 +2,7 @@
 """BASELINE config 5 readiness: Qwen2.5-32B 4-bit on one MI355X — init,
 quantize, fused-nf4 decode session, learner micro-step, memory report."""
 
+import argparse
 import sys
 import time
 
@@ -11,6 +12,10 @@ sys.path.insert(0, ".")
 
 
 def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="unsloth/Qwen2.5-32B-Instruct-bnb-4bit")
+    ap.add_argument("--pool", type=float, default=0.3)
+    args = ap.parse_args()
     from distrl_llm_amd.config import EngineConfig, SamplingParams
     from distrl_llm_amd.engine import Engine
     from distrl_llm_amd.models import CausalLM, get_spec
@@ -18,7 +23,7 @@ def main():
     from distrl_llm_amd.utils.tokenizer import ByteTokenizer
 
     dev = torch.device("cuda:0")
-    spec = get_spec("unsloth/Qwen2.5-32B-Instruct-bnb-4bit")
+    spec = get_spec(args.model)
     t0 = time.time()
     model = CausalLM(spec, lora_r=32, lora_alpha=16, dtype=torch.bfloat16,
                      device=dev).random_init(3407)
@@ -32,7 +37,7 @@ def main():
     tok = ByteTokenizer(vocab_size=spec.vocab_size)
     t0 = time.time()
     engine = Engine(model, EngineConfig(max_seq_length=1550,
-                                        gpu_memory_utilization=0.3),
+                                        gpu_memory_utilization=args.pool),
                     device=dev, seed=0)
     print(f"engine init {time.time()-t0:.1f}s, nf4_path={engine.fused.nf4}, "
           f"kv blocks {engine.pool.num_blocks}", flush=True)
