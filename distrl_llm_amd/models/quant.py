@@ -40,6 +40,16 @@ def prepack_nf4_fragments(packed: torch.Tensor, absmax: torch.Tensor,
     """
     assert N % 32 == 0 and K % 64 == 0
     device = packed.device
+    # pad K to the kernel's widest super-panel (512) — padded columns get
+    # absmax 0, so any nibble dequantizes to 0 and contributes nothing
+    # (needed for shapes like Qwen2.5-72B's intermediate 29568 = 128*231)
+    Kp = (K + 511) // 512 * 512
+    if Kp != K:
+        pb = packed.contiguous().view(N, K // 2)
+        packed = torch.cat([pb, pb.new_zeros(N, (Kp - K) // 2)], 1).reshape(-1)
+        am = absmax.contiguous().view(N, K // 64)
+        absmax = torch.cat([am, am.new_zeros(N, (Kp - K) // 64)], 1).reshape(-1)
+        K = Kp
     ntiles, ksteps = N // 16, K // 32
     ngr, nkb = N // 32, K // 64
     pd = packed.contiguous().view(N, K // 2).view(torch.int32)  # (N, K/8)

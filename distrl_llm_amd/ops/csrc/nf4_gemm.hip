@@ -450,6 +450,14 @@ torch::Tensor nf4_gemm(torch::Tensor x, torch::Tensor w4f, torch::Tensor amaxf,
   TORCH_CHECK(w4f.scalar_type() == at::kInt || w4f.scalar_type() == at::kUInt32);
   const int M = x.size(0);
   TORCH_CHECK(x.size(1) == K);
+  {
+    // the prepack pads K to a multiple of 512 (zero absmax): mirror it
+    const int64_t Kp = (K + 511) / 512 * 512;
+    if (Kp != K) {
+      x = torch::constant_pad_nd(x, {0, Kp - K}, 0);
+      K = Kp;
+    }
+  }
   TORCH_CHECK(N % 128 == 0, "nf4_gemm: N must be a multiple of 128, got ", N);
   TORCH_CHECK(K % 64 == 0);
   auto y = torch::empty({(int64_t)M, N}, x.options());
