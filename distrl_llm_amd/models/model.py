@@ -175,19 +175,34 @@ class CausalLM(nn.Module):
         return self
 
     @torch.no_grad()
-    def quantize_nf4_(self, block_size: int = 64) -> "CausalLM":
+    def quantize_nf4_(self, block_size: int = 64,
+                      keep_bf16: Optional[bool] = None) -> "CausalLM":
         """nf4-quantize every base projection weight in place: the bf16
         weight is replaced with its quantize->dequantize image (so training
         numerics match the 4-bit model) and the packed nf4 + absmax sidecar
-        is attached for the engine's fused 4-bit GEMM kernels."""
+        is attached for the engine's fused 4-bit GEMM kernels.
+
+        keep_bf16=False (default for >16B models on GPU) FREES the bf16
+        base image instead — every base projection (learner forward,
+        prefill) then runs through the fused nf4 GEMM with on-the-fly
+        dequant for dX (SURVEY.md §2.4-B "learner nf4 GEMM" row; removes
+        the 65 GiB redundant image at 32B / the 145 GiB one at 72B that
+        OOM'd the 72B dual-role trial)."""
+        if keep_bf16 is None:
+            dev = next(self.parameters()).device
+            n_params = sum(p.numel() for p in self.parameters())
+            keep_bf16 = not (dev.type == "cuda" and n_params > 16e9)
         for mod in self.modules():
             if isinstance(mod, LoRALinear):
                 packed, absmax = R.quantize_nf4(mod.weight, block_size)
                 mod.weight_nf4 = packed
                 mod.weight_absmax = absmax
-                mod.weight.copy_(R.dequantize_nf4(
-                    packed, absmax, mod.weight.shape, block_size,
-                    dtype=mod.weight.dtype))
+                if keep_bf16:
+                    mod.weight.copy_(R.dequantize_nf4(
+                        packed, absmax, mod.weight.shape, block_size,
+                        dtype=mod.weight.dtype))
+                else:
+                    mod.free_base_to_nf4_()
         return self
 
     # ----------------------------------------------------------- forward

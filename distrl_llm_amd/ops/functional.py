@@ -177,6 +177,42 @@ def sample_tokens(logits: torch.Tensor, temperature: float, top_p: float,
     return R.sample_tokens(logits, temperature, top_p, top_k, generator=generator)
 
 
+# ----------------------------------------------------- nf4 base linear
+
+class _NF4LinearFn(torch.autograd.Function):
+    """Base projection through the fused nf4-dequant MFMA GEMM when the
+    bf16 base image has been freed (LoRALinear.free_base_to_nf4_; big-
+    model learner path — SURVEY.md §2.4-B "learner nf4 GEMM" row). The
+    base is frozen, so backward only needs dX = dY @ W with W dequanted
+    on the fly (transient, never resident)."""
+
+    @staticmethod
+    def forward(ctx, x, w4f, amaxf, packed, absmax, bias, N, K):
+        ext = _require_ext("nf4_linear")
+        x2 = x.reshape(-1, K)
+        if x2.dtype != torch.bfloat16:
+            x2 = x2.to(torch.bfloat16)
+        y = ext.nf4_gemm(x2.contiguous(), w4f, amaxf, bias, None, None,
+                         N, K, 0)
+        ctx.save_for_backward(packed, absmax)
+        ctx.NK = (N, K)
+        ctx.in_dtype = x.dtype
+        return y.view(*x.shape[:-1], N)
+
+    @staticmethod
+    def backward(ctx, dy):
+        packed, absmax = ctx.saved_tensors
+        N, K = ctx.NK
+        W = R.dequantize_nf4(packed, absmax, (N, K), 64, torch.bfloat16)
+        dx = (dy.reshape(-1, N).to(torch.bfloat16) @ W).to(ctx.in_dtype)
+        return (dx.view(*dy.shape[:-1], K), None, None, None, None, None,
+                None, None)
+
+
+def nf4_linear(x, w4f, amaxf, packed, absmax, bias, N: int, K: int):
+    return _NF4LinearFn.apply(x, w4f, amaxf, packed, absmax, bias, N, K)
+
+
 # --------------------------------------------------- flash attention
 
 class _FlashAttnFn(torch.autograd.Function):

@@ -187,3 +187,34 @@ def test_long_prompt_prefill_no_cap(setup):
     expected = _naive_greedy(model, prompt, 4)
     agree = sum(a == b for a, b in zip(res[0][0], expected))
     assert agree >= 3, (res[0][0], expected)
+
+
+def test_nf4_freed_base_learner_parity():
+    """free_base_to_nf4_ (bf16 image dropped, fused nf4 GEMM forward +
+    on-the-fly dequant dX): forward and LoRA grads must match the
+    materialized nf4-image path (SURVEY.md §2.4-B learner nf4 GEMM)."""
+    from distrl_llm_amd.models import CausalLM, get_spec
+    spec = get_spec("small-qwen2")
+    dev = torch.device("cuda:0")
+    outs = {}
+    for freed in (False, True):
+        torch.manual_seed(0)
+        model = CausalLM(spec, lora_r=8, lora_alpha=16,
+                         dtype=torch.bfloat16, device=dev)
+        model.random_init(seed=3)
+        model.quantize_nf4_(keep_bf16=not freed)
+        ids = torch.randint(1, 500, (2, 48), device=dev)
+        logits = model(ids)
+        loss = logits.float().log_softmax(-1).mean()
+        loss.backward()
+        g = [m.lora_A.grad.clone() for m in model.modules()
+             if hasattr(m, "lora_A") and m.lora_A is not None
+             and m.lora_A.grad is not None]
+        outs[freed] = (logits.detach().float(), g)
+    lo, go = outs[False]
+    lf, gf = outs[True]
+    assert torch.allclose(lo, lf, atol=2e-2, rtol=2e-2), (lo - lf).abs().max()
+    for a, b in zip(go, gf):
+        err = (a - b).abs().max()
+        ref = a.abs().max().clamp_min(1e-6)
+        assert err <= 0.05 * ref + 1e-5, (err, ref)
