@@ -68,8 +68,14 @@ class FusedWeights:
         # GEMM path streams 4.3x fewer weight bytes and is selected with
         # DISTRL_DECODE_NF4=1 (kept fully tested).
         import os
+        freed = (q0.weight_nf4 is not None and q0.weight.numel() == 0)
         env = os.environ.get("DISTRL_DECODE_NF4")
-        if env is not None:
+        if freed:
+            # the bf16 base image was dropped (free_base_to_nf4_, big
+            # models) — there is nothing to merge, the fused nf4 path is
+            # the only decode path
+            want_nf4 = True
+        elif env is not None:
             want_nf4 = use_nf4 and env == "1"
         else:
             # default: merged cache for models whose bf16 copy is cheap;
@@ -79,6 +85,10 @@ class FusedWeights:
             want_nf4 = use_nf4 and n_params > 16e9
         self.nf4 = bool(want_nf4 and q0.weight_nf4 is not None
                         and q0.weight.is_cuda and shapes_ok)
+        if freed and not self.nf4:
+            raise RuntimeError(
+                "base image freed but the fused nf4 decode path is "
+                "unavailable for this shape — cannot fall back to merged")
         self.lora_r = q0.r
 
     @staticmethod
