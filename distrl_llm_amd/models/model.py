@@ -189,9 +189,12 @@ class CausalLM(nn.Module):
         the 65 GiB redundant image at 32B / the 145 GiB one at 72B that
         OOM'd the 72B dual-role trial)."""
         if keep_bf16 is None:
+            # auto: free only when the bf16 image threatens residency
+            # (72B: image 145 GiB; 32B keeps its image — measured round 1
+            # at 199 GiB peak with the faster hipBLASLt learner path)
             dev = next(self.parameters()).device
             n_params = sum(p.numel() for p in self.parameters())
-            keep_bf16 = not (dev.type == "cuda" and n_params > 16e9)
+            keep_bf16 = not (dev.type == "cuda" and n_params > 40e9)
         for mod in self.modules():
             if isinstance(mod, LoRALinear):
                 packed, absmax = R.quantize_nf4(mod.weight, block_size)
