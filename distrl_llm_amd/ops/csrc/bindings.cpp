@@ -35,6 +35,9 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kcache,
 torch::Tensor sample_tokens(torch::Tensor logits, double temperature,
                             double top_p, int64_t top_k, torch::Tensor seeds,
                             torch::Tensor step);
+torch::Tensor sample_tokens2(torch::Tensor logits, double temperature,
+                             double top_p, int64_t top_k, torch::Tensor seeds,
+                             torch::Tensor step);
 std::vector<torch::Tensor> logprob_lse_fwd(torch::Tensor logits,
                                            torch::Tensor targets);
 torch::Tensor logprob_loss_bwd(torch::Tensor logits, torch::Tensor targets,
@@ -78,6 +81,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "paged GQA decode attention");
   m.def("sample_tokens", &sample_tokens,
         "fused temperature/top-k/top-p categorical sampling");
+  m.def("sample_tokens2", &sample_tokens2,
+        "two-stage fused sampler (slice-parallel; same semantics/RNG)");
   m.def("logprob_lse_fwd", &logprob_lse_fwd,
         "per-token log p(target) + LSE, one streaming pass");
   m.def("logprob_loss_bwd", &logprob_loss_bwd,

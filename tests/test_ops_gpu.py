@@ -135,7 +135,8 @@ def test_paged_attention_decode(ext, D, H, KV, ctxs):
 
 # -------------------------------------------------------------- sampling
 
-def test_sample_tokens_distribution(ext):
+@pytest.mark.parametrize("impl", ["sample_tokens", "sample_tokens2"])
+def test_sample_tokens_distribution(ext, impl):
     """Statistical check: a peaked 3-token distribution sampled many times
     matches the renormalized top-p categorical within tolerance."""
     torch.manual_seed(6)
@@ -148,7 +149,8 @@ def test_sample_tokens_distribution(ext):
     rows = logits.expand(n, V).contiguous()
     seeds = torch.arange(n, device=_dev(), dtype=torch.int64) * 7919 + 13
     step = torch.zeros(1, dtype=torch.int64, device=_dev())
-    out = ext.sample_tokens(rows, 1.0, 1.0, 0, seeds, step)
+    fn = getattr(ext, impl)
+    out = fn(rows, 1.0, 1.0, 0, seeds, step)
     counts = torch.bincount(out.cpu(), minlength=V).float() / n
     assert abs(counts[10] - 0.6) < 0.04
     assert abs(counts[20] - 0.3) < 0.04
@@ -157,24 +159,25 @@ def test_sample_tokens_distribution(ext):
 
     # top_p = 0.65 keeps only token 10 (0.6 < 0.65 needs token 20 too: the
     # reference keeps the crossing token) -> {10, 20} renormalized
-    out2 = ext.sample_tokens(rows, 1.0, 0.65, 0, seeds, step)
+    out2 = fn(rows, 1.0, 0.65, 0, seeds, step)
     c2 = torch.bincount(out2.cpu(), minlength=V).float() / n
     assert c2[30] == 0.0
     assert abs(c2[10] - 0.6 / 0.9) < 0.05
 
     # top_k = 1 is greedy
-    out3 = ext.sample_tokens(rows, 1.0, 1.0, 1, seeds, step)
+    out3 = fn(rows, 1.0, 1.0, 1, seeds, step)
     assert (out3 == 10).all()
 
 
-def test_sample_tokens_temperature(ext):
+@pytest.mark.parametrize("impl", ["sample_tokens", "sample_tokens2"])
+def test_sample_tokens_temperature(ext, impl):
     V = 128
     logits = torch.zeros(2000, V, device=_dev())
     logits[:, 5] = 2.0
     seeds = torch.arange(2000, device=_dev(), dtype=torch.int64)
     step = torch.zeros(1, dtype=torch.int64, device=_dev())
     # low temperature sharpens: nearly all mass on token 5
-    out = ext.sample_tokens(logits, 0.05, 1.0, 0, seeds, step)
+    out = getattr(ext, impl)(logits, 0.05, 1.0, 0, seeds, step)
     assert (out == 5).float().mean() > 0.99
 
 
