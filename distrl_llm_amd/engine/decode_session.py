@@ -98,11 +98,11 @@ class DecodeSession:
             if logits.is_cuda:
                 from ..ops.build import get_extension
                 ext = get_extension()
-                sampled = ext.sample_tokens(logits,
-                                            float(self.sp.temperature),
-                                            float(self.sp.top_p),
-                                            int(self.sp.top_k), self.seeds,
-                                            self.step_idx)
+                sampled = ext.sample_tokens2(logits,
+                                             float(self.sp.temperature),
+                                             float(self.sp.top_p),
+                                             int(self.sp.top_k), self.seeds,
+                                             self.step_idx)
             else:
                 # CPU session (DISTRL_FORCE_SESSION=1 CI coverage of the
                 # state machine): reference sampler on the engine stream

@@ -172,8 +172,9 @@ def sample_tokens(logits: torch.Tensor, temperature: float, top_p: float,
                                       generator=generator)
             if step is None:
                 step = torch.zeros(1, dtype=torch.int64, device=logits.device)
-            return ext.sample_tokens(logits.contiguous(), float(temperature),
-                                     float(top_p), int(top_k), seeds, step)
+            return ext.sample_tokens2(logits.contiguous(),
+                                      float(temperature), float(top_p),
+                                      int(top_k), seeds, step)
     return R.sample_tokens(logits, temperature, top_p, top_k, generator=generator)
 
 
