@@ -98,11 +98,16 @@ class DecodeSession:
             if logits.is_cuda:
                 from ..ops.build import get_extension
                 ext = get_extension()
-                sampled = ext.sample_tokens2(logits,
-                                             float(self.sp.temperature),
-                                             float(self.sp.top_p),
-                                             int(self.sp.top_k), self.seeds,
-                                             self.step_idx)
+                # two-stage sampler fills the chip at decode batch
+                # sizes (160: 306 -> 242 us; 64: 299 -> 135 us); the
+                # single-kernel version wins again once B alone covers
+                # the CUs (512: 576 vs 626 us)
+                fn = (ext.sample_tokens2 if logits.shape[0] <= 256
+                      else ext.sample_tokens)
+                sampled = fn(logits, float(self.sp.temperature),
+                             float(self.sp.top_p),
+                             int(self.sp.top_k), self.seeds,
+                             self.step_idx)
             else:
                 # CPU session (DISTRL_FORCE_SESSION=1 CI coverage of the
                 # state machine): reference sampler on the engine stream

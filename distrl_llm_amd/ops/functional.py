@@ -172,9 +172,10 @@ def sample_tokens(logits: torch.Tensor, temperature: float, top_p: float,
                                       generator=generator)
             if step is None:
                 step = torch.zeros(1, dtype=torch.int64, device=logits.device)
-            return ext.sample_tokens2(logits.contiguous(),
-                                      float(temperature), float(top_p),
-                                      int(top_k), seeds, step)
+            fn = (ext.sample_tokens2 if logits.shape[0] <= 256
+                  else ext.sample_tokens)
+            return fn(logits.contiguous(), float(temperature),
+                      float(top_p), int(top_k), seeds, step)
     return R.sample_tokens(logits, temperature, top_p, top_k, generator=generator)
 
 
