@@ -297,11 +297,13 @@ class Engine:
 
         # per-step LoRA-u pool (zeroed ONCE per step; the split-K lora_u
         # kernels accumulate into per-(layer, site) slices)
+        hybrid_down = (not nf4) and getattr(self.fused, "hybrid_down", False)
         u_pool = None
         site_off = {}
-        if nf4 and r > 0:
+        if (nf4 or hybrid_down) and r > 0:
             off = 0
-            for site in ("qkv", "o", "gateup", "down"):
+            for site in (("qkv", "o", "gateup", "down") if nf4
+                         else ("down",)):
                 rp = getattr(lws[0], f"{site}_r")
                 site_off[site] = (off, rp)
                 off += rp
@@ -315,8 +317,10 @@ class Engine:
             u_pool.zero_()
 
         def proj(x, lw, li, site, N_out, K_in):
-            """One fused projection: nf4 GEMM + LoRA (or merged bf16)."""
-            if not nf4:
+            """One fused projection: nf4 GEMM + LoRA (or merged bf16;
+            the down site runs the fused nf4 kernel even in merged mode —
+            it beats hipBLASLt on that deep-K shape)."""
+            if not nf4 and not (hybrid_down and site == "down"):
                 if site == "qkv":
                     return F.linear(x, lw.qkv_w, lw.qkv_b)
                 return F.linear(x, getattr(lw, f"{site}_w"))

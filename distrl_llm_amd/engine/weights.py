@@ -89,6 +89,14 @@ class FusedWeights:
             raise RuntimeError(
                 "base image freed but the fused nf4 decode path is "
                 "unavailable for this shape — cannot fall back to merged")
+        # hybrid: even when the merged-bf16 cache wins overall, the fused
+        # nf4 kernel BEATS hipBLASLt on the deep-K down-projection
+        # (61+13 vs 80 us/layer at decode batch 160 — profiles/r02) —
+        # run just that site through it
+        self.hybrid_down = bool(
+            (not self.nf4) and q0.weight_nf4 is not None
+            and q0.weight.is_cuda and shapes_ok
+            and os.environ.get("DISTRL_HYBRID_DOWN", "1") == "1")
         self.lora_r = q0.r
 
     @staticmethod
@@ -113,7 +121,7 @@ class FusedWeights:
             setattr(lw, f"{name}_w4", w4f)
             setattr(lw, f"{name}_amax", amaxf)
 
-    def _refresh_nf4_adapters(self, lw: "LayerWeights", layer):
+    def _refresh_nf4_adapters(self, lw: "LayerWeights", layer, sites=None):
         """Per-round: stack the A matrices and build the block-diagonal,
         scale-folded B matrix per fused site, prepacked into fragments."""
         from ..models.quant import prepack_bf16_fragments
@@ -122,6 +130,8 @@ class FusedWeights:
                            ("o", [at.o_proj]),
                            ("gateup", [mlp.gate_proj, mlp.up_proj]),
                            ("down", [mlp.down_proj])):
+            if sites is not None and name not in sites:
+                continue
             A = torch.cat([m.lora_A.detach() for m in mods], dim=0)  # (r_tot, K)
             r_tot = A.shape[0]
             r_pad = (r_tot + 31) // 32 * 32  # kernel needs rank % 32 == 0
@@ -172,6 +182,14 @@ class FusedWeights:
                 self._refresh_nf4_adapters(lw, layer)
             else:
                 self._refresh_merged(lw, layer, first)
+                if self.hybrid_down:
+                    if first:
+                        from ..models.quant import prepack_nf4_fragments
+                        packed, absmax, N, K = self._nf4_cat(
+                            [layer.mlp.down_proj])
+                        lw.down_w4, lw.down_amax = prepack_nf4_fragments(
+                            packed, absmax, N, K)
+                    self._refresh_nf4_adapters(lw, layer, sites=("down",))
         self._built = True
 
     def _refresh_merged(self, lw: "LayerWeights", layer, first: bool):
