@@ -26,7 +26,7 @@
 
 namespace {
 
-constexpr int VTILE = 32;        // V tokens staged per LDS tile
+constexpr int VTILE = 64;        // V tokens staged per LDS tile
 
 template <int D, int GROUP>
 __global__ __launch_bounds__(256)
