@@ -159,6 +159,45 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
   if (kb0 + 1 < kb_end) LOAD_WCHUNK(kb0 + 1, 1);
   if (kb0 + 2 < kb_end) LOAD_WCHUNK(kb0 + 2, 2);
 
+  // double-buffered PRE-DEQUANTED B fragments: chunk kc's MFMAs consume
+  // bfr[kc&1] while chunk kc+1 dequants into bfr[(kc+1)&1] — the
+  // ds_bpermute latency hides under the matrix pipe instead of gating it
+  bf16v8 bfr[2][4];
+  #define DEQUANT_CHUNK(RS, PB)                                           \
+    {                                                                     \
+      if constexpr (DBG == 1) {                                           \
+        _Pragma("unroll")                                                 \
+        for (int q_ = 0; q_ < 4; ++q_) {                                  \
+          const uint32_t wbits_ = wb[RS][q_];                             \
+          _Pragma("unroll")                                               \
+          for (int j_ = 0; j_ < 8; ++j_)                                  \
+            bfr[PB][q_][j_] = (__bf16)(am[RS][q_ & 1]                     \
+                                       + (float)(wbits_ & 1));            \
+        }                                                                 \
+      } else {                                                            \
+        float cval_[4][8];                                                \
+        _Pragma("unroll")                                                 \
+        for (int q_ = 0; q_ < 4; ++q_) {                                  \
+          const uint32_t wbits_ = wb[RS][q_];                             \
+          _Pragma("unroll")                                               \
+          for (int j_ = 0; j_ < 8; ++j_) {                                \
+            const int addr_ = (j_ == 0) ? (int)(wbits_ << 2)              \
+                                        : (int)(wbits_ >> (4 * j_ - 2));  \
+            cval_[q_][j_] = __builtin_bit_cast(                           \
+                float, __builtin_amdgcn_ds_bpermute(addr_, lut_bits));    \
+          }                                                               \
+        }                                                                 \
+        _Pragma("unroll")                                                 \
+        for (int q_ = 0; q_ < 4; ++q_) {                                  \
+          const float amv_ = am[RS][q_ & 1];                              \
+          _Pragma("unroll")                                               \
+          for (int j_ = 0; j_ < 8; ++j_)                                  \
+            bfr[PB][q_][j_] = (__bf16)(cval_[q_][j_] * amv_);             \
+        }                                                                 \
+      }                                                                   \
+    }
+  if (kb0 < kb_end) DEQUANT_CHUNK(0, 0);
+
   // x is staged in SUPER-panels of SK columns: one barrier pair per
   // SK/64 weight chunks, so the chunk loop in between runs barrier-free
   // and the one-chunk-ahead weight prefetch is never drained (hipcc puts
@@ -221,10 +260,12 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
       // panel-relative phase aligned, so slot = kc & 3 in the unrolled
       // body (no register-shuffle rotation between chunks)
       if (kb + 3 < kb_end) LOAD_WCHUNK(kb + 3, (kc + 3) & 3);
-      const int cur = kc & 3;
+
+      // ---- A) MFMAs of the CURRENT chunk from the PRE-DEQUANTED
+      // fragments (bfr[kc&1], produced one chunk ago — the bpermute
+      // latency of the dequant overlaps the previous chunk's MFMAs)
       #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        // A fragments for every m-tile
         bf16v8 afrag[MT];
         #pragma unroll
         for (int mt = 0; mt < MT; ++mt) {
@@ -239,58 +280,22 @@ void nf4_gemm_kernel(const __hip_bfloat16* __restrict__ x,   // (M, K)
             afrag[mt] = lds_read_frag(x_lds, off);
           }
         }
-        // 2 n-tiles: dequant both B fragments first, THEN scale, THEN
-        // issue all MFMAs. The shuffle results land in 16 DISTINCT
-        // registers (cval) so the 16 ds_bpermutes issue as one batch and
-        // pipeline in the DS unit — when the destination doubles as the
-        // address register (what minimal-register codegen produces), the
-        // chain waits ~50-60 cycles per PAIR instead.
-        bf16v8 bfr[2];
-        if constexpr (DBG == 1) {
-          #pragma unroll
-          for (int nt = 0; nt < 2; ++nt) {
-            const uint32_t wbits = wb[cur][ks * 2 + nt];
-            #pragma unroll
-            for (int j = 0; j < 8; ++j)
-              bfr[nt][j] = (__bf16)(am[cur][nt] + (float)(wbits & 1));
-          }
-        } else {
-          // register LUT via raw-address ds_bpermute: the byte address
-          // only uses bits [7:2], so shifting the nibble to bits [5:2]
-          // needs ONE shift per element (no mask — bits [7:6] carry the
-          // next nibble's low bits and merely select one of the four
-          // 16-lane table replicas). No LDS traffic, no divergent-LDS
-          // hazards.
-          float cval[2][8];
-          #pragma unroll
-          for (int nt = 0; nt < 2; ++nt) {
-            const uint32_t wbits = wb[cur][ks * 2 + nt];
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const int addr = (j == 0) ? (int)(wbits << 2)
-                                        : (int)(wbits >> (4 * j - 2));
-              cval[nt][j] = __builtin_bit_cast(
-                  float, __builtin_amdgcn_ds_bpermute(addr, lut_bits));
-            }
-          }
-          #pragma unroll
-          for (int nt = 0; nt < 2; ++nt) {
-            const float amv = am[cur][nt];
-            #pragma unroll
-            for (int j = 0; j < 8; ++j)
-              bfr[nt][j] = (__bf16)(cval[nt][j] * amv);
-          }
-        }
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
           #pragma unroll
           for (int mt = 0; mt < MT; ++mt)
             acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag[mt], bfr[nt], acc[mt][nt], 0, 0, 0);
+                afrag[mt], bfr[kc & 1][ks * 2 + nt], acc[mt][nt], 0, 0, 0);
+      }
+
+      // ---- B) dequant the NEXT chunk into the spare fragment buffer
+      if (kb + 1 < kb_end) {
+        DEQUANT_CHUNK((kc + 1) & 3, (kc + 1) & 1);
       }
     }
   }
   #undef LOAD_WCHUNK
+  #undef DEQUANT_CHUNK
 
   // ---- LoRA epilogue: one extra MFMA k-step per rank-32 block ----
   // (slice 0 only under split-K; slices contribute additively)
