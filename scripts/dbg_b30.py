@@ -1,0 +1,44 @@
+#!/usr/bin/env python3
+"""Diagnose the batch-30 update skip."""
+import os, sys, time
+sys.path.insert(0, ".")
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+os.environ.setdefault("MASTER_PORT", "29741")
+os.environ.setdefault("RANK", "0")
+os.environ.setdefault("WORLD_SIZE", "1")
+import numpy as np
+from distrl_llm_amd.parallel.worker import build_worker
+from distrl_llm_amd.rl.data import ListDataset, process_dataset, r1_preprompt, synthetic_math_dataset
+from distrl_llm_amd.rl.rewards import reward_function as rf
+import zlib
+
+def reward_function(completions, solutions):
+    r = rf(completions, solutions)
+    jit = np.array([(zlib.crc32(c[-256:].encode()) % 1000) / 1000.0 * 0.1 for c in completions])
+    r[:, 0] += jit
+    return r
+
+B = int(os.environ.get("DBG_B", "30"))
+rows = process_dataset(None, synthetic_math_dataset(B * 2, seed=17), r1_preprompt)
+config = {
+    "run_name": "dbg", "project_name": "dbg", "lora_save_path": "/tmp/dbg_adapter",
+    "lr": 2e-5, "max_new_tokens": 64, "max_prompt_tokens": 350,
+    "num_candidates": 16, "episodes": 1, "batch_size": B, "train_batch_size": 16,
+    "temperature": 1.2, "save_every": 10**9, "eval_every": 0,
+    "model": "unsloth/Qwen2.5-7B-Instruct-bnb-4bit", "dataset": "synthetic",
+    "number_of_actors": 0, "number_of_learners": 1, "learner": "grpo",
+    "use_vllm": True, "max_lora_rank": 32, "topk": 16,
+    "learner_chunk_size": B, "actor_gpu_usage": 0.5, "learner_gpu_usage": 0.2,
+    "lora_alpha": 16, "lora_dropout": 0.0, "seed": 3407, "use_8bit_adam": True,
+}
+train = ListDataset(rows, seed=17)
+tr = build_worker(0, 1, config, train_dataset=train, test_dataset=ListDataset(rows[:2]),
+                  reward_function=reward_function)
+batch = next(iter(train.iter(batch_size=B)))
+stats = tr.rl_round(batch)
+for k in ("loss", "num_samples", "mean_format_reward", "timing/update_duration",
+          "timing/generation_duration", "mean_token_length"):
+    print(k, stats.get(k))
+# candidate diversity
+cands, _ = None, None
+print("sample answer repr:", repr(stats.get("sample_answer", ""))[:120])
