@@ -22,7 +22,7 @@ B = int(os.environ.get("DBG_B", "30"))
 rows = process_dataset(None, synthetic_math_dataset(B * 2, seed=17), r1_preprompt)
 config = {
     "run_name": "dbg", "project_name": "dbg", "lora_save_path": "/tmp/dbg_adapter",
-    "lr": 2e-5, "max_new_tokens": 64, "max_prompt_tokens": 350,
+    "lr": 2e-5, "max_new_tokens": int(os.environ.get("DBG_MAXNEW", "64")), "max_prompt_tokens": 350,
     "num_candidates": 16, "episodes": 1, "batch_size": B, "train_batch_size": 16,
     "temperature": 1.2, "save_every": 10**9, "eval_every": 0,
     "model": "unsloth/Qwen2.5-7B-Instruct-bnb-4bit", "dataset": "synthetic",
@@ -35,6 +35,17 @@ train = ListDataset(rows, seed=17)
 tr = build_worker(0, 1, config, train_dataset=train, test_dataset=ListDataset(rows[:2]),
                   reward_function=reward_function)
 batch = next(iter(train.iter(batch_size=B)))
+# instrument: capture the generate output before the update
+sp_dict = dict(tr.sampling_params.__dict__)
+cands, gen_dur = tr._cmd("generate", (batch, sp_dict))
+n_empty = sum(1 for c in cands for grp in c["answers"] for a in grp if len(a) == 0)
+tls = [t for c in cands for grp in c["token_lengths"] for t in grp]
+div = []
+for c in cands:
+    for grp in c["answers"]:
+        div.append(len(set(grp)))
+print("gen_dur", round(gen_dur, 1), "n_answers", len(tls), "n_empty", n_empty,
+      "tok_len min/max", min(tls), max(tls), "distinct-per-group", sorted(div)[:5], "...", sorted(div)[-3:])
 stats = tr.rl_round(batch)
 for k in ("loss", "num_samples", "mean_format_reward", "timing/update_duration",
           "timing/generation_duration", "mean_token_length"):
