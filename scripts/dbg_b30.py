@@ -46,6 +46,19 @@ for c in cands:
         div.append(len(set(grp)))
 print("gen_dur", round(gen_dur, 1), "n_answers", len(tls), "n_empty", n_empty,
       "tok_len min/max", min(tls), max(tls), "distinct-per-group", sorted(div)[:5], "...", sorted(div)[-3:])
+# manual pipeline with instrumentation
+from distrl_llm_amd.rl.advantage import process_candidates, merge_candidates, even_chunk_sizes
+rd = tr._compute_rewards(cands)
+cands2, st2 = process_candidates(cands, tr.learner_type, tr.topk)
+probs, ans, rews = merge_candidates(cands2)
+import numpy as _np
+ra = _np.array(rews)
+print("merged", len(probs), "rewards: nonzero", int((ra != 0).sum()),
+      "std", float(ra.std()), "first8", _np.round(ra[:8], 3).tolist())
+t0 = time.time()
+loss = tr.learner.accumulate_gradients(probs, ans, rews)
+tr.learner.step()
+print("direct accumulate:", round(time.time() - t0, 2), "s, loss", loss)
 stats = tr.rl_round(batch)
 for k in ("loss", "num_samples", "mean_format_reward", "timing/update_duration",
           "timing/generation_duration", "mean_token_length"):
