@@ -32,8 +32,11 @@ config = {
     "lora_alpha": 16, "lora_dropout": 0.0, "seed": 3407, "use_8bit_adam": True,
 }
 train = ListDataset(rows, seed=17)
+ov = {}
+if os.environ.get("DBG_POOL"):
+    ov = {"num_kv_blocks": int(os.environ["DBG_POOL"])}
 tr = build_worker(0, 1, config, train_dataset=train, test_dataset=ListDataset(rows[:2]),
-                  reward_function=reward_function)
+                  reward_function=reward_function, engine_overrides=ov)
 batch = next(iter(train.iter(batch_size=B)))
 # instrument: capture the generate output before the update
 sp_dict = dict(tr.sampling_params.__dict__)
@@ -46,6 +49,10 @@ for c in cands:
         div.append(len(set(grp)))
 print("gen_dur", round(gen_dur, 1), "n_answers", len(tls), "n_empty", n_empty,
       "tok_len min/max", min(tls), max(tls), "distinct-per-group", sorted(div)[:5], "...", sorted(div)[-3:])
+# second generation (cache-reuse path) diversity
+cands_b, gen_b = tr._cmd("generate", (batch, sp_dict))
+div_b = [len(set(grp)) for c in cands_b for grp in c["answers"]]
+print("gen2", round(gen_b, 1), "distinct-per-group", sorted(div_b)[:6], "...", sorted(div_b)[-3:])
 # manual pipeline with instrumentation
 from distrl_llm_amd.rl.advantage import process_candidates, merge_candidates, even_chunk_sizes
 rd = tr._compute_rewards(cands)
