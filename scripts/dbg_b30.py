@@ -53,23 +53,26 @@ print("gen_dur", round(gen_dur, 1), "n_answers", len(tls), "n_empty", n_empty,
 cands_b, gen_b = tr._cmd("generate", (batch, sp_dict))
 div_b = [len(set(grp)) for c in cands_b for grp in c["answers"]]
 print("gen2", round(gen_b, 1), "distinct-per-group", sorted(div_b)[:6], "...", sorted(div_b)[-3:])
-# manual pipeline with instrumentation
-from distrl_llm_amd.rl.advantage import process_candidates, merge_candidates, even_chunk_sizes
-rd = tr._compute_rewards(cands)
-cands2, st2 = process_candidates(cands, tr.learner_type, tr.topk)
-probs, ans, rews = merge_candidates(cands2)
+# analyze CALL-2 candidates (the cache-reuse path that fails)
+from distrl_llm_amd.rl.advantage import process_candidates, merge_candidates
 import numpy as _np
+rd = tr._compute_rewards(cands_b)
+raw_stds = []
+for c in cands_b:
+    for g in c["rewards"]:
+        raw_stds.append(float(_np.asarray(g).sum(axis=1).std()))
+print("call2 raw reward group stds: min", min(raw_stds), "max", max(raw_stds),
+      "zeros", sum(1 for x in raw_stds if x == 0.0))
+g0 = cands_b[0]["answers"][0]
+print("g0 ans lens", [len(a) for a in g0[:6]])
+print("g0 a0 tail", repr(g0[0][-48:]))
+print("g0 a1 tail", repr(g0[1][-48:]))
+cands2, st2 = process_candidates(cands_b, tr.learner_type, tr.topk)
+probs, ans, rews = merge_candidates(cands2)
 ra = _np.array(rews)
-print("merged", len(probs), "rewards: nonzero", int((ra != 0).sum()),
-      "std", float(ra.std()), "first8", _np.round(ra[:8], 3).tolist())
-t0 = time.time()
-loss = tr.learner.accumulate_gradients(probs, ans, rews)
-tr.learner.step()
-print("direct accumulate:", round(time.time() - t0, 2), "s, loss", loss)
-stats = tr.rl_round(batch)
-for k in ("loss", "num_samples", "mean_format_reward", "timing/update_duration",
-          "timing/generation_duration", "mean_token_length"):
-    print(k, stats.get(k))
+print("call2 advantages: nonzero", int((ra != 0).sum()), "std", float(ra.std()))
+stats = {"skip": 1}
+
 # candidate diversity
 cands, _ = None, None
 print("sample answer repr:", repr(stats.get("sample_answer", ""))[:120])
