@@ -307,13 +307,20 @@ class Engine:
                 rp = getattr(lws[0], f"{site}_r")
                 site_off[site] = (off, rp)
                 off += rp
-            key = (N_batch, off)
-            if getattr(self, "_u_pool_key", None) != key:
+            # the pool MUST keep one stable storage for the engine's
+            # lifetime: decode sessions capture hipGraphs that reference
+            # it, and waves of different sizes share those graphs across
+            # generation rounds. A per-wave reallocation left earlier
+            # waves' cached graphs pointing at freed memory (garbage u ->
+            # NaN logits -> the sampler degenerates to token 0; surfaced
+            # at batch 30 x 16 where pool pressure forces multiple wave
+            # sizes per round). Allocate once at max_num_seqs and slice.
+            if getattr(self, "_u_pool_off", None) != off:
                 self._u_pool = torch.zeros(
-                    len(lws), N_batch, off, dtype=torch.float32,
-                    device=input_ids.device)
-                self._u_pool_key = key
-            u_pool = self._u_pool
+                    len(lws), self.cfg.max_num_seqs, off,
+                    dtype=torch.float32, device=input_ids.device)
+                self._u_pool_off = off
+            u_pool = self._u_pool[:, :N_batch, :]
             u_pool.zero_()
 
         def proj(x, lw, li, site, N_out, K_in):
