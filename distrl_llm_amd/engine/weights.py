@@ -145,10 +145,27 @@ class FusedWeights:
                     m.lora_B.detach() * m.scale)
                 n0 += m.out_features
                 r0 += m.r
-            setattr(lw, f"{name}_afrag",
-                    prepack_bf16_fragments(A.to(torch.bfloat16)))
-            setattr(lw, f"{name}_bfrag",
-                    prepack_bf16_fragments(B.to(torch.bfloat16)))
+            # IN-PLACE refresh: decode sessions capture hipGraphs that
+            # reference these tensors, and the graph cache reuses those
+            # graphs across generation rounds — a fresh allocation here
+            # would leave every cached graph reading freed memory (at
+            # single-size waves the allocator happens to hand the same
+            # block back, which HID this; multi-wave rounds surfaced it
+            # as NaN logits -> the sampler degenerating to token 0)
+            af = prepack_bf16_fragments(A.to(torch.bfloat16))
+            bf = prepack_bf16_fragments(B.to(torch.bfloat16))
+            try:
+                old_af = getattr(lw, f"{name}_afrag")
+                old_bf = getattr(lw, f"{name}_bfrag")
+            except AttributeError:
+                old_af = old_bf = None
+            if (old_af is not None and old_af.shape == af.shape
+                    and old_bf.shape == bf.shape):
+                old_af.copy_(af)
+                old_bf.copy_(bf)
+            else:
+                setattr(lw, f"{name}_afrag", af)
+                setattr(lw, f"{name}_bfrag", bf)
             setattr(lw, f"{name}_r", r_pad)
 
     @torch.no_grad()
