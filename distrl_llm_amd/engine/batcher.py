@@ -102,8 +102,11 @@ class DynamicBatcher:
         # belongs to one caller)
         nonce = (id(req) if sp.seed is not None or req.stream_q is not None
                  else None)
-        return (sp.max_tokens, sp.temperature, sp.top_p, sp.top_k, sp.n,
-                req.eos, nonce)
+        # max_tokens is NOT part of the key: requests with different
+        # output caps batch together via the engine's per-sequence token
+        # limits (in-wave retirement frees a short request's lanes while
+        # longer ones keep decoding)
+        return (sp.temperature, sp.top_p, sp.top_k, sp.n, req.eos, nonce)
 
     def _loop(self):
         shutdown = False
@@ -149,9 +152,18 @@ class DynamicBatcher:
             cb = lambda pi, ci, toks: stream_q.put(("tok", pi, ci, toks))
         try:
             self.calls += 1
-            outs = self.engine.generate(merged, reqs[0].sp,
+            sp = reqs[0].sp
+            limits = None
+            caps = [r.sp.max_tokens for r in reqs]
+            if len(set(caps)) > 1:
+                import dataclasses
+                sp = dataclasses.replace(sp, max_tokens=max(caps))
+                limits = [[r.sp.max_tokens] * sp.n
+                          for r in reqs for _ in r.prompts]
+            outs = self.engine.generate(merged, sp,
                                         eos_token_id=reqs[0].eos,
-                                        stream_cb=cb)
+                                        stream_cb=cb,
+                                        token_limits=limits)
             off = 0
             for r in reqs:
                 r.result = outs[off:off + len(r.prompts)]
