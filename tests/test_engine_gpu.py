@@ -218,3 +218,38 @@ def test_nf4_freed_base_learner_parity():
         err = (a - b).abs().max()
         ref = a.abs().max().clamp_min(1e-6)
         assert err <= 0.05 * ref + 1e-5, (err, ref)
+
+
+def test_graph_cache_stable_across_weight_refresh():
+    """Cached hipGraphs must survive weight syncs: every tensor a captured
+    graph references (adapter fragment packs, the LoRA-u pool) must keep
+    ONE storage for the cache's lifetime. Regression test for the
+    multi-wave batch-30 silent-NaN bug (profiles/r02_summary.md): run a
+    generate, perturb the adapter (forces a refresh), run again with the
+    cache, and compare against a FRESH engine with identical weights."""
+    import copy
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    spec = get_spec("small-qwen2")
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    model = CausalLM(spec, lora_r=8, lora_alpha=16, dtype=torch.bfloat16,
+                     device=dev)
+    model.random_init(seed=21)
+    cfg = dict(max_seq_length=256, kv_block_size=16, num_kv_blocks=512,
+               max_num_seqs=64)
+    prompts = [[5, 9, 2, 7], list(range(30, 60))]
+    sp = SamplingParams(max_tokens=10, temperature=0.0, n=2)
+
+    eng = Engine(model, EngineConfig(**cfg), device=dev, seed=3)
+    eng.generate(prompts, sp, eos_token_id=None)    # captures + caches
+    with torch.no_grad():                           # a "weight sync"
+        for m in model.modules():
+            if hasattr(m, "lora_B") and m.lora_B is not None:
+                m.lora_B.add_(torch.randn_like(m.lora_B) * 0.05)
+    out_cached = eng.generate(prompts, sp, eos_token_id=None)  # reuses graphs
+
+    fresh = Engine(model, EngineConfig(**cfg), device=dev, seed=3)
+    out_fresh = fresh.generate(prompts, sp, eos_token_id=None)
+    assert out_cached == out_fresh, (out_cached, out_fresh)
