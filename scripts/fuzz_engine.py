@@ -67,13 +67,19 @@ for trial in range(args.trials):
     eos = naive(prompts[0][:min(len(prompts[0]), msl - 1)], 1)[0] \
         if rng.random() < 0.5 else None
     sp = SamplingParams(max_tokens=mt, temperature=0.0, n=n)
+    # sometimes: per-candidate output caps (token_limits -> in-wave
+    # retirement paths); naive() then compares against each cap
+    limits = None
+    if rng.random() < 0.4:
+        limits = [[rng.randint(1, mt) for _ in range(n)] for _ in prompts]
     streamed = {}
     cb = (lambda pi, ci, toks: streamed.setdefault((pi, ci), []).extend(toks)) \
         if rng.random() < 0.5 else None
     if trial < args.start:
         continue
     try:
-        res = eng.generate(prompts, sp, eos_token_id=eos, stream_cb=cb)
+        res = eng.generate(prompts, sp, eos_token_id=eos, stream_cb=cb,
+                           token_limits=limits)
     except MemoryError:
         continue
     ok = eng.pool.allocator.num_free == nb
@@ -81,8 +87,9 @@ for trial in range(args.trials):
         fails += 1; print("LEAK", trial); continue
     for pi, (p, r) in enumerate(zip(prompts, res)):
         L = min(len(p), msl - 1)
-        exp = naive(p[:L], min(mt, msl - L), eos)
         for ci, ids in enumerate(r):
+            cap = limits[pi][ci] if limits is not None else mt
+            exp = naive(p[:L], min(cap, msl - L), eos)
             if ids != exp:
                 fails += 1
                 print("MISMATCH", trial, force, bs, nb, mt, n, msl, eos, p)
