@@ -143,7 +143,8 @@ class Engine:
         causal prefill HIP kernel (q-row tiles, lane-per-kv scoring —
         ops/csrc/paged_attn.hip); batched-SDPA fallback only for prompts
         too long for its LDS score tiles. CPU: reference varlen."""
-        if q.is_cuda and max(lens) <= 2200:
+        if (q.is_cuda and max(lens) <= 2200
+                and os.environ.get("DISTRL_PREFILL_FLASH") != "1"):
             from ..ops.build import get_extension
             ext = get_extension()
             if ext is None:
