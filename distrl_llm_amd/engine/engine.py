@@ -139,12 +139,14 @@ class Engine:
         return self.model.logits(x[last_idx])
 
     def _prefill_attention(self, q, k, v, lens: List[int]) -> torch.Tensor:
-        """Causal attention over concatenated prompts. GPU: the varlen
-        causal prefill HIP kernel (q-row tiles, lane-per-kv scoring —
-        ops/csrc/paged_attn.hip); batched-SDPA fallback only for prompts
-        too long for its LDS score tiles. CPU: reference varlen."""
+        """Causal attention over concatenated prompts. GPU default: the
+        first-party MFMA flash kernel (ops/csrc/attention.hip) on a
+        per-prompt padded batch — measured 10x the varlen scalar-scoring
+        kernel (1.8 vs 19.6 s for 160 ~400-token prompts) and no length
+        cap. DISTRL_PREFILL_VARLEN=1 selects the legacy varlen kernel
+        (<=2200 tokens) for comparison. CPU: reference varlen."""
         if (q.is_cuda and max(lens) <= 2200
-                and os.environ.get("DISTRL_PREFILL_FLASH") != "1"):
+                and os.environ.get("DISTRL_PREFILL_VARLEN") == "1"):
             from ..ops.build import get_extension
             ext = get_extension()
             if ext is None:
