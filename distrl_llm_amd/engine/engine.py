@@ -427,6 +427,12 @@ class Engine:
         stream_cb: optional ``f(prompt_index, cand_index, new_token_ids)``
         called as tokens become known — per token on the eager path, per
         decode chunk on the session path (serving SSE streaming).
+
+        token_limits: optional per-prompt lists of per-candidate output
+        caps (each <= sp.max_tokens); finished lanes retire in-wave and
+        free their decode slots (the serving batcher uses this to merge
+        requests with different max_tokens; the EOS-realistic bench mode
+        draws them from an exponential via sp.geom_len_mean).
         """
         was_training = self.model.training
         self.model.eval()
