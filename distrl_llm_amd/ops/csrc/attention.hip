@@ -72,17 +72,17 @@ DEV_INLINE bf16v8 relayout_frag(const f32x4& t0, const f32x4& t1) {
   const uint32_t a1_1 = cvt_pk_bf16(t1[2], t1[3]);
   const int s0 = ((g & 1) << 5) + c;        // source lane: group 2*(g&1)
   const int s1 = s0 + 16;                   // group 2*(g&1)+1
-  uint32_t w[4];
+  union { uint32_t w[4]; bf16v8 v; } out;
   const uint32_t p00 = __shfl((int)a0_0, s0, 64), p01 = __shfl((int)a1_0, s0, 64);
   const uint32_t p02 = __shfl((int)a0_0, s1, 64), p03 = __shfl((int)a1_0, s1, 64);
   const uint32_t p10 = __shfl((int)a0_1, s0, 64), p11 = __shfl((int)a1_1, s0, 64);
   const uint32_t p12 = __shfl((int)a0_1, s1, 64), p13 = __shfl((int)a1_1, s1, 64);
   const bool hi = g >= 2;                   // k 16..31 -> T1
-  w[0] = hi ? p10 : p00;
-  w[1] = hi ? p11 : p01;
-  w[2] = hi ? p12 : p02;
-  w[3] = hi ? p13 : p03;
-  return *reinterpret_cast<const bf16v8*>(w);
+  out.w[0] = hi ? p10 : p00;
+  out.w[1] = hi ? p11 : p01;
+  out.w[2] = hi ? p12 : p02;
+  out.w[3] = hi ? p13 : p03;
+  return out.v;
 }
 
 // LDS tile staging: [ROWS][COLS] bf16 row-major with an XOR swizzle on

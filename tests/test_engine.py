@@ -188,3 +188,21 @@ def test_derive_num_blocks_cpu_sizing(setup):
     per_block = (2 * spec.num_layers * 8 * spec.num_kv_heads * spec.head_dim
                  * 4)  # fp32 on CPU
     assert e2.pool.num_blocks == max(64 * 1024 * 1024 // per_block, 16)
+
+
+def test_geom_len_mean_draws_deterministic_limits(setup):
+    """sp.geom_len_mean: per-candidate exponential caps drawn from the
+    engine's seeded generator — deterministic per (engine seed, call)."""
+    model, _ = setup
+    cfg = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=256,
+                       max_num_seqs=16)
+    sp = SamplingParams(max_tokens=20, temperature=0.0, n=3,
+                        geom_len_mean=6.0)
+    outs = []
+    for _ in range(2):
+        eng = Engine(model, cfg, device=torch.device("cpu"), seed=9)
+        outs.append(eng.generate([[3, 5, 7], [2, 4]], sp, eos_token_id=None))
+    assert outs[0] == outs[1]
+    lens = [len(ids) for per in outs[0] for ids in per]
+    assert all(1 <= L <= 20 for L in lens)
+    assert len(set(lens)) > 1  # the exponential actually varies the caps
