@@ -55,7 +55,21 @@ class Attention(nn.Module):
         q = self.q_proj(x).view(B, T, s.num_heads, s.head_dim)
         k = self.k_proj(x).view(B, T, s.num_kv_heads, s.head_dim)
         v = self.v_proj(x).view(B, T, s.num_kv_heads, s.head_dim)
-        q, k = R.apply_rope(q, k, cos, sin)
+        import os
+        if (attn_bias is None and q.is_cuda and q.dtype == torch.bfloat16
+                and os.environ.get("DISTRL_ROPE_KERNEL") == "1"):
+            # first-party RoPE fwd/bwd (fused HIP kernel; backward =
+            # negated-frequency rotation — ops/functional._RopeTrainFn).
+            # Opt-in: formula CPU-proven + GPU-tested; the default stays
+            # the torch path pending an end-to-end perf pass.
+            from ..ops import functional as OF
+            pos = torch.arange(T, device=x.device).repeat(B)
+            inv_freq = 1.0 / (s.rope_theta ** (
+                torch.arange(0, s.head_dim, 2, device=x.device,
+                             dtype=torch.float32) / s.head_dim))
+            q, k = OF.rope_training(q, k, pos, inv_freq)
+        else:
+            q, k = R.apply_rope(q, k, cos, sin)
         if (attn_bias is None and q.is_cuda and q.dtype == torch.bfloat16
                 and s.head_dim in (64, 128)):
             # first-party CDNA4 flash attention (ops/csrc/attention.hip):

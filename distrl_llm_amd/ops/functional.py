@@ -179,6 +179,46 @@ def sample_tokens(logits: torch.Tensor, temperature: float, top_p: float,
     return R.sample_tokens(logits, temperature, top_p, top_k, generator=generator)
 
 
+# ------------------------------------------------- training RoPE fwd/bwd
+
+class _RopeTrainFn(torch.autograd.Function):
+    """First-party RoPE for the TRAINING path (SURVEY.md §2.4-B "RoPE
+    fwd/bwd" row): forward rotates q/k with the fused HIP kernel
+    (ops/csrc/rope.hip, already numerics-tested on the inference path);
+    backward is the SAME kernel with NEGATED frequencies — the gradient
+    of a rotation is the transposed rotation, and R(-theta) = R(theta)^T
+    exactly (sin is odd, cos even). Formula verified on CPU against
+    torch autograd (test_autograd_formulas), kernel fwd/bwd verified on
+    GPU (test_ops_gpu::test_rope_training)."""
+
+    @staticmethod
+    def forward(ctx, q, k, positions, inv_freq):
+        ext = _require_ext("rope")
+        B, T, H, D = q.shape
+        KV = k.shape[2]
+        qf = q.reshape(B * T, H, D).contiguous().clone()
+        kf = k.reshape(B * T, KV, D).contiguous().clone()
+        ext.rope_inplace(qf, kf, positions.to(torch.int32), inv_freq)
+        ctx.save_for_backward(positions, inv_freq)
+        ctx.shapes = (B, T, H, KV, D)
+        return qf.view(B, T, H, D), kf.view(B, T, KV, D)
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        positions, inv_freq = ctx.saved_tensors
+        B, T, H, KV, D = ctx.shapes
+        ext = _require_ext("rope")
+        dqf = dq.reshape(B * T, H, D).contiguous().clone()
+        dkf = dk.reshape(B * T, KV, D).contiguous().clone()
+        ext.rope_inplace(dqf, dkf, positions.to(torch.int32), -inv_freq)
+        return (dqf.view(B, T, H, D), dkf.view(B, T, KV, D), None, None)
+
+
+def rope_training(q, k, positions, inv_freq):
+    """q (B,T,H,D), k (B,T,KV,D) bf16 on GPU; positions flat (B*T,)."""
+    return _RopeTrainFn.apply(q, k, positions, inv_freq)
+
+
 # ----------------------------------------------------- nf4 base linear
 
 class _NF4LinearFn(torch.autograd.Function):

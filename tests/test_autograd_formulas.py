@@ -101,3 +101,35 @@ def test_logprob_loss_zero_mask_row_safe():
     assert torch.isfinite(loss)
     assert torch.isfinite(logits.grad).all()
     assert logits.grad[1].abs().sum() == 0
+
+
+def test_rope_backward_is_negated_frequency_rotation():
+    """The training-RoPE backward formula (ops/functional._RopeTrainFn):
+    grad of y = R(theta) x is R(-theta) dy. Verified against torch
+    autograd through the reference rope on CPU fp64."""
+    import torch
+    from distrl_llm_amd.ops import reference as R
+    torch.manual_seed(0)
+    B, T, H, KV, D = 2, 5, 3, 1, 8
+    theta = 1e4
+    pos = torch.arange(T).repeat(B)
+    inv_freq = 1.0 / (theta ** (torch.arange(0, D, 2).double() / D))
+    q = torch.randn(B * T, H, D, dtype=torch.float64, requires_grad=True)
+    k = torch.randn(B * T, KV, D, dtype=torch.float64, requires_grad=True)
+    cos, sin = R.rope_cos_sin(pos, D, theta, dtype=torch.float64)
+    qr, kr = R.apply_rope(q.view(B, T, H, D), k.view(B, T, KV, D),
+                          cos.view(B, T, -1), sin.view(B, T, -1))
+    dq = torch.randn_like(qr)
+    dk = torch.randn_like(kr)
+    (qr * dq).sum().backward(retain_graph=True)
+    gq_auto = q.grad.clone()
+    q.grad = None
+    (kr * dk).sum().backward()
+    gk_auto = k.grad.clone()
+
+    # formula: rotate the output grads with NEGATED frequencies
+    ncos, nsin = R.rope_cos_sin(pos, D, theta, dtype=torch.float64)
+    gq_f, gk_f = R.apply_rope(dq, dk, ncos.view(B, T, -1),
+                              -nsin.view(B, T, -1))
+    assert torch.allclose(gq_f.reshape(B * T, H, D), gq_auto, atol=1e-10)
+    assert torch.allclose(gk_f.reshape(B * T, KV, D), gk_auto, atol=1e-10)

@@ -422,3 +422,41 @@ def test_flash_attention_fwd_bwd(ext, B, Hq, Hkv, T, D):
         (kg.grad.float() - kr.grad).abs().max()
     assert torch.allclose(vg.grad.float(), vr.grad, atol=8e-2, rtol=8e-2), \
         (vg.grad.float() - vr.grad).abs().max()
+
+
+def test_rope_training_fwd_bwd(ext):
+    """Training RoPE fwd/bwd (ops/functional.rope_training): the fused
+    HIP kernel forward + negated-frequency backward vs fp32 torch
+    autograd (SURVEY.md §2.4-B RoPE fwd/bwd row; backward formula proven
+    in test_autograd_formulas)."""
+    from distrl_llm_amd.ops import functional as OF
+    from distrl_llm_amd.ops import reference as R
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    B, T, H, KV, D = 2, 33, 4, 2, 64
+    theta = 1e4
+    pos = torch.arange(T, device=dev).repeat(B)
+    inv_freq = (1.0 / (theta ** (torch.arange(0, D, 2, device=dev,
+                                              dtype=torch.float32) / D)))
+    q = (torch.randn(B, T, H, D, device=dev) * 0.5).to(torch.bfloat16)
+    k = (torch.randn(B, T, KV, D, device=dev) * 0.5).to(torch.bfloat16)
+    dq = torch.randn_like(q)
+    dk = torch.randn_like(k)
+
+    qg = q.clone().requires_grad_(True)
+    kg = k.clone().requires_grad_(True)
+    qo, ko = OF.rope_training(qg, kg, pos, inv_freq)
+    (qo * dq.float()).sum().backward(retain_graph=True)
+    (ko * dk.float()).sum().backward()
+
+    qr = q.float().requires_grad_(True)
+    kr = k.float().requires_grad_(True)
+    cos, sin = R.rope_cos_sin(pos.view(B, T), D, theta, dtype=torch.float32)
+    qor, kor = R.apply_rope(qr, kr, cos, sin)
+    (qor * dq.float()).sum().backward(retain_graph=True)
+    (kor * dk.float()).sum().backward()
+
+    assert torch.allclose(qo.float(), qor, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(ko.float(), kor, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(qg.grad.float(), qr.grad, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(kg.grad.float(), kr.grad, atol=2e-2, rtol=2e-2)
