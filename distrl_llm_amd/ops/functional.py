@@ -207,11 +207,19 @@ class _RopeTrainFn(torch.autograd.Function):
     def backward(ctx, dq, dk):
         positions, inv_freq = ctx.saved_tensors
         B, T, H, KV, D = ctx.shapes
-        ext = _require_ext("rope")
-        dqf = dq.reshape(B * T, H, D).contiguous().clone()
-        dkf = dk.reshape(B * T, KV, D).contiguous().clone()
-        ext.rope_inplace(dqf, dkf, positions.to(torch.int32), -inv_freq)
-        return (dqf.view(B, T, H, D), dkf.view(B, T, KV, D), None, None)
+        if dq.dtype == torch.bfloat16 and dk.dtype == torch.bfloat16:
+            ext = _require_ext("rope")
+            dqf = dq.reshape(B * T, H, D).contiguous().clone()
+            dkf = dk.reshape(B * T, KV, D).contiguous().clone()
+            ext.rope_inplace(dqf, dkf, positions.to(torch.int32), -inv_freq)
+            return (dqf.view(B, T, H, D), dkf.view(B, T, KV, D), None, None)
+        # non-bf16 upstream grads (e.g. fp32 test harnesses): exact torch
+        # rotation with negated sin — the kernel is bf16-only
+        freqs = positions.to(torch.float32).unsqueeze(-1) * inv_freq
+        cos = freqs.cos().view(B, T, -1)
+        sin = freqs.sin().view(B, T, -1)
+        gq, gk = R.apply_rope(dq, dk, cos, -sin)
+        return gq, gk, None, None
 
 
 def rope_training(q, k, positions, inv_freq):
