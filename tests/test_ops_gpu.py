@@ -446,8 +446,9 @@ def test_rope_training_fwd_bwd(ext):
     qg = q.clone().requires_grad_(True)
     kg = k.clone().requires_grad_(True)
     qo, ko = OF.rope_training(qg, kg, pos, inv_freq)
-    (qo * dq.float()).sum().backward(retain_graph=True)
-    (ko * dk.float()).sum().backward()
+    # bf16 upstream grads so backward exercises the KERNEL path (fp32
+    # grads take the exact-torch fallback)
+    torch.autograd.backward([qo, ko], [dq, dk])
 
     qr = q.float().requires_grad_(True)
     kr = k.float().requires_grad_(True)
