@@ -12,6 +12,7 @@ weight tensors from ``engine/``.
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
@@ -55,7 +56,6 @@ class Attention(nn.Module):
         q = self.q_proj(x).view(B, T, s.num_heads, s.head_dim)
         k = self.k_proj(x).view(B, T, s.num_kv_heads, s.head_dim)
         v = self.v_proj(x).view(B, T, s.num_kv_heads, s.head_dim)
-        import os
         if (attn_bias is None and q.is_cuda and q.dtype == torch.bfloat16
                 and os.environ.get("DISTRL_ROPE_KERNEL") == "1"):
             # first-party RoPE fwd/bwd (fused HIP kernel; backward =
