@@ -323,6 +323,12 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kcache,
 // north star); replaces the torch-SDPA library call in the engine.
 namespace {
 
+// the legacy prefill kernel keeps the original 32-token V tile: the
+// decode kernel's VTILE=64 retune doubled this kernel's LDS and cost it
+// ~10x in occupancy (one reason the MFMA flash kernel is now the
+// prefill default)
+constexpr int PF_VTILE = 32;
+
 template <int D>
 __global__ __launch_bounds__(256)
 void prefill_attn_kernel(const __hip_bfloat16* __restrict__ q,   // (T,H,D)
@@ -349,7 +355,7 @@ void prefill_attn_kernel(const __hip_bfloat16* __restrict__ q,   // (T,H,D)
   float* denom = q_lds + QT * D;                              // QT
   float* scores = denom + QT;                                 // QT*Lpad
   __hip_bfloat16* v_lds = reinterpret_cast<__hip_bfloat16*>(
-      scores + (size_t)QT * Lpad);                            // VTILE*D
+      scores + (size_t)QT * Lpad);                            // PF_VTILE*D
 
   for (int i = tid; i < QT * D; i += blockDim.x) {
     const int r = i / D;
@@ -413,8 +419,8 @@ void prefill_attn_kernel(const __hip_bfloat16* __restrict__ q,   // (T,H,D)
   float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   const int u = tid;
   const int ur = u / DV, ud = u % DV;
-  for (int base = 0; base < Lmax; base += VTILE) {
-    const int tl = min(VTILE, Lmax - base);
+  for (int base = 0; base < Lmax; base += PF_VTILE) {
+    const int tl = min(PF_VTILE, Lmax - base);
     for (int i = tid; i < tl * DV; i += blockDim.x) {
       reinterpret_cast<bf16x8*>(v_lds)[i] =
           *reinterpret_cast<const bf16x8*>(
@@ -461,7 +467,7 @@ torch::Tensor prefill_attention(torch::Tensor q, torch::Tensor k,
   if (n_tiles == 0) return out;
   const int Lpad = (int)max_len + 4;
   size_t smem = 16 * D * 4 + 16 * 4 + (size_t)16 * Lpad * 4
-                + (size_t)VTILE * D * 2;
+                + (size_t)PF_VTILE * D * 2;
   TORCH_CHECK(smem <= 160 * 1024, "prompt too long for prefill kernel: ",
               max_len);
   dim3 grid(n_tiles, H), block(256);
