@@ -461,6 +461,15 @@ class Engine:
         # caps (each <= sp.max_tokens) — the EOS-realistic bench mode and
         # per-request limits use these; admission still reserves the
         # worst case from sp.max_tokens.
+        if token_limits is not None:
+            if len(token_limits) != len(prompts):
+                raise ValueError("token_limits must have one list per prompt")
+            for per in token_limits:
+                if len(per) != sp.n or any(
+                        not (1 <= int(v) <= sp.max_tokens) for v in per):
+                    raise ValueError(
+                        "each token_limits entry needs sp.n caps in "
+                        f"[1, {sp.max_tokens}]")
         if sp.n > self.cfg.max_num_seqs:
             # the n-candidate fan-out of one prompt is admitted atomically,
             # so it can never fit — fail with the real reason instead of
