@@ -283,8 +283,8 @@ class _FlashAttnFn(torch.autograd.Function):
     def backward(ctx, dout):
         q, k, v, o, lse = ctx.saved_tensors
         ext = _require_ext("flash_attention")
-        dq, dk, dv = ext.flash_attn_bwd(dout.contiguous(), q, k, v, o, lse,
-                                        ctx.scale)
+        dq, dk, dv = ext.flash_attn_bwd(dout.to(q.dtype).contiguous(),
+                                        q, k, v, o, lse, ctx.scale)
         return dq, dk, dv, None
 
 
