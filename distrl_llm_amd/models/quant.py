@@ -1,12 +1,12 @@
 """nf4 packing utilities + MFMA fragment prepacking for the gfx950 kernels.
 
-The fused nf4 GEMM consumes weights in B-fragment order, layout v2: the
-lane's 8 nibble-dwords for one 64-deep K chunk of a wave's 4 n-tiles are
-contiguous ([N/64][K/64][64][8] — see ops/csrc/nf4_gemm.hip), so one chunk
-streams as two dwordx4 loads per lane. Because the base quantizer packs
-two nibbles per byte K-contiguously, each fragment dword is exactly 4
-consecutive packed bytes, so prepacking is a pure int32 gather (no bit
-twiddling).
+The fused nf4 GEMM consumes weights in B-fragment order, layout v3: the
+lane's 4 nibble-dwords for one 64-deep K chunk of a wave's 2 n-tiles are
+contiguous ([N/32][K/64][64][4] — see ops/csrc/nf4_gemm.hip), so one chunk
+streams as one dwordx4 load per lane (plus one dwordx2 of absmax).
+Because the base quantizer packs two nibbles per byte K-contiguously,
+each fragment dword is exactly 4 consecutive packed bytes, so prepacking
+is a pure int32 gather (no bit twiddling).
 """
 
 from __future__ import annotations
