@@ -43,8 +43,18 @@ def spec_from_hf_config(path: str, name: Optional[str] = None) -> ModelSpec:
         cfg = json.load(f)
     archs = cfg.get("architectures") or [""]
     arch = archs[0]
-    if not any(a in arch for a in ("Qwen2", "Llama")):
+    if not any(a in arch for a in ("Qwen2", "Llama", "Mistral")):
         raise ValueError(f"unsupported architecture {arch!r} in {path}")
+    sw = cfg.get("sliding_window")
+    if arch.startswith("Mistral") and sw is not None \
+            and sw < cfg.get("max_position_embeddings", 32768):
+        # pre-v0.3 Mistral uses sliding-window attention; this stack
+        # implements plain causal attention (all BASELINE.json families),
+        # so accepting the checkpoint would silently change semantics
+        # beyond `sliding_window` tokens of context
+        raise ValueError(
+            f"windowed Mistral checkpoint (sliding_window={sw}) is not "
+            f"supported; use a v0.3+ checkpoint (sliding_window null)")
     heads = cfg["num_attention_heads"]
     head_dim = cfg.get("head_dim") or cfg["hidden_size"] // heads
     return ModelSpec(
@@ -128,10 +138,15 @@ def save_hf_checkpoint(model, path: str) -> None:
 
 
 def _write_hf_config(s: ModelSpec, path: str) -> None:
+    if s.qkv_bias:
+        arch, mtype = "Qwen2ForCausalLM", "qwen2"
+    elif "mistral" in s.name:
+        arch, mtype = "MistralForCausalLM", "mistral"
+    else:
+        arch, mtype = "LlamaForCausalLM", "llama"
     cfg = {
-        "architectures": ["Qwen2ForCausalLM" if s.qkv_bias
-                          else "LlamaForCausalLM"],
-        "model_type": "qwen2" if s.qkv_bias else "llama",
+        "architectures": [arch],
+        "model_type": mtype,
         "hidden_size": s.hidden_size,
         "intermediate_size": s.intermediate_size,
         "num_hidden_layers": s.num_layers,
@@ -146,6 +161,9 @@ def _write_hf_config(s: ModelSpec, path: str) -> None:
         "max_position_embeddings": s.max_position,
         "torch_dtype": "bfloat16",
     }
+    if mtype == "mistral":
+        # v0.3 semantics: no window (transformers defaults to 4096)
+        cfg["sliding_window"] = None
     with open(os.path.join(path, "config.json"), "w") as f:
         json.dump(cfg, f, indent=2)
 

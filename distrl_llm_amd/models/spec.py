@@ -55,6 +55,11 @@ _REGISTRY = {
                             5e5, 1e-5, False, False),
     "llama-3-70b": ModelSpec("llama-3-70b", 128256, 8192, 28672, 80, 64, 8, 128,
                              5e5, 1e-5, False, False),
+    # Mistral-7B v0.3 (sliding_window null => plain causal attention,
+    # which is the attention this stack implements; pre-v0.3 windowed
+    # checkpoints are rejected by hf_io.spec_from_hf_config)
+    "mistral-7b": ModelSpec("mistral-7b", 32768, 4096, 14336, 32, 32, 8, 128,
+                            1e6, 1e-5, False, False),
     # tiny models for CPU tests / the gloo plumbing config
     "tiny-qwen2": ModelSpec("tiny-qwen2", 2048, 64, 128, 2, 4, 2, 16,
                             1e4, 1e-6, True, True, max_position=512),
@@ -63,6 +68,8 @@ _REGISTRY = {
                              1e5, 1e-6, True, True, max_position=2048),
     "tiny-llama": ModelSpec("tiny-llama", 2048, 64, 128, 2, 4, 2, 16,
                             1e4, 1e-5, False, False, max_position=512),
+    "tiny-mistral": ModelSpec("tiny-mistral", 2048, 64, 128, 2, 4, 2, 16,
+                              1e4, 1e-5, False, False, max_position=512),
 }
 
 
@@ -70,7 +77,7 @@ def get_spec(model_name: str) -> ModelSpec:
     """Resolve a model-name string (e.g. the reference's default
     'unsloth/Qwen2.5-7B-Instruct-bnb-4bit') to an architecture spec."""
     low = model_name.lower()
-    for key in ("tiny-qwen2", "tiny-llama", "small-qwen2"):
+    for key in ("tiny-qwen2", "tiny-llama", "tiny-mistral", "small-qwen2"):
         if key in low:
             return _REGISTRY[key]
     if "qwen2.5-0.5b" in low or "qwen2-0.5b" in low:
@@ -89,6 +96,8 @@ def get_spec(model_name: str) -> ModelSpec:
         return _REGISTRY["llama-3-8b"]
     if "llama-3" in low and "70b" in low:
         return _REGISTRY["llama-3-70b"]
+    if "mistral" in low and "7b" in low:
+        return _REGISTRY["mistral-7b"]
     raise ValueError(f"Unknown model architecture for name: {model_name!r}; "
                      f"known: {sorted(_REGISTRY)}")
 

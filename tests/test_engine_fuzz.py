@@ -134,3 +134,36 @@ def _llama_model():
         assert m.lm_head is not None  # untied head actually exercised
         _LLAMA["m"] = m
     return _LLAMA["m"]
+
+
+@settings(max_examples=8, deadline=None)
+@given(
+    prompts=st.lists(
+        st.lists(st.integers(min_value=1, max_value=500), min_size=1,
+                 max_size=20),
+        min_size=1, max_size=4),
+    max_tokens=st.integers(min_value=1, max_value=6),
+)
+def test_mistral_family_greedy_equals_naive(prompts, max_tokens):
+    """The Mistral (v0.3-style) architecture branch — same stack as
+    Llama but its own spec/name resolution and HF round-trip — through
+    the same engine paths."""
+    if "m" not in _MISTRAL:
+        m = CausalLM(get_spec("tiny-mistral"), lora_r=4, lora_alpha=8,
+                     dtype=torch.float32)
+        m.random_init(seed=78)
+        _MISTRAL["m"] = m
+    m = _MISTRAL["m"]
+    cfg = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=128,
+                       max_num_seqs=32)
+    engine = Engine(m, cfg, device=torch.device("cpu"), seed=0)
+    sp = SamplingParams(max_tokens=max_tokens, temperature=0.0, n=2)
+    results = engine.generate(prompts, sp, eos_token_id=None)
+    for p, res in zip(prompts, results):
+        expected = _naive_greedy(m, p, max_tokens)
+        for ids in res:
+            assert ids == expected
+    assert engine.pool.allocator.num_free == 128
+
+
+_MISTRAL = {}

@@ -226,3 +226,34 @@ def _worker_main(rank, ckpt, port, ok_file):
     assert torch.equal(got[key], want[key])
     trainer.fabric.close()
     open(ok_file, "w").write("ok")
+
+
+def test_transformers_cross_check_mistral(tmp_path):
+    """Mistral family (v0.3-style: no qkv bias, untied embeddings, no
+    sliding window): transformers' MistralForCausalLM must reproduce our
+    logits from our own checkpoint, and windowed (pre-v0.3) configs must
+    be rejected loudly instead of silently running full attention."""
+    d = str(tmp_path / "mistral_ckpt")
+    m = CausalLM(get_spec("tiny-mistral"), lora_r=0,
+                 dtype=torch.float32).random_init(31)
+    save_hf_checkpoint(m, d)
+    import json
+    with open(os.path.join(d, "config.json")) as f:
+        cfg = json.load(f)
+    assert cfg["architectures"] == ["MistralForCausalLM"]
+    assert cfg["sliding_window"] is None
+    from distrl_llm_amd.models.hf_io import spec_from_hf_config
+    s = spec_from_hf_config(d)
+    assert (s.hidden_size, s.num_kv_heads, s.qkv_bias) == (64, 2, False)
+    from transformers import AutoModelForCausalLM
+    hf = AutoModelForCausalLM.from_pretrained(d, torch_dtype=torch.float32)
+    assert type(hf).__name__ == "MistralForCausalLM"
+    ids = torch.randint(0, 2048, (2, 9))
+    torch.testing.assert_close(m(ids), hf(ids).logits, rtol=2e-4, atol=2e-4)
+    # pre-v0.3 windowed checkpoint (window < max_position, i.e. the
+    # window actually binds): rejected
+    cfg["sliding_window"] = 128
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg, f)
+    with pytest.raises(ValueError, match="sliding_window"):
+        spec_from_hf_config(d)
