@@ -601,9 +601,13 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, double scale) {
   // q: (B, Hq, T, D); k, v: (B, Hkv, T, D) — causal, bf16
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(k.scalar_type() == at::kBFloat16 &&
+              v.scalar_type() == at::kBFloat16);
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
   const int B = q.size(0), Hq = q.size(1), T = q.size(2), D = q.size(3);
   const int Hkv = k.size(1);
+  TORCH_CHECK(k.size(0) == B && k.size(2) == T && k.size(3) == D &&
+              v.sizes() == k.sizes(), "flash_attn_fwd: k/v shape mismatch");
   TORCH_CHECK(Hq % Hkv == 0);
   TORCH_CHECK(D == 64 || D == 128, "flash_attn_fwd: head_dim 64/128 only");
   auto qc = q.contiguous(), kc = k.contiguous();
@@ -635,8 +639,17 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                                           torch::Tensor o, torch::Tensor lse,
                                           double scale) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(dout.scalar_type() == at::kBFloat16 &&
+              k.scalar_type() == at::kBFloat16 &&
+              v.scalar_type() == at::kBFloat16 &&
+              o.scalar_type() == at::kBFloat16 &&
+              lse.scalar_type() == at::kFloat);
+  TORCH_CHECK(dout.sizes() == q.sizes() && o.sizes() == q.sizes(),
+              "flash_attn_bwd: dout/o shape mismatch");
   const int B = q.size(0), Hq = q.size(1), T = q.size(2), D = q.size(3);
   const int Hkv = k.size(1);
+  TORCH_CHECK(k.size(0) == B && k.size(2) == T && k.size(3) == D &&
+              v.sizes() == k.sizes(), "flash_attn_bwd: k/v shape mismatch");
   auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
   auto doc = dout.contiguous();
   const int Tp = (T + 31) / 32 * 32;
