@@ -106,7 +106,8 @@ class DynamicBatcher:
         # output caps batch together via the engine's per-sequence token
         # limits (in-wave retirement frees a short request's lanes while
         # longer ones keep decoding)
-        return (sp.temperature, sp.top_p, sp.top_k, sp.n, req.eos, nonce)
+        return (sp.temperature, sp.top_p, sp.top_k, sp.n, sp.geom_len_mean,
+                req.eos, nonce)
 
     def _loop(self):
         shutdown = False
