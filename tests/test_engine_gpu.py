@@ -253,3 +253,41 @@ def test_graph_cache_stable_across_weight_refresh():
     fresh = Engine(model, EngineConfig(**cfg), device=dev, seed=3)
     out_fresh = fresh.generate(prompts, sp, eos_token_id=None)
     assert out_cached == out_fresh, (out_cached, out_fresh)
+
+
+@pytest.mark.gpu
+def test_graph_cache_stable_across_refresh_nf4_hybrid():
+    """Same graph-stability contract through the QUANTIZED decode paths:
+    with nf4 sidecars attached, the 7B-class default runs merged bf16 +
+    the fused nf4 down-projection (hybrid_down), whose adapter fragment
+    packs are rebuilt per weight sync — the exact tensors whose per-round
+    reallocation caused the multi-wave batch-30 silent-NaN bug. The
+    in-place refresh (weights.py::_refresh_nf4_adapters) must keep the
+    captured graphs valid across syncs."""
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    spec = get_spec("small-qwen2")
+    dev = torch.device("cuda:0")
+    model = CausalLM(spec, lora_r=8, lora_alpha=16, dtype=torch.bfloat16,
+                     device=dev)
+    model.random_init(seed=22)
+    model.quantize_nf4_()   # sidecars on -> hybrid_down active
+    cfg = dict(max_seq_length=256, kv_block_size=16, num_kv_blocks=512,
+               max_num_seqs=64)
+    prompts = [[5, 9, 2, 7], list(range(30, 60)), [1000, 2000, 3000]]
+    sp = SamplingParams(max_tokens=10, temperature=0.0, n=2)
+
+    eng = Engine(model, EngineConfig(**cfg), device=dev, seed=3)
+    assert getattr(eng.fused, "hybrid_down", False), \
+        "test requires the hybrid nf4 down-proj decode path"
+    eng.generate(prompts, sp, eos_token_id=None)    # captures + caches
+    with torch.no_grad():                           # a "weight sync"
+        for m in model.modules():
+            if hasattr(m, "lora_B") and m.lora_B is not None:
+                m.lora_B.add_(torch.randn_like(m.lora_B) * 0.05)
+    out_cached = eng.generate(prompts, sp, eos_token_id=None)
+
+    fresh = Engine(model, EngineConfig(**cfg), device=dev, seed=3)
+    out_fresh = fresh.generate(prompts, sp, eos_token_id=None)
+    assert out_cached == out_fresh, (out_cached, out_fresh)
