@@ -28,9 +28,9 @@ _SHUTDOWN = object()
 
 class _Request:
     __slots__ = ("prompts", "sp", "eos", "event", "result", "error",
-                 "stream_q")
+                 "stream_q", "cancel_event")
 
-    def __init__(self, prompts, sp, eos, stream=False):
+    def __init__(self, prompts, sp, eos, stream=False, cancel_event=None):
         self.prompts = prompts
         self.sp = sp
         self.eos = eos
@@ -41,6 +41,9 @@ class _Request:
         # ("tok", prompt_i, cand_i, [ids...]) then ("done", result) or
         # ("err", exception)
         self.stream_q = queue.Queue() if stream else None
+        # optional abort signal (e.g. SSE client disconnect): once set,
+        # the engine retires this request's lanes with partial output
+        self.cancel_event = cancel_event
 
 
 class DynamicBatcher:
@@ -65,11 +68,14 @@ class DynamicBatcher:
     # ---------------------------------------------------------------- API
 
     def submit(self, prompts: List[List[int]], sp: SamplingParams,
-               eos_token_id: Optional[int] = None) -> List[List[List[int]]]:
-        """Blocking: returns the engine.generate result for ``prompts``."""
+               eos_token_id: Optional[int] = None,
+               cancel_event: Optional[threading.Event] = None
+               ) -> List[List[List[int]]]:
+        """Blocking: returns the engine.generate result for ``prompts``.
+        A set ``cancel_event`` aborts the request (partial outputs)."""
         if self._closed:
             raise RuntimeError("DynamicBatcher is closed")
-        req = _Request(prompts, sp, eos_token_id)
+        req = _Request(prompts, sp, eos_token_id, cancel_event=cancel_event)
         self._q.put(req)
         req.event.wait()
         if req.error is not None:
@@ -77,13 +83,18 @@ class DynamicBatcher:
         return req.result
 
     def submit_stream(self, prompts: List[List[int]], sp: SamplingParams,
-                      eos_token_id: Optional[int] = None) -> "queue.Queue":
+                      eos_token_id: Optional[int] = None,
+                      cancel_event: Optional[threading.Event] = None
+                      ) -> "queue.Queue":
         """Non-blocking: enqueue a streaming request and return its event
         queue. Items: ("tok", prompt_i, cand_i, [token_ids]) deltas, then
-        a final ("done", full_result) or ("err", exception)."""
+        a final ("done", full_result) or ("err", exception). Setting
+        ``cancel_event`` (client disconnect) aborts generation within one
+        decode chunk."""
         if self._closed:
             raise RuntimeError("DynamicBatcher is closed")
-        req = _Request(prompts, sp, eos_token_id, stream=True)
+        req = _Request(prompts, sp, eos_token_id, stream=True,
+                       cancel_event=cancel_event)
         self._q.put(req)
         return req.stream_q
 
@@ -161,10 +172,17 @@ class DynamicBatcher:
                 sp = dataclasses.replace(sp, max_tokens=max(caps))
                 limits = [[r.sp.max_tokens] * sp.n
                           for r in reqs for _ in r.prompts]
+            cancel_check = None
+            if any(r.cancel_event is not None for r in reqs):
+                # map merged prompt index -> owning request's abort event
+                events = [r.cancel_event for r in reqs for _ in r.prompts]
+                cancel_check = lambda pi: (events[pi] is not None
+                                           and events[pi].is_set())
             outs = self.engine.generate(merged, sp,
                                         eos_token_id=reqs[0].eos,
                                         stream_cb=cb,
-                                        token_limits=limits)
+                                        token_limits=limits,
+                                        cancel_check=cancel_check)
             off = 0
             for r in reqs:
                 r.result = outs[off:off + len(r.prompts)]

@@ -155,12 +155,18 @@ class DecodeSession:
 
     # -------------------------------------------------------------- run
 
-    def run(self, chunk: int = 16, stream_cb=None, retire_at=None):
+    def run(self, chunk: int = 16, stream_cb=None, retire_at=None,
+            stop_check=None):
         """Decode until every lane finishes, or — when ``retire_at`` is
         set — until at least that many lanes have finished (in-wave
         retirement: the engine then retires them, admits waiting prompts
         and re-waves the survivors; reference parity: vLLM continuous
         batching inside fast_generate, distributed_actor.py:147-172).
+
+        stop_check: optional zero-arg callable polled once per chunk —
+        returning True exits the wave early (request cancellation: the
+        engine then retires the cancelled lanes with partial output,
+        bounding abort latency to one chunk of steps).
 
         Returns (new_tokens_per_seq, finished_per_seq): only the tokens
         generated by THIS session (the caller extends q.output_ids)."""
@@ -198,6 +204,8 @@ class DecodeSession:
                 if n_fin >= self.finished.numel():  # incl. padded dummies
                     break
                 if retire_at is not None and n_fin >= retire_at:
+                    break
+                if stop_check is not None and stop_check():
                     break
 
         # ---- extraction: this session's new tokens + finished flags ----
@@ -322,12 +330,13 @@ class CachedDecodeSession(DecodeSession):
     # the retirement/extraction logic never sees them (finished count is
     # compared against n_pad via retire_at offsetting in SessionCache).
 
-    def run(self, chunk: int = 16, stream_cb=None, retire_at=None):
+    def run(self, chunk: int = 16, stream_cb=None, retire_at=None,
+            stop_check=None):
         # dummy lanes count as finished: shift the retirement threshold
         if retire_at is not None:
             retire_at = retire_at + (self.n_pad - len(self.seqs))
         return super().run(chunk=chunk, stream_cb=stream_cb,
-                           retire_at=retire_at)
+                           retire_at=retire_at, stop_check=stop_check)
 
 
 class SessionCache:

@@ -416,7 +416,8 @@ class Engine:
                  eos_token_id: Optional[int] = None,
                  prefill_token_budget: int = 8192,
                  stream_cb=None,
-                 token_limits: Optional[List[List[int]]] = None
+                 token_limits: Optional[List[List[int]]] = None,
+                 cancel_check=None
                  ) -> List[List[List[int]]]:
         """Generate sp.n completions per prompt.
 
@@ -433,6 +434,13 @@ class Engine:
         free their decode slots (the serving batcher uses this to merge
         requests with different max_tokens; the EOS-realistic bench mode
         draws them from an exponential via sp.geom_len_mean).
+
+        cancel_check: optional ``f(prompt_index) -> bool`` polled between
+        decode chunks (request abort, e.g. an SSE client disconnecting —
+        vLLM's ``abort_request`` analogue). A cancelled prompt's
+        candidates finish immediately with their partial outputs (or
+        ``[]`` if never admitted) and their KV blocks are freed; abort
+        latency is bounded by one decode chunk (16 steps).
         """
         was_training = self.model.training
         self.model.eval()
@@ -443,7 +451,7 @@ class Engine:
         try:
             return self._generate_inner(prompts, sp, eos_token_id,
                                         prefill_token_budget, stream_cb,
-                                        token_limits)
+                                        token_limits, cancel_check)
         except Exception:
             # a failure mid-generation strands this call's in-flight
             # sequences' KV blocks; generate calls are serialized, so no
@@ -456,7 +464,7 @@ class Engine:
                 self.model.train()
 
     def _generate_inner(self, prompts, sp, eos_token_id, prefill_token_budget,
-                        stream_cb=None, token_limits=None):
+                        stream_cb=None, token_limits=None, cancel_check=None):
         # token_limits: optional per-prompt lists of per-candidate output
         # caps (each <= sp.max_tokens) — the EOS-realistic bench mode and
         # per-request limits use these; admission still reserves the
@@ -511,6 +519,13 @@ class Engine:
         def try_admit():
             """Prefill + fork as many waiting prompts as memory allows
             (continuous batching admission)."""
+            if cancel_check is not None:
+                # purge cancelled prompts that never started: their
+                # candidates resolve to empty outputs (a prompt is
+                # admitted atomically, so all its slots are still None)
+                for pi in [p for p in waiting if cancel_check(p)]:
+                    waiting.remove(pi)
+                    results[pi] = [[] for _ in range(sp.n)]
             batch: List[Sequence] = []
             batch_tokens = 0
             reserved = future_need()
@@ -592,10 +607,16 @@ class Engine:
                         break
                 session = self._make_session(running, sp, eos_token_id)
                 retire_at = retire_unit if retire_unit > 0 else None
+                stop = None
+                if cancel_check is not None:
+                    wave = running
+                    stop = lambda: any(cancel_check(q.parent_prompt)
+                                       for q in wave)
                 try:
                     with trace_range(f"engine/decode_wave[{len(running)}]"):
                         new_toks, fin = session.run(stream_cb=stream_cb,
-                                                    retire_at=retire_at)
+                                                    retire_at=retire_at,
+                                                    stop_check=stop)
                 finally:
                     if self._session_cache is not None:
                         from .decode_session import CachedDecodeSession
@@ -604,6 +625,9 @@ class Engine:
                 still = []
                 for q, toks, f in zip(running, new_toks, fin):
                     q.output_ids.extend(toks)
+                    if cancel_check is not None and \
+                            cancel_check(q.parent_prompt):
+                        f = True  # aborted: partial output, free the KV
                     if f:
                         results[q.parent_prompt][q.cand_index] = q.output_ids
                         self._finish(q)
@@ -634,7 +658,9 @@ class Engine:
                     stream_cb(q.parent_prompt, q.cand_index, [t])
                 done = ((eos_token_id is not None and t == eos_token_id)
                         or len(q.output_ids) >= (q.max_tokens or sp.max_tokens)
-                        or q.total_len >= max_total)
+                        or q.total_len >= max_total
+                        or (cancel_check is not None
+                            and cancel_check(q.parent_prompt)))
                 if done:
                     results[q.parent_prompt][q.cand_index] = q.output_ids
                     self._finish(q)

@@ -164,10 +164,24 @@ def create_app(engine, tokenizer, model_name: str,
         """SSE generator over one streaming request (single prompt, n=1;
         per-token events on CPU, per-decode-chunk on GPU). Stop-string
         truncation is not applied mid-stream (tokens are emitted as
-        sampled)."""
+        sampled). Client disconnect aborts generation within one decode
+        chunk (the cancel event reaches the engine's cancel_check) so
+        dropped connections stop consuming decode slots."""
         import json as _json
+        import threading as _threading
         eos = getattr(tokenizer, "eos_token_id", None)
-        q = batcher.submit_stream(prompt_ids, sp, eos_token_id=eos)
+        cancel = _threading.Event()
+        q = batcher.submit_stream(prompt_ids, sp, eos_token_id=eos,
+                                  cancel_event=cancel)
+        try:
+            yield from _sse_events(q, rid, obj, delta_fn, endpoint, eos,
+                                   _json)
+        finally:
+            # normal completion: no-op (generation already finished);
+            # disconnect (GeneratorExit): aborts the in-flight request
+            cancel.set()
+
+    def _sse_events(q, rid, obj, delta_fn, endpoint, eos, _json):
         finish = "length"
         while True:
             kind, *rest = q.get()

@@ -170,3 +170,37 @@ def test_mixed_cap_stress_many_concurrent(engine):
     assert batcher.calls < 24
     # no leaked KV blocks after the storm
     assert engine.pool.allocator.num_free == engine.pool.num_blocks
+
+
+def test_stream_cancel_event_aborts_request(engine):
+    """A streaming request whose cancel_event fires mid-generation (SSE
+    client disconnect) finishes early with partial output; a co-batched
+    request is unaffected and the pool is fully freed."""
+    import queue as _queue
+    sp = SamplingParams(max_tokens=40, temperature=0.0, n=1)
+    expect_b = engine.generate([[4, 5, 6]], sp, eos_token_id=None)
+
+    batcher = DynamicBatcher(engine, max_wait_ms=80.0)
+    cancel = threading.Event()
+    try:
+        with ThreadPoolExecutor(max_workers=1) as ex:
+            fb = ex.submit(batcher.submit, [[4, 5, 6]], sp, None)
+            q = batcher.submit_stream([[1, 2, 3]], sp, eos_token_id=None,
+                                      cancel_event=cancel)
+            # wait for the first token delta, then "disconnect"
+            kind, *rest = q.get(timeout=60)
+            assert kind == "tok"
+            cancel.set()
+            result = None
+            while True:
+                kind, *rest = q.get(timeout=60)
+                if kind == "done":
+                    result = rest[0]
+                    break
+                assert kind == "tok"
+            rb = fb.result(timeout=60)
+        assert 1 <= len(result[0][0]) < 40   # aborted early
+        assert rb == expect_b                # co-batched request intact
+    finally:
+        batcher.close()
+    assert engine.pool.allocator.num_free == engine.pool.num_blocks

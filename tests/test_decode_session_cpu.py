@@ -140,3 +140,52 @@ def test_in_wave_retirement_and_per_seq_limits(tiny_engine_factory=None):
     for pi, per_prompt in enumerate(outs["eager"]):
         for ci, ids in enumerate(per_prompt):
             assert len(ids) == limits[pi][ci]  # greedy, no EOS -> exact cap
+
+
+def test_cancel_check_aborts_mid_generation(model, force_session):
+    """Request-abort support (vLLM abort_request analogue): a cancelled
+    prompt retires with partial output within one decode chunk and frees
+    its KV blocks; co-batched prompts are unaffected."""
+    eng = _engine(model)
+    sp = SamplingParams(max_tokens=40, temperature=0.0, n=1)
+    prompts = [[3, 1, 4], [2, 7, 2]]
+    full = eng.generate(prompts, sp, eos_token_id=None)
+
+    # cancel prompt 0 once it is in flight (first token streamed), as a
+    # disconnecting client would — an immediately-set cancel is purged
+    # pre-admission instead (covered below)
+    seen = []
+    out = eng.generate(prompts, sp, eos_token_id=None,
+                       stream_cb=lambda pi, ci, t: seen.append(pi),
+                       cancel_check=lambda pi: pi == 0 and 0 in seen)
+    # prompt 0 aborted: partial (prefill token + at most one chunk of 16)
+    assert 1 <= len(out[0][0]) <= 17 < 40
+    assert out[0][0] == full[0][0][:len(out[0][0])]  # prefix of greedy
+    # prompt 1 ran to its cap untouched
+    assert out[1][0] == full[1][0]
+    assert eng.pool.allocator.num_free == eng.pool.num_blocks
+
+
+def test_cancel_before_admission_returns_empty(model, force_session):
+    eng = _engine(model)
+    sp = SamplingParams(max_tokens=5, temperature=0.0, n=2)
+    out = eng.generate([[5, 6], [7, 8]], sp, eos_token_id=None,
+                       cancel_check=lambda pi: True)
+    assert out == [[[], []], [[], []]]
+    assert eng.pool.allocator.num_free == eng.pool.num_blocks
+
+
+def test_cancel_check_eager_path(model, monkeypatch):
+    monkeypatch.setenv("DISTRL_FORCE_SESSION", "0")
+    eng = _engine(model)
+    sp = SamplingParams(max_tokens=30, temperature=0.0, n=1)
+    prompts = [[3, 1, 4], [2, 7, 2]]
+    full = eng.generate(prompts, sp, eos_token_id=None)
+    seen = []
+    out = eng.generate(prompts, sp, eos_token_id=None,
+                       stream_cb=lambda pi, ci, t: seen.append(pi),
+                       cancel_check=lambda pi: pi == 0 and seen.count(0) >= 3)
+    assert 3 <= len(out[0][0]) < 30
+    assert out[0][0] == full[0][0][:len(out[0][0])]
+    assert out[1][0] == full[1][0]
+    assert eng.pool.allocator.num_free == eng.pool.num_blocks
