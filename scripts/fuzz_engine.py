@@ -75,11 +75,26 @@ for trial in range(args.trials):
     streamed = {}
     cb = (lambda pi, ci, toks: streamed.setdefault((pi, ci), []).extend(toks)) \
         if rng.random() < 0.5 else None
+    # sometimes: request cancellation (abort support) — cancel one prompt
+    # after it has streamed >= k tokens; its outputs must be PREFIXES of
+    # the uncancelled greedy result, other prompts stay exact
+    cancel_pi, cancel_after, cancel_check = None, 0, None
+    if rng.random() < 0.25:
+        cancel_pi = rng.randrange(len(prompts))
+        cancel_after = rng.randint(0, mt)
+        cstream = {}
+        ucb = cb
+        def cb(pi, ci, toks, _u=ucb):
+            cstream.setdefault(pi, []).extend(toks)
+            if _u is not None:
+                _u(pi, ci, toks)
+        cancel_check = (lambda pi: pi == cancel_pi
+                        and len(cstream.get(pi, [])) >= cancel_after)
     if trial < args.start:
         continue
     try:
         res = eng.generate(prompts, sp, eos_token_id=eos, stream_cb=cb,
-                           token_limits=limits)
+                           token_limits=limits, cancel_check=cancel_check)
     except MemoryError:
         continue
     ok = eng.pool.allocator.num_free == nb
@@ -90,7 +105,13 @@ for trial in range(args.trials):
         for ci, ids in enumerate(r):
             cap = limits[pi][ci] if limits is not None else mt
             exp = naive(p[:L], min(cap, msl - L), eos)
-            if ids != exp:
+            if pi == cancel_pi:
+                if ids != exp[:len(ids)]:
+                    fails += 1
+                    print("CANCEL-MISMATCH", trial, force, bs, nb, mt, n,
+                          msl, eos, p)
+                    break
+            elif ids != exp:
                 fails += 1
                 print("MISMATCH", trial, force, bs, nb, mt, n, msl, eos, p)
                 break
@@ -98,7 +119,11 @@ for trial in range(args.trials):
         # streamed deltas must prefix-match finals (session path may fork
         # results in cand order; greedy so all candidates identical)
         for (pi, ci), toks in streamed.items():
-            if toks != res[pi][ci]:
+            if pi == cancel_pi:
+                # aborted lanes may have unsent tokens at cancel time
+                if toks != res[pi][ci][:len(toks)]:
+                    fails += 1; print("STREAM-MISMATCH", trial, pi, ci)
+            elif toks != res[pi][ci]:
                 fails += 1; print("STREAM-MISMATCH", trial, pi, ci)
     if trial % 100 == 0:
         print("trial", trial, "ok", flush=True)
