@@ -201,3 +201,68 @@ def test_streaming_disconnect_aborts_generation(client):
                             "temperature": 0.0})
     assert out.status_code == 200
     assert out.json()["choices"][0]["text"] is not None
+
+
+def test_concurrent_mixed_storm(client):
+    """Production-shaped concurrency: blocking completions, chats,
+    seeded requests, streams that complete, and streams that disconnect
+    mid-generation — all in flight together. Every response must be
+    well-formed and the server must stay healthy throughout."""
+    import json as _json
+    from concurrent.futures import ThreadPoolExecutor
+
+    def blocking(i):
+        r = client.post("/v1/completions",
+                        json={"prompt": f"req {i}", "max_tokens": 6,
+                              "temperature": 0.0})
+        assert r.status_code == 200
+        return r.json()["choices"][0]["text"]
+
+    def chat(i):
+        r = client.post("/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": f"chat {i}"}],
+            "max_tokens": 5, "temperature": 0.0})
+        assert r.status_code == 200
+        return r.json()["choices"][0]["message"]["content"]
+
+    def seeded(i):
+        r = client.post("/v1/completions",
+                        json={"prompt": "seeded", "max_tokens": 5,
+                              "temperature": 0.9, "seed": 42})
+        assert r.status_code == 200
+        return r.json()["choices"][0]["text"]
+
+    def stream_full(i):
+        chunks = []
+        with client.stream("POST", "/v1/completions",
+                           json={"prompt": f"stream {i}", "max_tokens": 6,
+                                 "temperature": 0.0, "stream": True}) as r:
+            assert r.status_code == 200
+            for line in r.iter_lines():
+                if line.startswith("data:") and "[DONE]" not in line:
+                    chunks.append(_json.loads(line[5:]))
+        assert chunks
+        return "".join(c["choices"][0].get("text", "") for c in chunks)
+
+    def stream_abort(i):
+        with client.stream("POST", "/v1/completions",
+                           json={"prompt": f"abort {i}", "max_tokens": 100,
+                                 "temperature": 0.0, "stream": True}) as r:
+            assert r.status_code == 200
+            next(line for line in r.iter_lines()
+                 if line.startswith("data:"))
+        return "aborted"
+
+    jobs = []
+    with ThreadPoolExecutor(max_workers=12) as ex:
+        for i in range(4):
+            jobs += [ex.submit(blocking, i), ex.submit(chat, i),
+                     ex.submit(seeded, i), ex.submit(stream_full, i),
+                     ex.submit(stream_abort, i)]
+        results = [f.result(timeout=180) for f in jobs]
+    assert len(results) == 20
+    # seeded requests are deterministic even under the storm
+    seeds = [r for f, r in zip(jobs, results)][2::5]
+    assert all(s == seeds[0] for s in seeds)
+    # server healthy after the storm
+    assert client.get("/health").json() == {"status": "ok"}
