@@ -291,3 +291,27 @@ def test_graph_cache_stable_across_refresh_nf4_hybrid():
     fresh = Engine(model, EngineConfig(**cfg), device=dev, seed=3)
     out_fresh = fresh.generate(prompts, sp, eos_token_id=None)
     assert out_cached == out_fresh, (out_cached, out_fresh)
+
+
+@pytest.mark.gpu
+def test_cancel_check_aborts_on_gpu(setup):
+    """Request abort over the real hipGraph session path: stop_check is
+    polled between replay chunks, the cancelled prompt retires with a
+    greedy-prefix partial output, co-batched prompts are unaffected and
+    no KV blocks leak (CPU counterpart:
+    test_decode_session_cpu.py::test_cancel_check_aborts_mid_generation)."""
+    from distrl_llm_amd.config import SamplingParams
+    model, engine = setup
+    sp = SamplingParams(max_tokens=40, temperature=0.0, n=1)
+    prompts = [[3, 1, 4], [2, 7, 2]]
+    free0 = engine.pool.allocator.num_free
+    full = engine.generate(prompts, sp, eos_token_id=None)
+
+    seen = []
+    out = engine.generate(prompts, sp, eos_token_id=None,
+                          stream_cb=lambda pi, ci, t: seen.append(pi),
+                          cancel_check=lambda pi: pi == 0 and 0 in seen)
+    assert 1 <= len(out[0][0]) <= 17 < 40
+    assert out[0][0] == full[0][0][:len(out[0][0])]
+    assert out[1][0] == full[1][0]
+    assert engine.pool.allocator.num_free == free0
