@@ -122,6 +122,8 @@ class Engine:
             q = self._proj(layer.self_attn.q_proj, h).view(total, s.num_heads, s.head_dim)
             k = self._proj(layer.self_attn.k_proj, h).view(total, s.num_kv_heads, s.head_dim)
             v = self._proj(layer.self_attn.v_proj, h).view(total, s.num_kv_heads, s.head_dim)
+            if s.qk_norm:
+                q, k = self._qk_norm(layer.self_attn, q, k)
             q, k = OF.apply_rope_inplace(q, k, positions, self._inv_freq,
                                          s.rope_theta)
             OF.kv_cache_scatter(k, v, self.pool.key[li], self.pool.value[li],
@@ -137,6 +139,16 @@ class Engine:
         last_idx = torch.tensor([sum(lens[:i + 1]) - 1 for i in range(len(lens))],
                                 dtype=torch.long, device=device)
         return self.model.logits(x[last_idx])
+
+    def _qk_norm(self, at, q, k):
+        """Qwen3 per-head q/k RMSNorm over head_dim, applied before RoPE
+        (same op as HF Qwen3Attention.q_norm/k_norm)."""
+        D = self.spec.head_dim
+        q = OF.rmsnorm(q.reshape(-1, D), at.q_norm.weight,
+                       self.spec.rms_norm_eps).view_as(q)
+        k = OF.rmsnorm(k.reshape(-1, D), at.k_norm.weight,
+                       self.spec.rms_norm_eps).view_as(k)
+        return q, k
 
     def _prefill_attention(self, q, k, v, lens: List[int]) -> torch.Tensor:
         """Causal attention over concatenated prompts. GPU default: the
@@ -268,6 +280,8 @@ class Engine:
             q = self._proj(layer.self_attn.q_proj, h).view(N, s.num_heads, s.head_dim)
             k = self._proj(layer.self_attn.k_proj, h).view(N, s.num_kv_heads, s.head_dim)
             v = self._proj(layer.self_attn.v_proj, h).view(N, s.num_kv_heads, s.head_dim)
+            if s.qk_norm:
+                q, k = self._qk_norm(layer.self_attn, q, k)
             q, k = OF.apply_rope_inplace(q, k, positions, self._inv_freq,
                                          s.rope_theta)
             OF.kv_cache_scatter(k, v, self.pool.key[li], self.pool.value[li],
@@ -349,10 +363,26 @@ class Engine:
                                 getattr(lw, f"{site}_amax"), bias, None,
                                 None, N_out, K_in, 0)
 
+        def qk_norm_packed(qkv, li):
+            """Qwen3 per-head q/k RMSNorm applied in place on the packed
+            qkv GEMM output before the fused rope+scatter. Plain torch
+            ops (graph-capturable); fp32 internals match HF Qwen3. The
+            norm weights are engine-lifetime parameter storages, so
+            captured graphs stay valid across weight refreshes."""
+            at = m.model.layers[li].self_attn
+            D = s.head_dim
+            for sl, w in ((qkv[:, :qs], at.q_norm.weight),
+                          (qkv[:, qs:qs + kvs], at.k_norm.weight)):
+                hv = sl.unflatten(1, (-1, D)).float()
+                inv = hv.pow(2).mean(-1, keepdim=True).add(eps).rsqrt()
+                sl.copy_((hv * inv * w.float()).to(qkv.dtype).flatten(1))
+
         res = m.model.embed_tokens(input_ids)
         h = ext.rmsnorm_fwd(res, lws[0].in_norm, eps)
         for li, lw in enumerate(lws):
             qkv = proj(h, lw, li, "qkv", qs + 2 * kvs, s.hidden_size)
+            if s.qk_norm:
+                qk_norm_packed(qkv, li)
             ext.rope_scatter_qkv(qkv, pos32, slot_mapping, self._inv_freq,
                                  self.pool.key[li], self.pool.value[li],
                                  s.num_heads, s.num_kv_heads, s.head_dim)

@@ -43,7 +43,7 @@ def spec_from_hf_config(path: str, name: Optional[str] = None) -> ModelSpec:
         cfg = json.load(f)
     archs = cfg.get("architectures") or [""]
     arch = archs[0]
-    if not any(a in arch for a in ("Qwen2", "Llama", "Mistral")):
+    if not any(a in arch for a in ("Qwen2", "Qwen3", "Llama", "Mistral")):
         raise ValueError(f"unsupported architecture {arch!r} in {path}")
     sw = cfg.get("sliding_window")
     if arch.startswith("Mistral") and sw is not None \
@@ -70,9 +70,10 @@ def spec_from_hf_config(path: str, name: Optional[str] = None) -> ModelSpec:
         rms_norm_eps=float(cfg.get("rms_norm_eps", 1e-6)),
         tie_word_embeddings=bool(cfg.get("tie_word_embeddings", False)),
         # Llama exposes attention_bias (default False); Qwen2 always
-        # biases q/k/v
+        # biases q/k/v; Qwen3 drops the bias and adds per-head q/k norms
         qkv_bias=bool(cfg.get("attention_bias", arch.startswith("Qwen2"))),
         max_position=int(cfg.get("max_position_embeddings", 32768)),
+        qk_norm=arch.startswith("Qwen3"),
     )
 
 
@@ -138,7 +139,9 @@ def save_hf_checkpoint(model, path: str) -> None:
 
 
 def _write_hf_config(s: ModelSpec, path: str) -> None:
-    if s.qkv_bias:
+    if s.qk_norm:
+        arch, mtype = "Qwen3ForCausalLM", "qwen3"
+    elif s.qkv_bias:
         arch, mtype = "Qwen2ForCausalLM", "qwen2"
     elif "mistral" in s.name:
         arch, mtype = "MistralForCausalLM", "mistral"

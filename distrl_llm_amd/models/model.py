@@ -46,6 +46,13 @@ class Attention(nn.Module):
         self.k_proj = LoRALinear(h, kvs, spec.qkv_bias, r, alpha, dropout, dtype, device)
         self.v_proj = LoRALinear(h, kvs, spec.qkv_bias, r, alpha, dropout, dtype, device)
         self.o_proj = LoRALinear(qs, h, False, r, alpha, dropout, dtype, device)
+        if spec.qk_norm:
+            # Qwen3: per-head RMSNorm on q/k over head_dim, before RoPE
+            # (HF Qwen3Attention.q_norm/k_norm)
+            self.q_norm = RMSNorm(spec.head_dim, spec.rms_norm_eps, dtype,
+                                  device)
+            self.k_norm = RMSNorm(spec.head_dim, spec.rms_norm_eps, dtype,
+                                  device)
         self.scale = spec.head_dim ** -0.5
 
     def forward(self, x, cos, sin, attn_bias):
@@ -56,6 +63,9 @@ class Attention(nn.Module):
         q = self.q_proj(x).view(B, T, s.num_heads, s.head_dim)
         k = self.k_proj(x).view(B, T, s.num_kv_heads, s.head_dim)
         v = self.v_proj(x).view(B, T, s.num_kv_heads, s.head_dim)
+        if s.qk_norm:
+            q = self.q_norm(q.reshape(-1, s.head_dim)).view_as(q)
+            k = self.k_norm(k.reshape(-1, s.head_dim)).view_as(k)
         if (attn_bias is None and q.is_cuda and q.dtype == torch.bfloat16
                 and os.environ.get("DISTRL_ROPE_KERNEL") == "1"):
             # first-party RoPE fwd/bwd (fused HIP kernel; backward =

@@ -28,6 +28,8 @@ class ModelSpec:
     tie_word_embeddings: bool
     qkv_bias: bool  # Qwen2 uses biases on q/k/v projections
     max_position: int = 32768
+    # Qwen3: per-head RMSNorm on q/k (over head_dim, before RoPE)
+    qk_norm: bool = False
 
     @property
     def q_size(self) -> int:
@@ -60,6 +62,14 @@ _REGISTRY = {
     # checkpoints are rejected by hf_io.spec_from_hf_config)
     "mistral-7b": ModelSpec("mistral-7b", 32768, 4096, 14336, 32, 32, 8, 128,
                             1e6, 1e-5, False, False),
+    # Qwen3 dense (per-head q/k RMSNorm before RoPE, no qkv bias;
+    # note q_size != hidden for 4B: heads*head_dim = 32*128 = 4096)
+    "qwen3-4b": ModelSpec("qwen3-4b", 151936, 2560, 9728, 36, 32, 8, 128,
+                          1e6, 1e-6, True, False, qk_norm=True),
+    "qwen3-8b": ModelSpec("qwen3-8b", 151936, 4096, 12288, 36, 32, 8, 128,
+                          1e6, 1e-6, False, False, qk_norm=True),
+    "qwen3-32b": ModelSpec("qwen3-32b", 151936, 5120, 25600, 64, 64, 8, 128,
+                           1e6, 1e-6, False, False, qk_norm=True),
     # tiny models for CPU tests / the gloo plumbing config
     "tiny-qwen2": ModelSpec("tiny-qwen2", 2048, 64, 128, 2, 4, 2, 16,
                             1e4, 1e-6, True, True, max_position=512),
@@ -70,6 +80,10 @@ _REGISTRY = {
                             1e4, 1e-5, False, False, max_position=512),
     "tiny-mistral": ModelSpec("tiny-mistral", 2048, 64, 128, 2, 4, 2, 16,
                               1e4, 1e-5, False, False, max_position=512),
+    # tiny Qwen3 exercises qk_norm + q_size != hidden (4 heads x 24 = 96)
+    "tiny-qwen3": ModelSpec("tiny-qwen3", 2048, 64, 128, 2, 4, 2, 24,
+                            1e4, 1e-6, False, False, max_position=512,
+                            qk_norm=True),
 }
 
 
@@ -77,7 +91,8 @@ def get_spec(model_name: str) -> ModelSpec:
     """Resolve a model-name string (e.g. the reference's default
     'unsloth/Qwen2.5-7B-Instruct-bnb-4bit') to an architecture spec."""
     low = model_name.lower()
-    for key in ("tiny-qwen2", "tiny-llama", "tiny-mistral", "small-qwen2"):
+    for key in ("tiny-qwen2", "tiny-llama", "tiny-mistral", "tiny-qwen3",
+                "small-qwen2"):
         if key in low:
             return _REGISTRY[key]
     if "qwen2.5-0.5b" in low or "qwen2-0.5b" in low:
@@ -98,6 +113,10 @@ def get_spec(model_name: str) -> ModelSpec:
         return _REGISTRY["llama-3-70b"]
     if "mistral" in low and "7b" in low:
         return _REGISTRY["mistral-7b"]
+    if "qwen3" in low:
+        for size in ("4b", "8b", "32b"):
+            if size in low:
+                return _REGISTRY[f"qwen3-{size}"]
     raise ValueError(f"Unknown model architecture for name: {model_name!r}; "
                      f"known: {sorted(_REGISTRY)}")
 

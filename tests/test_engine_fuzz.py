@@ -167,3 +167,49 @@ def test_mistral_family_greedy_equals_naive(prompts, max_tokens):
 
 
 _MISTRAL = {}
+
+
+@settings(max_examples=8, deadline=None)
+@given(
+    prompts=st.lists(
+        st.lists(st.integers(min_value=1, max_value=500), min_size=1,
+                 max_size=20),
+        min_size=1, max_size=4),
+    max_tokens=st.integers(min_value=1, max_value=6),
+)
+def test_qwen3_family_greedy_equals_naive(prompts, max_tokens):
+    """The Qwen3 architecture branch (per-head q/k RMSNorm before RoPE,
+    q_size != hidden) through the same engine paths — including the
+    session state machine (DISTRL_FORCE_SESSION flips inside the run)."""
+    import os
+    if "m" not in _QWEN3:
+        m = CausalLM(get_spec("tiny-qwen3"), lora_r=4, lora_alpha=8,
+                     dtype=torch.float32)
+        m.random_init(seed=79)
+        with torch.no_grad():  # non-trivial norms
+            for layer in m.model.layers:
+                at = layer.self_attn
+                at.q_norm.weight.add_(
+                    torch.rand(at.q_norm.weight.shape) * 0.5 - 0.25)
+                at.k_norm.weight.add_(
+                    torch.rand(at.k_norm.weight.shape) * 0.5 - 0.25)
+        _QWEN3["m"] = m
+    m = _QWEN3["m"]
+    cfg = EngineConfig(max_seq_length=64, kv_block_size=8, num_kv_blocks=128,
+                       max_num_seqs=32)
+    for force in ("0", "1"):
+        os.environ["DISTRL_FORCE_SESSION"] = force
+        try:
+            engine = Engine(m, cfg, device=torch.device("cpu"), seed=0)
+            sp = SamplingParams(max_tokens=max_tokens, temperature=0.0, n=2)
+            results = engine.generate(prompts, sp, eos_token_id=None)
+            for p, res in zip(prompts, results):
+                expected = _naive_greedy(m, p, max_tokens)
+                for ids in res:
+                    assert ids == expected, f"force={force}"
+            assert engine.pool.allocator.num_free == 128
+        finally:
+            os.environ.pop("DISTRL_FORCE_SESSION", None)
+
+
+_QWEN3 = {}

@@ -257,3 +257,72 @@ def test_transformers_cross_check_mistral(tmp_path):
         json.dump(cfg, f)
     with pytest.raises(ValueError, match="sliding_window"):
         spec_from_hf_config(d)
+
+
+def test_transformers_cross_check_qwen3(tmp_path):
+    """Qwen3 family: per-head q/k RMSNorm before RoPE, no qkv bias, and
+    q_size != hidden_size (heads*head_dim decoupled from hidden).
+    transformers' Qwen3ForCausalLM must reproduce our logits from our
+    own checkpoint — with q/k norm weights perturbed away from 1 so the
+    extra norms are exercised non-trivially."""
+    d = str(tmp_path / "qwen3_ckpt")
+    m = CausalLM(get_spec("tiny-qwen3"), lora_r=0,
+                 dtype=torch.float32).random_init(41)
+    assert m.spec.q_size != m.spec.hidden_size  # 4*24=96 vs 64
+    g = torch.Generator().manual_seed(7)
+    with torch.no_grad():
+        for layer in m.model.layers:
+            at = layer.self_attn
+            at.q_norm.weight.add_(torch.rand(at.q_norm.weight.shape,
+                                             generator=g) - 0.5)
+            at.k_norm.weight.add_(torch.rand(at.k_norm.weight.shape,
+                                             generator=g) - 0.5)
+    save_hf_checkpoint(m, d)
+    import json
+    with open(os.path.join(d, "config.json")) as f:
+        cfg = json.load(f)
+    assert cfg["architectures"] == ["Qwen3ForCausalLM"]
+    from distrl_llm_amd.models.hf_io import spec_from_hf_config
+    s = spec_from_hf_config(d)
+    assert s.qk_norm and not s.qkv_bias and s.head_dim == 24
+    from transformers import AutoModelForCausalLM
+    hf = AutoModelForCausalLM.from_pretrained(d, torch_dtype=torch.float32)
+    assert type(hf).__name__ == "Qwen3ForCausalLM"
+    ids = torch.randint(0, 2048, (2, 11))
+    torch.testing.assert_close(m(ids), hf(ids).logits, rtol=2e-4, atol=2e-4)
+
+
+def test_engine_generation_matches_transformers_qwen3(tmp_path):
+    """End-to-end Qwen3 decode cross-validation: the engine's prefill +
+    paged decode (with the per-head q/k norms applied before RoPE) must
+    agree token-for-token with transformers' KV-cached generate."""
+    from transformers import AutoModelForCausalLM
+
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+
+    d = str(tmp_path / "qwen3_gen")
+    m = CausalLM(get_spec("tiny-qwen3"), lora_r=0,
+                 dtype=torch.float32).random_init(43)
+    g = torch.Generator().manual_seed(9)
+    with torch.no_grad():
+        for layer in m.model.layers:
+            at = layer.self_attn
+            at.q_norm.weight.add_(torch.rand(at.q_norm.weight.shape,
+                                             generator=g) - 0.5)
+            at.k_norm.weight.add_(torch.rand(at.k_norm.weight.shape,
+                                             generator=g) - 0.5)
+    save_hf_checkpoint(m, d)
+    hf = AutoModelForCausalLM.from_pretrained(d, dtype=torch.float32)
+    eng = Engine(m, EngineConfig(max_seq_length=128, kv_block_size=8,
+                                 num_kv_blocks=256, max_num_seqs=16),
+                 device=torch.device("cpu"), seed=0)
+    prompts = [[1, 5, 9, 2, 7], [17] * 12, list(range(2, 25))]
+    ours = eng.generate(prompts, SamplingParams(max_tokens=8,
+                                                temperature=0.0, n=1),
+                        eos_token_id=None)
+    for p, o in zip(prompts, ours):
+        with torch.no_grad():
+            out = hf.generate(torch.tensor([p]), max_new_tokens=8,
+                              do_sample=False, use_cache=True)
+        assert o[0] == out[0, len(p):].tolist()

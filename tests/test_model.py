@@ -183,6 +183,9 @@ def test_spec_registry_resolution_and_shapes():
         "meta-llama/Meta-Llama-3-8B": (4096, 32, 32, 8),
         "meta-llama/Meta-Llama-3-70B-Instruct": (8192, 80, 64, 8),
         "mistralai/Mistral-7B-Instruct-v0.3": (4096, 32, 32, 8),
+        "Qwen/Qwen3-8B": (4096, 36, 32, 8),
+        "Qwen/Qwen3-4B-Instruct-2507": (2560, 36, 32, 8),
+        "Qwen/Qwen3-32B": (5120, 64, 64, 8),
     }
     for name, (h, L, q, kv) in cases.items():
         s = get_spec(name)
