@@ -73,9 +73,12 @@ _REGISTRY = {
     # tiny models for CPU tests / the gloo plumbing config
     "tiny-qwen2": ModelSpec("tiny-qwen2", 2048, 64, 128, 2, 4, 2, 16,
                             1e4, 1e-6, True, True, max_position=512),
-    # small model for GPU kernel/engine tests (head_dim 64 = kernel-supported)
+    # small models for GPU kernel/engine tests (head_dim 64 = kernel-supported)
     "small-qwen2": ModelSpec("small-qwen2", 4096, 512, 1024, 2, 8, 2, 64,
                              1e5, 1e-6, True, True, max_position=2048),
+    "small-qwen3": ModelSpec("small-qwen3", 4096, 512, 1024, 2, 8, 2, 64,
+                             1e5, 1e-6, True, False, max_position=2048,
+                             qk_norm=True),
     "tiny-llama": ModelSpec("tiny-llama", 2048, 64, 128, 2, 4, 2, 16,
                             1e4, 1e-5, False, False, max_position=512),
     "tiny-mistral": ModelSpec("tiny-mistral", 2048, 64, 128, 2, 4, 2, 16,
@@ -92,7 +95,7 @@ def get_spec(model_name: str) -> ModelSpec:
     'unsloth/Qwen2.5-7B-Instruct-bnb-4bit') to an architecture spec."""
     low = model_name.lower()
     for key in ("tiny-qwen2", "tiny-llama", "tiny-mistral", "tiny-qwen3",
-                "small-qwen2"):
+                "small-qwen2", "small-qwen3"):
         if key in low:
             return _REGISTRY[key]
     if "qwen2.5-0.5b" in low or "qwen2-0.5b" in low:

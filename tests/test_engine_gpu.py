@@ -315,3 +315,38 @@ def test_cancel_check_aborts_on_gpu(setup):
     assert out[0][0] == full[0][0][:len(out[0][0])]
     assert out[1][0] == full[1][0]
     assert engine.pool.allocator.num_free == free0
+
+
+@pytest.mark.gpu
+def test_qwen3_qk_norm_gpu_decode():
+    """Qwen3 family on the real GPU decode path: the per-head q/k norms
+    run as in-place torch ops on the packed qkv buffer inside the
+    hipGraph-captured step (engine.py qk_norm_packed). Greedy engine
+    output must match the naive layer-by-layer forward (which applies
+    the norms through models/model.py — a different code path)."""
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    dev = torch.device("cuda:0")
+    model = CausalLM(get_spec("small-qwen3"), lora_r=8, lora_alpha=16,
+                     dtype=torch.bfloat16, device=dev)
+    model.random_init(seed=33)
+    with torch.no_grad():  # non-trivial norms
+        for layer in model.model.layers:
+            at = layer.self_attn
+            at.q_norm.weight.add_(
+                (torch.rand_like(at.q_norm.weight.float()) * 0.5 - 0.25)
+                .to(at.q_norm.weight.dtype))
+            at.k_norm.weight.add_(
+                (torch.rand_like(at.k_norm.weight.float()) * 0.5 - 0.25)
+                .to(at.k_norm.weight.dtype))
+    eng = Engine(model, EngineConfig(max_seq_length=256, kv_block_size=16,
+                                     num_kv_blocks=512, max_num_seqs=64),
+                 device=dev, seed=0)
+    prompts = [[5, 9, 2, 7], list(range(30, 50))]
+    sp = SamplingParams(max_tokens=8, temperature=0.0, n=2)
+    out = eng.generate(prompts, sp, eos_token_id=None)
+    for p, res in zip(prompts, out):
+        exp = _naive_greedy(model, p, 8)
+        for ids in res:
+            assert ids == exp
