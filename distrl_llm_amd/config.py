@@ -83,3 +83,10 @@ class EngineConfig:
     # from free memory * gpu_memory_utilization.
     num_kv_blocks: int = 0
     enforce_eager: bool = False  # True disables hipGraph capture of decode
+    # Automatic cross-request prefix caching (vLLM APC analogue): full
+    # prompt blocks are kept in the pool keyed by their token prefix and
+    # reused by later requests sharing the prefix (classic win: a long
+    # system prompt + short user tails). Cache-hit tails prefill through
+    # the decode step (existing kernels only); LRU-evicted under pool
+    # pressure. Off by default: RL rollouts use unique prompts.
+    enable_prefix_caching: bool = False

@@ -17,6 +17,7 @@ broadcast into these tensors.
 from __future__ import annotations
 
 import os
+from collections import OrderedDict
 from typing import List, Optional
 
 import torch
@@ -41,6 +42,13 @@ class Engine:
         self.scale = self.spec.head_dim ** -0.5
         self._seq_counter = 0
         self._session_cache = None  # DISTRL_GRAPH_CACHE=1 lazily creates
+        # automatic prefix caching (cfg.enable_prefix_caching): LRU map
+        # {tuple(prompt[:k*block_size]): block_id}; the cache holds its
+        # own refcount on every entry so blocks survive sequence finish
+        self._prefix_cache = (OrderedDict()
+                              if cfg.enable_prefix_caching else None)
+        self._prefix_hits = 0   # blocks reused (tests/metrics)
+        self._prefix_evicts = 0
         self.generator = torch.Generator(device=self.device)
         self.generator.manual_seed(seed)
 
@@ -93,7 +101,129 @@ class Engine:
     def _prefill_batch(self, seqs: List[Sequence]) -> torch.Tensor:
         """Run the prompt phase for a batch of fresh sequences (one per
         prompt; fan-out happens after). Writes prompt KV into the pool and
-        returns last-token logits (len(seqs), V)."""
+        returns last-token logits (len(seqs), V).
+
+        With ``enable_prefix_caching`` on, sequences whose prompt starts
+        with already-cached full blocks reuse those blocks (incref, no
+        recompute) and prefill only the tail — through the DECODE step,
+        so no new attention kernel is involved (classic serving win: a
+        long shared system prompt + short per-user tails). Full blocks of
+        every prefilled prompt are then inserted into the LRU cache."""
+        if self._prefix_cache is None:
+            return self._prefill_flat(seqs)
+        bs = self.pool.block_size
+        hits, misses, logits = [], [], {}
+        for q in seqs:
+            ids = q.prompt_ids
+            blocks = []
+            # cap at (L-1)//bs blocks so >= 1 tail token yields logits
+            for i in range((len(ids) - 1) // bs):
+                key = tuple(ids[:(i + 1) * bs])
+                blk = self._prefix_cache.get(key)
+                if blk is None:
+                    break
+                blocks.append(blk)
+                self._prefix_cache.move_to_end(key)
+            if blocks:
+                for b in blocks:
+                    self.pool.allocator.incref(b)
+                q.block_table = list(blocks)
+                q.context_len = len(blocks) * bs
+                self._prefix_hits += len(blocks)
+                hits.append(q)
+            else:
+                misses.append(q)
+        if misses:
+            lg = self._prefill_flat(misses)
+            for i, q in enumerate(misses):
+                logits[id(q)] = lg[i:i + 1]
+        if hits:
+            lg = self._prefill_tail(hits)
+            for i, q in enumerate(hits):
+                logits[id(q)] = lg[i:i + 1]
+        self._prefix_insert(seqs)
+        return torch.cat([logits[id(q)] for q in seqs], 0)
+
+    def _prefill_tail(self, seqs: List[Sequence]) -> torch.Tensor:
+        """Teacher-force the uncached prompt tail of cache-hit sequences
+        through the decode step (KV scatter + paged attention against the
+        cached context — existing, GPU-validated kernels only). Returns
+        last-token logits per sequence. Sequential over the longest tail;
+        the cache trades that for skipping the cached prefix entirely."""
+        device = self.device
+        bs = self.pool.block_size
+        out: List[Optional[torch.Tensor]] = [None] * len(seqs)
+        active = list(range(len(seqs)))
+        while active:
+            tokens, positions, slots, ctx = [], [], [], []
+            for i in active:
+                q = seqs[i]
+                pos = q.context_len
+                tokens.append(q.prompt_ids[pos])
+                positions.append(pos)
+                bi = pos // bs
+                if bi == len(q.block_table):
+                    q.block_table.append(self.pool.allocator.alloc())
+                slots.append(q.block_table[bi] * bs + pos % bs)
+                ctx.append(pos + 1)
+            max_nb = max(len(seqs[i].block_table) for i in active)
+            block_tables = torch.zeros(len(active), max_nb,
+                                       dtype=torch.int32, device=device)
+            for row, i in enumerate(active):
+                bt = seqs[i].block_table
+                block_tables[row, :len(bt)] = torch.tensor(
+                    bt, dtype=torch.int32)
+            lg = self._decode_forward(
+                torch.tensor(tokens, dtype=torch.long, device=device),
+                torch.tensor(positions, dtype=torch.long, device=device),
+                torch.tensor(slots, dtype=torch.long, device=device),
+                block_tables,
+                torch.tensor(ctx, dtype=torch.int32, device=device))
+            nxt = []
+            for row, i in enumerate(active):
+                q = seqs[i]
+                q.context_len += 1
+                if q.context_len >= len(q.prompt_ids):
+                    out[i] = lg[row:row + 1]
+                else:
+                    nxt.append(i)
+            active = nxt
+        return torch.cat([o for o in out], 0)
+
+    def _prefix_insert(self, seqs: List[Sequence]) -> None:
+        """Insert every fully-written prompt block into the LRU cache
+        (the cache takes its own refcount, so blocks outlive their
+        sequences). Duplicate keys keep the first block. Note: evicting
+        a shorter-prefix entry orphans its longer continuations until
+        LRU reclaims them — lookups walk prefixes in order."""
+        bs = self.pool.block_size
+        for q in seqs:
+            ids = q.prompt_ids
+            for i in range(len(ids) // bs):
+                key = tuple(ids[:(i + 1) * bs])
+                if key in self._prefix_cache:
+                    self._prefix_cache.move_to_end(key)
+                    continue
+                blk = q.block_table[i]
+                self.pool.allocator.incref(blk)
+                self._prefix_cache[key] = blk
+
+    def _evict_prefix(self, target_free: int) -> None:
+        """Drop LRU cache entries until the allocator has target_free
+        blocks (or the cache is empty). Freeing only drops the cache's
+        own refcount — blocks shared with live sequences stay alive."""
+        while (self._prefix_cache
+               and self.pool.allocator.num_free < target_free):
+            _, blk = self._prefix_cache.popitem(last=False)
+            self.pool.allocator.free(blk)
+            self._prefix_evicts += 1
+
+    def clear_prefix_cache(self) -> None:
+        if self._prefix_cache:
+            self._evict_prefix(self.pool.num_blocks + 1)
+
+    def _prefill_flat(self, seqs: List[Sequence]) -> torch.Tensor:
+        """Full varlen prefill from scratch (no cached context)."""
         s = self.spec
         device = self.device
         lens = [len(q.prompt_ids) for q in seqs]
@@ -488,6 +618,10 @@ class Engine:
             # other sequence is live — reset the pool instead of leaking
             # (long-lived serving processes would otherwise exhaust it)
             self.pool.allocator.reset()
+            if self._prefix_cache:
+                # the reset dropped every refcount; cached block ids are
+                # stale — start the cache over rather than dangle
+                self._prefix_cache.clear()
             raise
         finally:
             if was_training:
@@ -563,6 +697,9 @@ class Engine:
                 pi = waiting[0]
                 need = blocks_needed(pi)
                 L = min(len(prompts[pi]), max_total - 1)
+                if self._prefix_cache:
+                    # reclaim LRU-cached prefix blocks before refusing
+                    self._evict_prefix(need + reserved + len(batch))
                 if need > self.pool.allocator.num_free - reserved - len(batch):
                     break
                 reserved += need

@@ -60,13 +60,25 @@ for trial in range(args.trials):
         else:
             p = [rng.randint(1, 500) for _ in range(rng.randint(1, 30))]
         prompts.append(p)
+    # sometimes: automatic prefix caching, warmed with a prefix of the
+    # shared base so the main call's shared-prefix prompts hit the cache
+    use_cache = rng.random() < 0.3
     cfg = EngineConfig(max_seq_length=msl, kv_block_size=bs, num_kv_blocks=nb,
-                       max_num_seqs=rng.choice([8, 16, 64]))
+                       max_num_seqs=rng.choice([8, 16, 64]),
+                       enable_prefix_caching=use_cache)
     eng = Engine(model, cfg, device=dev, seed=trial)
     # adversarial EOS: the naive first token of prompt 0
     eos = naive(prompts[0][:min(len(prompts[0]), msl - 1)], 1)[0] \
         if rng.random() < 0.5 else None
     sp = SamplingParams(max_tokens=mt, temperature=0.0, n=n)
+    if use_cache and trial >= args.start:
+        warm = base[:rng.randint(2, len(base))]
+        try:
+            eng.generate([warm], SamplingParams(max_tokens=1,
+                                                temperature=0.0, n=1),
+                         eos_token_id=None)
+        except MemoryError:
+            continue
     # sometimes: per-candidate output caps (token_limits -> in-wave
     # retirement paths); naive() then compares against each cap
     limits = None
@@ -97,9 +109,14 @@ for trial in range(args.trials):
                            token_limits=limits, cancel_check=cancel_check)
     except MemoryError:
         continue
-    ok = eng.pool.allocator.num_free == nb
+    held = len(eng._prefix_cache) if use_cache else 0
+    ok = eng.pool.allocator.num_free == nb - held
     if not ok:
         fails += 1; print("LEAK", trial); continue
+    if use_cache:
+        eng.clear_prefix_cache()
+        if eng.pool.allocator.num_free != nb:
+            fails += 1; print("CACHE-LEAK", trial); continue
     for pi, (p, r) in enumerate(zip(prompts, res)):
         L = min(len(p), msl - 1)
         for ci, ids in enumerate(r):
