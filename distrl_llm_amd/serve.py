@@ -336,6 +336,9 @@ def main():
     ap.add_argument("--gpu-memory-utilization", type=float, default=0.9)
     ap.add_argument("--adapter", type=str, default=None,
                     help="PEFT adapter directory to apply before serving")
+    ap.add_argument("--enable-prefix-caching", action="store_true",
+                    help="reuse KV blocks of shared prompt prefixes "
+                         "across requests (vLLM APC analogue)")
     args = ap.parse_args()
 
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
@@ -358,7 +361,8 @@ def main():
     tokenizer = load_tokenizer(args.model, spec.vocab_size)
     engine = Engine(model, EngineConfig(
         max_seq_length=args.max_seq_length,
-        gpu_memory_utilization=args.gpu_memory_utilization),
+        gpu_memory_utilization=args.gpu_memory_utilization,
+        enable_prefix_caching=args.enable_prefix_caching),
         device=device, seed=args.seed)
 
     import uvicorn
