@@ -350,3 +350,33 @@ def test_qwen3_qk_norm_gpu_decode():
         exp = _naive_greedy(model, p, 8)
         for ids in res:
             assert ids == exp
+
+
+@pytest.mark.gpu
+def test_prefix_caching_gpu_equality(setup):
+    """Automatic prefix caching on the GPU: cache-hit tails prefill
+    through the fused decode step; outputs must equal the uncached
+    engine's, and clear_prefix_cache must return every block."""
+    from distrl_llm_amd.config import EngineConfig, SamplingParams
+    from distrl_llm_amd.engine import Engine
+    model, _ = setup
+    dev = torch.device("cuda:0")
+    cfg = dict(max_seq_length=256, kv_block_size=16, num_kv_blocks=512,
+               max_num_seqs=64)
+    sys_p = list(range(700, 740))  # 40-token shared prefix (2 full blocks)
+    prompts_a = [sys_p + [7, 8, 9], sys_p + [1]]
+    prompts_b = [sys_p + [4, 4], sys_p + [2, 3, 5]]
+    sp = SamplingParams(max_tokens=8, temperature=0.0, n=2)
+
+    plain = Engine(model, EngineConfig(**cfg), device=dev, seed=3)
+    exp_a = plain.generate(prompts_a, sp, eos_token_id=None)
+    exp_b = plain.generate(prompts_b, sp, eos_token_id=None)
+
+    eng = Engine(model, EngineConfig(enable_prefix_caching=True, **cfg),
+                 device=dev, seed=3)
+    assert eng.generate(prompts_a, sp, eos_token_id=None) == exp_a
+    hits0 = eng._prefix_hits
+    assert eng.generate(prompts_b, sp, eos_token_id=None) == exp_b
+    assert eng._prefix_hits > hits0  # the shared prefix was reused
+    eng.clear_prefix_cache()
+    assert eng.pool.allocator.num_free == eng.pool.num_blocks
