@@ -266,3 +266,41 @@ def test_concurrent_mixed_storm(client):
     assert all(s == seeds[0] for s in seeds)
     # server healthy after the storm
     assert client.get("/health").json() == {"status": "ok"}
+
+
+def test_serving_with_prefix_caching():
+    """A cache-enabled server handling chat traffic with a shared system
+    prompt: responses equal the uncached server's, and the cache is
+    actually hit on the repeated prefix."""
+    from fastapi.testclient import TestClient
+
+    from distrl_llm_amd.config import EngineConfig
+    from distrl_llm_amd.engine import Engine
+    from distrl_llm_amd.models import CausalLM, get_spec
+    from distrl_llm_amd.serve import create_app
+    from distrl_llm_amd.utils.tokenizer import ByteTokenizer
+
+    spec = get_spec("tiny-qwen2")
+    model = CausalLM(spec, lora_r=0, dtype=torch.float32).random_init(5)
+    tok = ByteTokenizer(vocab_size=spec.vocab_size)
+    cfg = dict(max_seq_length=128, kv_block_size=8, num_kv_blocks=256,
+               max_num_seqs=32)
+    plain_eng = Engine(model, EngineConfig(**cfg),
+                       device=torch.device("cpu"), seed=0)
+    cached_eng = Engine(model, EngineConfig(enable_prefix_caching=True,
+                                            **cfg),
+                        device=torch.device("cpu"), seed=0)
+    plain = TestClient(create_app(plain_eng, tok, "tiny-qwen2"))
+    cached = TestClient(create_app(cached_eng, tok, "tiny-qwen2"))
+
+    system = ("You are a terse assistant. Answer in one word whenever "
+              "possible and never apologize.")
+    for user in ("hi there", "what is 2+2?", "name a color"):
+        body = {"messages": [{"role": "system", "content": system},
+                             {"role": "user", "content": user}],
+                "max_tokens": 6, "temperature": 0.0}
+        a = plain.post("/v1/chat/completions", json=body).json()
+        b = cached.post("/v1/chat/completions", json=body).json()
+        assert (a["choices"][0]["message"]["content"]
+                == b["choices"][0]["message"]["content"])
+    assert cached_eng._prefix_hits > 0  # the system prompt was reused
