@@ -20,7 +20,8 @@ def test_serve_main_boots_and_serves(tmp_path):
     proc = subprocess.Popen(
         [sys.executable, "-m", "distrl_llm_amd.serve",
          "--model", "tiny-qwen2", "--host", "127.0.0.1",
-         "--port", str(port), "--max-seq-length", "128"],
+         "--port", str(port), "--max-seq-length", "128",
+         "--enable-prefix-caching"],
         cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
         text=True)
     base = f"http://127.0.0.1:{port}"
